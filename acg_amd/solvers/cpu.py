@@ -1,0 +1,180 @@
+"""Host (CPU) CG solvers — the reference oracle path.
+
+Reference: acg/cg.c — acgsolver_solve / acgsolver_solvempi: textbook CG
+with a halo exchange before each SpMV and an all-reduce per dot
+(cg.c:408-649).  Unlike the GPU solvers it also supports the
+diff_atol/diff_rtol stopping criteria.  solve_pipelined mirrors the
+Ghysels–Vanroose recurrences used by the GPU pipelined solver so the
+algorithm is testable without a GPU.
+
+Runs on torch CPU tensors (gloo for multi-rank tests); the same code
+drives the ops in acg_amd.ops.torch_ref that the HIP kernels are tested
+against.
+"""
+
+from __future__ import annotations
+
+import math
+import time
+
+import numpy as np
+import torch
+
+from ..dist.halo import HaloExchange
+from ..ops import torch_ref as ops
+from ..part.subdomain import LocalSystem
+from .base import SolveResult, cg_flops_per_iter
+
+
+class CGSolverCPU:
+    """Distributed classic / pipelined CG on host tensors."""
+
+    def __init__(self, local: LocalSystem, comm=None, device="cpu"):
+        self.local = local
+        self.comm = comm
+        self.device = torch.device(device)
+        L = local
+        self.A_rowptr = torch.from_numpy(np.ascontiguousarray(L.A_rowptr)).to(self.device)
+        self.A_colidx = torch.from_numpy(np.ascontiguousarray(L.A_colidx)).to(self.device)
+        self.A_vals = torch.from_numpy(np.ascontiguousarray(L.A_vals)).to(self.device)
+        self.O_rowptr = torch.from_numpy(np.ascontiguousarray(L.O_rowptr)).to(self.device)
+        self.O_colidx = torch.from_numpy(np.ascontiguousarray(L.O_colidx)).to(self.device)
+        self.O_vals = torch.from_numpy(np.ascontiguousarray(L.O_vals)).to(self.device)
+        self.halo = HaloExchange(L.halo, L.nowned, self.device, comm)
+        self.n = L.nowned
+        self.nlocal = L.nowned + L.nghost
+
+    # -- helpers ----------------------------------------------------------
+
+    def _allreduce(self, t: torch.Tensor) -> torch.Tensor:
+        if self.comm is not None:
+            self.comm.allreduce_(t)
+        return t
+
+    def _dot(self, a: torch.Tensor, b: torch.Tensor) -> float:
+        v = torch.dot(a[: self.n], b[: self.n]).reshape(1)
+        return float(self._allreduce(v)[0])
+
+    def _spmv(self, xfull: torch.Tensor, y: torch.Tensor) -> None:
+        """y = A_local x (halo exchange + split matA/matO SpMV)."""
+        self.halo.exchange(xfull)
+        ops.spmv(self.A_rowptr, self.A_colidx, self.A_vals, xfull, y)
+        ops.spmv(self.O_rowptr, self.O_colidx, self.O_vals, xfull, y,
+                 rowbase=self.local.ninterior, accum=True)
+
+    def _vec(self, nghost: bool = False) -> torch.Tensor:
+        return torch.zeros(self.nlocal if nghost else self.n,
+                           dtype=torch.float64, device=self.device)
+
+    # -- classic CG (reference acgsolver_solvempi, cg.c:408-649) ----------
+
+    def solve(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
+              res_atol: float = 0.0, res_rtol: float = 1e-9,
+              diff_atol: float = 0.0, diff_rtol: float = 0.0) -> SolveResult:
+        res = SolveResult(solver="cg-cpu", maxits=maxits, res_atol=res_atol,
+                          res_rtol=res_rtol,
+                          nranks=self.comm.size if self.comm else 1)
+        n = self.n
+        t0 = time.perf_counter()
+        bnrm2 = math.sqrt(self._dot(b, b))
+        res.bnrm2 = bnrm2
+        r = self._vec()
+        t = self._vec()
+        p = self._vec(nghost=True)
+        # r0 = b - A x0
+        self._spmv(x, t)
+        r[:] = b[:n] - t
+        p[:n] = r
+        rr = self._dot(r, r)
+        res.r0nrm2 = math.sqrt(rr)
+        rtol2 = max(res_atol, res_rtol * bnrm2) ** 2
+        dtol = max(diff_atol, diff_rtol * bnrm2)
+        if rr <= rtol2 and (dtol == 0.0):
+            res.converged = True
+            res.rnrm2 = math.sqrt(rr)
+            res.tsolve = time.perf_counter() - t0
+            return res
+        for k in range(maxits):
+            self._spmv(p, t)
+            pt = self._dot(p, t)
+            alpha = rr / pt
+            r -= alpha * t
+            x[:n] += alpha * p[:n]
+            rr_new = self._dot(r, r)
+            res.niterations = k + 1
+            converged = False
+            if rtol2 > 0 and rr_new <= rtol2:
+                converged = True
+            if dtol > 0:
+                dx = abs(alpha) * math.sqrt(self._dot(p, p))
+                if dx <= dtol:
+                    converged = True
+            if converged:
+                res.converged = True
+                res.rnrm2 = math.sqrt(rr_new)
+                break
+            beta = rr_new / rr
+            p[:n] = r + beta * p[:n]
+            rr = rr_new
+            res.rnrm2 = math.sqrt(rr_new)
+        res.tsolve = time.perf_counter() - t0
+        nnz_full = self.local.nnzA + 2 * self.local.nnzO
+        res.nflops = res.niterations * cg_flops_per_iter(nnz_full, n)
+        res.halo_bytes_sent = self.halo.bytes_sent
+        res.halo_msgs_sent = self.halo.nmsgs_sent
+        return res
+
+    # -- pipelined CG (reference acgsolverhip_solve_pipelined, §3.3) ------
+
+    def solve_pipelined(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
+                        res_atol: float = 0.0, res_rtol: float = 1e-9) -> SolveResult:
+        res = SolveResult(solver="cg-pipelined-cpu", maxits=maxits,
+                          res_atol=res_atol, res_rtol=res_rtol,
+                          nranks=self.comm.size if self.comm else 1)
+        n = self.n
+        t0 = time.perf_counter()
+        bnrm2 = math.sqrt(self._dot(b, b))
+        res.bnrm2 = bnrm2
+        tmp = self._vec()
+        r = self._vec(nghost=True)
+        w = self._vec(nghost=True)
+        q = self._vec()
+        z = self._vec()
+        t = self._vec()
+        p = self._vec()
+        self._spmv(x, tmp)
+        r[:n] = b[:n] - tmp
+        self._spmv(r, w)  # w = A r  (writes w[:n])
+        rtol2 = max(res_atol, res_rtol * bnrm2) ** 2
+        gamma_prev = alpha_prev = None
+        for k in range(maxits):
+            gd = torch.stack([torch.dot(r[:n], r[:n]), torch.dot(w[:n], r[:n])])
+            self._allreduce(gd)
+            gamma, delta = float(gd[0]), float(gd[1])
+            if k == 0:
+                res.r0nrm2 = math.sqrt(gamma)
+            res.rnrm2 = math.sqrt(gamma)
+            if rtol2 > 0 and gamma <= rtol2:
+                res.converged = True
+                res.niterations = k
+                break
+            self._spmv(w, q)  # q = A w (halo on w inside)
+            if k == 0:
+                beta, alpha = 0.0, gamma / delta
+            else:
+                beta = gamma / gamma_prev
+                alpha = gamma / (delta - beta * gamma / alpha_prev)
+            z[:] = q + beta * z
+            t[:] = w[:n] + beta * t
+            p[:] = r[:n] + beta * p
+            x[:n] += alpha * p
+            r[:n] -= alpha * t
+            w[:n] -= alpha * z
+            gamma_prev, alpha_prev = gamma, alpha
+            res.niterations = k + 1
+        res.tsolve = time.perf_counter() - t0
+        nnz_full = self.local.nnzA + 2 * self.local.nnzO
+        res.nflops = res.niterations * (cg_flops_per_iter(nnz_full, n) + 8.0 * n)
+        res.halo_bytes_sent = self.halo.bytes_sent
+        res.halo_msgs_sent = self.halo.nmsgs_sent
+        return res

@@ -1,0 +1,159 @@
+"""End-to-end CPU CG tests (reference test strategy: SURVEY.md §4 --
+manufactured solutions + independent scipy oracle on 5-pt Poisson)."""
+
+import numpy as np
+import pytest
+import torch
+
+from acg_amd.gen import STENCIL_5PT_2D, STENCIL_27PT_3D, queen_like_spec, stencil_global
+from acg_amd.part import extract_subdomains, partition_rows
+from acg_amd.solvers import CGSolverCPU
+
+
+def _build(nx=32, ny=32, nparts=1, spec=STENCIL_5PT_2D, gz=1, method="block"):
+    A = stencil_global(nx, ny, gz, spec)
+    part = partition_rows(A, nparts, method=method)
+    systems = extract_subdomains(A, part, nparts)
+    return A, systems
+
+
+def test_poisson_cg_serial_vs_scipy():
+    A, systems = _build(32, 32, 1)
+    S = systems[0]
+    n = S.nowned
+    rng = np.random.default_rng(0)
+    xsol = rng.standard_normal(A.n)
+    b_global = A.dsymv(xsol)
+    b = torch.from_numpy(b_global[S.owned_global])
+    x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64)
+    solver = CGSolverCPU(S, comm=None)
+    res = solver.solve(b, x, maxits=2000, res_rtol=1e-10)
+    assert res.converged, res.summary()
+    err = np.linalg.norm(x[:n].numpy() - xsol[S.owned_global]) / np.linalg.norm(xsol)
+    assert err < 1e-7, err
+
+
+def test_poisson_cg_pipelined_matches_classic():
+    A, systems = _build(24, 24, 1)
+    S = systems[0]
+    rng = np.random.default_rng(1)
+    b_np = rng.standard_normal(A.n)
+    b = torch.from_numpy(b_np[S.owned_global])
+    solver = CGSolverCPU(S, comm=None)
+    x1 = torch.zeros(S.nowned + S.nghost, dtype=torch.float64)
+    r1 = solver.solve(b.clone(), x1, maxits=500, res_rtol=1e-10)
+    x2 = torch.zeros(S.nowned + S.nghost, dtype=torch.float64)
+    r2 = solver.solve_pipelined(b.clone(), x2, maxits=500, res_rtol=1e-10)
+    assert r1.converged and r2.converged
+    # pipelined CG is algebraically equivalent; allow fp jitter
+    assert abs(r1.niterations - r2.niterations) <= 3
+    np.testing.assert_allclose(x1[:S.nowned].numpy(), x2[:S.nowned].numpy(),
+                               rtol=1e-6, atol=1e-8)
+
+
+@pytest.mark.parametrize("method", ["block", "rgb"])
+def test_partitioned_serial_consistency(method):
+    """Multi-part extraction run serially (all parts in one process) must
+    reproduce the global SpMV exactly."""
+    A, systems = _build(16, 16, 4, method=method)
+    n = A.n
+    rng = np.random.default_rng(2)
+    xg = rng.standard_normal(n)
+    yg = A.dsymv(xg)
+    # emulate the halo by hand: fill ghost tails from the global vector
+    for S in systems:
+        xl = np.concatenate([xg[S.owned_global], xg[S.ghost_global]])
+        import acg_amd.ops.torch_ref as ops
+
+        xt = torch.from_numpy(xl)
+        yt = torch.zeros(S.nowned, dtype=torch.float64)
+        ops.spmv(torch.from_numpy(S.A_rowptr), torch.from_numpy(S.A_colidx),
+                 torch.from_numpy(S.A_vals), xt, yt)
+        ops.spmv(torch.from_numpy(S.O_rowptr), torch.from_numpy(S.O_colidx),
+                 torch.from_numpy(S.O_vals), xt, yt,
+                 rowbase=S.ninterior, accum=True)
+        np.testing.assert_allclose(yt.numpy(), yg[S.owned_global], rtol=1e-12, atol=1e-12)
+
+
+def test_subdomain_send_recv_pairing():
+    """Sender block ordering must match receiver ghost ordering."""
+    A, systems = _build(16, 16, 4)
+    for S in systems:
+        h = S.halo
+        for qi, q in enumerate(h.recipients):
+            R = systems[q]
+            # R's ghosts owned by S.rank
+            gmask = np.isin(R.ghost_global, S.owned_global)
+            gset = R.ghost_global[gmask]
+            owners = np.array([systems[S.rank].owned_global[i]
+                               for i in h.sendidx[h.sdispls[qi]:h.sdispls[qi] + h.sendcounts[qi]]])
+            # what S sends to q == q's ghosts from S, in q's tail order
+            si = np.where(np.asarray(R.halo.senders) == S.rank)[0][0]
+            lo = int(R.halo.rdispls[si])
+            hi = lo + int(R.halo.recvcounts[si])
+            np.testing.assert_array_equal(owners, R.ghost_global[lo:hi])
+
+
+def test_queen_like_spec_shape():
+    spec = queen_like_spec(dof=3)
+    A = stencil_global(6, 6, 6, spec)
+    n = 6 * 6 * 6 * 3
+    assert A.n == n
+    F = A.to_full_csr()
+    # interior rows have 27 neighbours x 3 dof = 81 nonzeros (Queen_4147: ~79)
+    counts = np.diff(F.rowptr)
+    assert counts.max() == 27 * 3
+    # SPD sanity: diagonally dominant
+    X = A.to_scipy_full()
+    d = X.diagonal()
+    offsum = np.abs(X).sum(axis=1).A1 - np.abs(d)
+    assert np.all(d > offsum * 0.99)
+
+
+@pytest.mark.parametrize("spec,g,nranks", [
+    (STENCIL_5PT_2D, (12, 12, 1), 1),
+    (STENCIL_27PT_3D, (6, 6, 8), 3),
+    (queen_like_spec(3), (5, 5, 7), 2),
+])
+def test_slab_generator_matches_global(spec, g, nranks):
+    """stencil_local_slab must produce the same operator as the global
+    assembly + hand-filled halo."""
+    from acg_amd.gen import stencil_local_slab
+    import acg_amd.ops.torch_ref as ops
+
+    gx, gy, gz = g
+    A = stencil_global(gx, gy, gz, spec)
+    rng = np.random.default_rng(3)
+    xg = rng.standard_normal(A.n)
+    yg = A.dsymv(xg)
+    covered = np.zeros(A.n, dtype=bool)
+    for rank in range(nranks):
+        S = stencil_local_slab(gx, gy, gz, spec, rank, nranks)
+        xl = np.concatenate([xg[S.owned_global], xg[S.ghost_global]])
+        xt = torch.from_numpy(xl)
+        yt = torch.zeros(S.nowned, dtype=torch.float64)
+        ops.spmv(torch.from_numpy(S.A_rowptr), torch.from_numpy(S.A_colidx.astype(np.int64)),
+                 torch.from_numpy(S.A_vals), xt, yt)
+        ops.spmv(torch.from_numpy(S.O_rowptr), torch.from_numpy(S.O_colidx.astype(np.int64)),
+                 torch.from_numpy(S.O_vals), xt, yt, rowbase=S.ninterior, accum=True)
+        np.testing.assert_allclose(yt.numpy(), yg[S.owned_global], rtol=1e-12, atol=1e-10)
+        assert not covered[S.owned_global].any()
+        covered[S.owned_global] = True
+    assert covered.all()
+
+
+def test_slab_halo_pairing():
+    from acg_amd.gen import stencil_local_slab
+
+    spec = queen_like_spec(3)
+    systems = [stencil_local_slab(5, 5, 9, spec, r, 3) for r in range(3)]
+    for S in systems:
+        h = S.halo
+        for qi, q in enumerate(h.recipients):
+            R = systems[q]
+            sl = h.sendidx[int(h.sdispls[qi]):int(h.sdispls[qi]) + int(h.sendcounts[qi])]
+            sent_globals = S.owned_global[np.asarray(sl, dtype=np.int64)]
+            si = np.where(np.asarray(R.halo.senders) == S.rank)[0][0]
+            lo = int(R.halo.rdispls[si])
+            hi = lo + int(R.halo.recvcounts[si])
+            np.testing.assert_array_equal(sent_globals, R.ghost_global[lo:hi])
