@@ -118,7 +118,7 @@ class CGSolverCPU:
             rr = rr_new
             res.rnrm2 = math.sqrt(rr_new)
         res.tsolve = time.perf_counter() - t0
-        nnz_full = self.local.nnzA + 2 * self.local.nnzO
+        nnz_full = self.local.nnzA + self.local.nnzO
         res.nflops = res.niterations * cg_flops_per_iter(nnz_full, n)
         res.halo_bytes_sent = self.halo.bytes_sent
         res.halo_msgs_sent = self.halo.nmsgs_sent
@@ -173,7 +173,7 @@ class CGSolverCPU:
             gamma_prev, alpha_prev = gamma, alpha
             res.niterations = k + 1
         res.tsolve = time.perf_counter() - t0
-        nnz_full = self.local.nnzA + 2 * self.local.nnzO
+        nnz_full = self.local.nnzA + self.local.nnzO
         res.nflops = res.niterations * (cg_flops_per_iter(nnz_full, n) + 8.0 * n)
         res.halo_bytes_sent = self.halo.bytes_sent
         res.halo_msgs_sent = self.halo.nmsgs_sent

@@ -1,0 +1,129 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: distributed CG iterations/s on Queen_4147-shaped SPD.
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W` (launched
+via torch.distributed.run for N>1, one rank per GPU over RCCL).  Does W
+untimed warmup CG iterations, then times EXACTLY K iterations bracketed by
+barrier + torch.cuda.synchronize on both sides, takes the MAX time over
+ranks, and rank 0 prints ONE JSON line.
+
+Metric (BASELINE.json): CG iter/s (whole node) on Queen_4147-shaped data
+(4.1M rows / ~330M nnz fp64; synthetic 27-pt dof-3 stencil -- no network,
+no SuiteSparse downloads), strong scaling across 1/2/4/8 MI355X.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+
+def main() -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=20)
+    ap.add_argument("--solver", choices=["pipelined", "classic"], default="pipelined")
+    ap.add_argument("--grid", type=int, default=111,
+                    help="grid edge G; rows = 3*G^3 (111 -> 4.10M rows, Queen_4147 scale)")
+    ap.add_argument("--dof", type=int, default=3)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    ngpus = max(world, 1)
+
+    from acg_amd.dist.comm import Comm
+    from acg_amd.gen import queen_like_spec, stencil_local_slab
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    device = torch.device("cuda", local_rank % max(torch.cuda.device_count(), 1))
+    torch.cuda.set_device(device)
+    comm = Comm("rccl", device=device) if world > 1 else None
+
+    G = args.grid
+    spec = queen_like_spec(args.dof)
+    S = stencil_local_slab(G, G, G, spec, rank, ngpus)
+    solver = CGSolverHIP(S, comm=comm, device=device)
+
+    rloc = np.random.default_rng(10_000 + rank)
+    b = torch.from_numpy(rloc.standard_normal(S.nowned)).to(device)
+    x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device=device)
+
+    solve = solver.solve_pipelined if args.solver == "pipelined" else solver.solve
+
+    # warmup (untimed; also JITs RCCL channels and fills caches)
+    if args.warmup > 0:
+        solve(b, x.clone(), maxits=args.warmup, res_rtol=0.0)
+
+    if comm is not None:
+        comm.barrier()
+    torch.cuda.synchronize(device)
+    t0 = time.perf_counter()
+    res = solve(b, x, maxits=args.steps, res_rtol=0.0)
+    torch.cuda.synchronize(device)
+    if comm is not None:
+        comm.barrier()
+    t1 = time.perf_counter()
+    elapsed = t1 - t0
+    # MAX over ranks
+    if comm is not None:
+        et = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        import torch.distributed as dist
+
+        dist.all_reduce(et, op=dist.ReduceOp.MAX)
+        elapsed = float(et.item())
+
+    assert res.niterations == args.steps, (res.niterations, args.steps)
+    nrows_global = args.dof * G * G * G
+    nnz_local = S.nnzA + S.nnzO
+    if comm is not None:
+        import torch.distributed as dist
+
+        nt = torch.tensor([float(nnz_local)], dtype=torch.float64, device=device)
+        dist.all_reduce(nt, op=dist.ReduceOp.SUM)
+        nnz_global = float(nt.item())
+    else:
+        nnz_global = float(nnz_local)
+
+    iters_per_s = args.steps / elapsed
+    ms_per_step = 1000.0 * elapsed / args.steps
+    if rank == 0:
+        out = {
+            "metric": "CG iter/s (whole node), Queen_4147-shaped fp64",
+            "value": iters_per_s,
+            "unit": "iter/s",
+            "n_gpus": ngpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": None,
+            "dtype": "fp64",
+            "data": "synthetic SPD (27-pt dof-3 stencil, Queen_4147 shape; random RHS)",
+            "config": {
+                "model": f"queen4147-like-27pt-dof{args.dof}-G{G}",
+                "rows": nrows_global,
+                "nnz": nnz_global,
+                "solver": f"cg-{args.solver}",
+                "time_to_solution_s": elapsed,
+                "gflops": args.steps * (2.0 * nnz_global + 10.0 * nrows_global) / elapsed / 1e9,
+                "parallelism": f"slab{ngpus}-rccl",
+            },
+        }
+        print(json.dumps(out), flush=True)
+    if comm is not None:
+        comm.finalize()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

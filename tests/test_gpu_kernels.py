@@ -1,0 +1,165 @@
+"""GPU kernel numerics: every gfx950 HIP kernel vs the plain-torch fp64
+reference (ops.torch_ref), per the test strategy in SURVEY.md §4."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _rand_csr(nrows, ncols, nnz_per_row, seed=0, col64=False):
+    rng = np.random.default_rng(seed)
+    counts = rng.integers(1, 2 * nnz_per_row, size=nrows)
+    rowptr = np.zeros(nrows + 1, dtype=np.int64)
+    np.cumsum(counts, out=rowptr[1:])
+    nnz = int(rowptr[-1])
+    cols = rng.integers(0, ncols, size=nnz)
+    # sort within rows
+    rows = np.repeat(np.arange(nrows), counts)
+    order = np.lexsort((cols, rows))
+    cols = cols[order]
+    vals = rng.standard_normal(nnz)
+    cdt = np.int64 if col64 else np.int32
+    return (torch.from_numpy(rowptr), torch.from_numpy(cols.astype(cdt)),
+            torch.from_numpy(vals))
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    return torch.device("cuda:0")
+
+
+@pytest.mark.parametrize("lanes", [4, 8, 16, 32, 64])
+@pytest.mark.parametrize("col64", [False, True])
+def test_spmv_vector(dev, lanes, col64):
+    from acg_amd.ops import gpu_ops, torch_ref
+
+    rowptr, colidx, vals = _rand_csr(5000, 6000, 40, seed=lanes, col64=col64)
+    x = torch.randn(6000, dtype=torch.float64)
+    y_ref = torch.zeros(5000, dtype=torch.float64)
+    torch_ref.spmv(rowptr, colidx, vals, x, y_ref)
+    yg = torch.zeros(5000, dtype=torch.float64, device=dev)
+    gpu_ops.spmv(rowptr.to(dev), colidx.to(dev), vals.to(dev), x.to(dev), yg,
+                 lanes=lanes)
+    torch.testing.assert_close(yg.cpu(), y_ref, rtol=1e-12, atol=1e-10)
+
+
+def test_spmv_accum_rowbase_fused_dot(dev):
+    from acg_amd.ops import gpu_ops, torch_ref
+
+    n = 4000
+    rowptr, colidx, vals = _rand_csr(1000, n, 30, seed=7)
+    x = torch.randn(n, dtype=torch.float64)
+    y0 = torch.randn(n, dtype=torch.float64)
+    scal_ref = torch_ref.alloc_scalars()
+    y_ref = y0.clone()
+    torch_ref.spmv(rowptr, colidx, vals, x, y_ref, rowbase=500, accum=True,
+                   scal=scal_ref, dotslot=torch_ref.S_PT)
+    scal = gpu_ops.alloc_scalars(dev)
+    yg = y0.to(dev)
+    gpu_ops.spmv(rowptr.to(dev), colidx.to(dev), vals.to(dev), x.to(dev), yg,
+                 rowbase=500, accum=True, lanes=8, scal=scal, dotslot=gpu_ops.S_PT)
+    torch.testing.assert_close(yg.cpu(), y_ref, rtol=1e-12, atol=1e-10)
+    torch.testing.assert_close(scal.cpu()[gpu_ops.S_PT], scal_ref[torch_ref.S_PT],
+                               rtol=1e-10, atol=1e-10)
+
+
+def test_dot_and_dot2(dev):
+    from acg_amd.ops import gpu_ops, torch_ref
+
+    n = 100_003
+    r = torch.randn(n, dtype=torch.float64)
+    w = torch.randn(n, dtype=torch.float64)
+    sr = torch_ref.alloc_scalars()
+    torch_ref.dot(r, w, sr, torch_ref.S_PT)
+    torch_ref.dot2(r, w, sr, n)
+    sg = gpu_ops.alloc_scalars(dev)
+    gpu_ops.dot(r.to(dev), w.to(dev), sg, gpu_ops.S_PT)
+    gpu_ops.dot2(r.to(dev), w.to(dev), sg, n)
+    torch.testing.assert_close(sg.cpu(), sr, rtol=1e-10, atol=1e-8)
+
+
+def test_fused_update_and_daypx(dev):
+    from acg_amd.ops import gpu_ops, torch_ref
+
+    n = 50_001
+    g = torch.Generator().manual_seed(5)
+    r = torch.randn(n, dtype=torch.float64, generator=g)
+    x = torch.randn(n, dtype=torch.float64, generator=g)
+    p = torch.randn(n, dtype=torch.float64, generator=g)
+    t = torch.randn(n, dtype=torch.float64, generator=g)
+    sr = torch_ref.alloc_scalars()
+    sr[torch_ref.S_RR] = 3.7
+    sr[torch_ref.S_PT] = 1.9
+    sg = sr.clone().to(dev)
+    rr, xr, pr = r.clone(), x.clone(), p.clone()
+    torch_ref.cg_prep_rr(sr)
+    torch_ref.cg_fused_update(rr, xr, pr, t, sr, n)
+    torch_ref.daypx_ratio(pr, rr, sr, torch_ref.S_RR, torch_ref.S_RR_PREV)
+    rg, xg, pg, tg = r.to(dev), x.to(dev), p.to(dev), t.to(dev)
+    gpu_ops.cg_prep_rr(sg)
+    gpu_ops.cg_fused_update(rg, xg, pg, tg, sg, n)
+    gpu_ops.daypx_ratio(pg, rg, sg, gpu_ops.S_RR, gpu_ops.S_RR_PREV)
+    torch.testing.assert_close(rg.cpu(), rr, rtol=1e-12, atol=1e-12)
+    torch.testing.assert_close(xg.cpu(), xr, rtol=1e-12, atol=1e-12)
+    torch.testing.assert_close(pg.cpu(), pr, rtol=1e-12, atol=1e-12)
+    torch.testing.assert_close(sg.cpu()[gpu_ops.S_RR], sr[torch_ref.S_RR],
+                               rtol=1e-10, atol=1e-8)
+
+
+@pytest.mark.parametrize("first", [True, False])
+def test_pipelined_fused(dev, first):
+    from acg_amd.ops import gpu_ops, torch_ref
+
+    n = 30_000
+    g = torch.Generator().manual_seed(9)
+    vecs = {k: torch.randn(n, dtype=torch.float64, generator=g)
+            for k in "ztpxrwq"}
+    sr = torch_ref.alloc_scalars()
+    sr[torch_ref.S_GAMMA] = 2.1
+    sr[torch_ref.S_DELTA] = 4.3
+    sr[torch_ref.S_GAMMA_PREV] = 1.7
+    sr[torch_ref.S_ALPHA_PREV] = 0.9
+    sg = sr.clone().to(dev)
+    ref = {k: v.clone() for k, v in vecs.items()}
+    torch_ref.pipelined_fused(ref["z"], ref["t"], ref["p"], ref["x"], ref["r"],
+                              ref["w"], ref["q"], sr, n, first)
+    torch_ref.pipelined_reset(sr, first)
+    gv = {k: v.to(dev) for k, v in vecs.items()}
+    gpu_ops.pipelined_fused(gv["z"], gv["t"], gv["p"], gv["x"], gv["r"],
+                            gv["w"], gv["q"], sg, n, first)
+    gpu_ops.pipelined_reset(sg, first)
+    for k in "ztpxrw":
+        torch.testing.assert_close(gv[k].cpu(), ref[k], rtol=1e-12, atol=1e-12,
+                                   msg=f"vec {k}")
+    torch.testing.assert_close(sg.cpu(), sr, rtol=1e-12, atol=1e-12)
+
+
+def test_pack_gather(dev):
+    from acg_amd.ops import gpu_ops, torch_ref
+
+    n = 10000
+    x = torch.randn(n, dtype=torch.float64)
+    idx = torch.randint(0, n, (777,), dtype=torch.int32)
+    ref = torch.empty(777, dtype=torch.float64)
+    torch_ref.pack_gather(ref, x, idx)
+    out = torch.empty(777, dtype=torch.float64, device=dev)
+    gpu_ops.pack_gather(out, x.to(dev), idx.to(dev))
+    torch.testing.assert_close(out.cpu(), ref)
+
+
+def test_axpy_ratio(dev):
+    from acg_amd.ops import gpu_ops, torch_ref
+
+    n = 12345
+    y = torch.randn(n, dtype=torch.float64)
+    x = torch.randn(n, dtype=torch.float64)
+    s = torch_ref.alloc_scalars()
+    s[0], s[1] = 2.5, 0.5
+    yr = y.clone()
+    torch_ref.axpy_ratio(yr, x, s, 0, 1, sign=-1.0)
+    yg = y.to(dev)
+    gpu_ops.axpy_ratio(yg, x.to(dev), s.to(dev), 0, 1, sign=-1.0)
+    torch.testing.assert_close(yg.cpu(), yr, rtol=1e-14, atol=1e-14)
