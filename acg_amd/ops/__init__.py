@@ -17,8 +17,9 @@ class _LazyGpuOps:
 
     def _load(self):
         if self._mod is None:
-            from . import gpu_ops as m
-            self._mod = m
+            import importlib
+
+            self._mod = importlib.import_module("acg_amd.ops.gpu_ops")
         return self._mod
 
     def __getattr__(self, name):

@@ -33,6 +33,8 @@ def main() -> int:
     ap.add_argument("--grid", type=int, default=111,
                     help="grid edge G; rows = 3*G^3 (111 -> 4.10M rows, Queen_4147 scale)")
     ap.add_argument("--dof", type=int, default=3)
+    ap.add_argument("--lanes", type=int, default=None,
+                    help="override SpMV lanes-per-row (4/8/16/32/64)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -51,7 +53,7 @@ def main() -> int:
     G = args.grid
     spec = queen_like_spec(args.dof)
     S = stencil_local_slab(G, G, G, spec, rank, ngpus)
-    solver = CGSolverHIP(S, comm=comm, device=device)
+    solver = CGSolverHIP(S, comm=comm, device=device, lanes=args.lanes)
 
     rloc = np.random.default_rng(10_000 + rank)
     b = torch.from_numpy(rloc.standard_normal(S.nowned)).to(device)
