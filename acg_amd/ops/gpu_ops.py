@@ -3,6 +3,10 @@
 Raw pointers + the caller's current HIP stream are passed to the
 extension; all type/shape checks happen here.  If the extension is
 missing on a machine with a GPU, import fails loudly (no eager fallback).
+
+Reductions are two-phase (per-block partials + finalize kernel): callers
+pass a ``partials`` scratch tensor of size 2*MAXG allocated once via
+:func:`alloc_partials`.
 """
 
 from __future__ import annotations
@@ -28,19 +32,19 @@ S_DELTA = K.S_DELTA
 S_GAMMA_PREV = K.S_GAMMA_PREV
 S_ALPHA_PREV = K.S_ALPHA_PREV
 S_NSLOTS = K.S_NSLOTS
+MAXG = K.MAXG
 
 
 def _stream() -> int:
     return torch.cuda.current_stream().cuda_stream
 
 
-def _chk(t: torch.Tensor, dtype, name: str):
-    assert t.is_cuda and t.dtype == dtype and t.is_contiguous(), \
-        f"{name}: need contiguous cuda {dtype}, got {t.device} {t.dtype} contig={t.is_contiguous()}"
-
-
 def alloc_scalars(device) -> torch.Tensor:
     return torch.zeros(S_NSLOTS, dtype=torch.float64, device=device)
+
+
+def alloc_partials(device) -> torch.Tensor:
+    return torch.zeros(2 * MAXG, dtype=torch.float64, device=device)
 
 
 def pick_lanes(mean_nnz_per_row: float) -> int:
@@ -58,20 +62,38 @@ def pick_lanes(mean_nnz_per_row: float) -> int:
 def spmv(rowptr: torch.Tensor, colidx: torch.Tensor, vals: torch.Tensor,
          x: torch.Tensor, y: torch.Tensor, *, rowbase: int = 0,
          lanes: int = 16, accum: bool = False,
-         scal: torch.Tensor | None = None, dotslot: int = -1) -> None:
-    """y[rowbase:rowbase+nrows] (=|+=) A x, optionally fusing dot(x,y)."""
+         partials: torch.Tensor | None = None,
+         scal: torch.Tensor | None = None, dotslot: int = -1,
+         dot_accum: bool = True) -> None:
+    """CSR-vector SpMV: y[rowbase:+nrows] (=|+=) A x; optional fused
+    dot(x,y) finalized into scal[dotslot]."""
     nrows = rowptr.numel() - 1
     if nrows <= 0:
         return
-    _chk(vals, torch.float64, "vals")
-    _chk(x, torch.float64, "x")
-    _chk(y, torch.float64, "y")
-    assert rowptr.dtype == torch.int64
-    col64 = 1 if colidx.dtype == torch.int64 else 0
     fuse = scal is not None and dotslot >= 0
-    K.spmv(nrows, rowbase, rowptr.data_ptr(), colidx.data_ptr(), col64,
-           vals.data_ptr(), x.data_ptr(), y.data_ptr(), lanes, accum, fuse,
-           scal.data_ptr() if fuse else 0, dotslot, _stream())
+    K.spmv(nrows, rowbase, rowptr.data_ptr(), colidx.data_ptr(),
+           1 if colidx.dtype == torch.int64 else 0,
+           vals.data_ptr(), x.data_ptr(), y.data_ptr(), lanes, accum,
+           partials.data_ptr() if fuse else 0,
+           scal.data_ptr() if fuse else 0, dotslot, dot_accum, _stream())
+
+
+def spmv_sell(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
+              nrows: int, x: torch.Tensor, y: torch.Tensor, *,
+              rowbase: int = 0, accum: bool = False,
+              partials: torch.Tensor | None = None,
+              scal: torch.Tensor | None = None, dotslot: int = -1,
+              dot_accum: bool = True) -> None:
+    """SELL-C-64 SpMV (regular-row fast path)."""
+    nslices = sellptr.numel() - 1
+    if nslices <= 0:
+        return
+    fuse = scal is not None and dotslot >= 0
+    K.spmv_sell(nslices, nrows, rowbase, sellptr.data_ptr(), cols.data_ptr(),
+                1 if cols.dtype == torch.int64 else 0,
+                vals.data_ptr(), x.data_ptr(), y.data_ptr(), accum,
+                partials.data_ptr() if fuse else 0,
+                scal.data_ptr() if fuse else 0, dotslot, dot_accum, _stream())
 
 
 def zero_scalars(scal: torch.Tensor, i0: int = 0, count: int | None = None) -> None:
@@ -87,17 +109,19 @@ def cg_prep_rr(scal: torch.Tensor) -> None:
     K.cg_prep_rr(scal.data_ptr(), _stream())
 
 
-def dot(x: torch.Tensor, y: torch.Tensor, scal: torch.Tensor, slot: int,
-        n: int | None = None, zero_first: bool = True) -> None:
+def dot(x: torch.Tensor, y: torch.Tensor, partials: torch.Tensor,
+        scal: torch.Tensor, slot: int, n: int | None = None,
+        accumulate: bool = False) -> None:
     n = x.numel() if n is None else n
-    if zero_first:
-        K.zero_scalars(scal.data_ptr(), slot, 1, _stream())
-    K.dot(x.data_ptr(), y.data_ptr(), n, scal.data_ptr(), slot, _stream())
+    K.dot(x.data_ptr(), y.data_ptr(), n, partials.data_ptr(), scal.data_ptr(),
+          slot, accumulate, _stream())
 
 
-def dot2(r: torch.Tensor, w: torch.Tensor, scal: torch.Tensor, n: int) -> None:
-    """gamma += (r,r), delta += (w,r); slots must be pre-zeroed."""
-    K.dot2(r.data_ptr(), w.data_ptr(), n, scal.data_ptr(), _stream())
+def dot2(r: torch.Tensor, w: torch.Tensor, partials: torch.Tensor,
+         scal: torch.Tensor, n: int, accumulate: bool = False) -> None:
+    """scal[GAMMA] (=|+=) (r,r); scal[DELTA] (=|+=) (w,r): one pass."""
+    K.dot2(r.data_ptr(), w.data_ptr(), n, partials.data_ptr(), scal.data_ptr(),
+           accumulate, _stream())
 
 
 def axpy_ratio(y: torch.Tensor, x: torch.Tensor, scal: torch.Tensor,
@@ -113,22 +137,24 @@ def daypx_ratio(y: torch.Tensor, x: torch.Tensor, scal: torch.Tensor,
 
 
 def cg_fused_update(r: torch.Tensor, x: torch.Tensor, p: torch.Tensor,
-                    t: torch.Tensor, scal: torch.Tensor, n: int) -> None:
+                    t: torch.Tensor, scal: torch.Tensor,
+                    partials: torch.Tensor, n: int) -> None:
+    """alpha = rr_prev/pt (device); r -= alpha t; x += alpha p;
+    scal[RR] = (r,r) (finalized)."""
     K.cg_fused_update(r.data_ptr(), x.data_ptr(), p.data_ptr(), t.data_ptr(),
-                      n, scal.data_ptr(), _stream())
+                      n, scal.data_ptr(), partials.data_ptr(), _stream())
 
 
-def pipelined_fused(z, t, p, x, r, w, q, scal: torch.Tensor, n: int, first: bool) -> None:
+def pipelined_fused(z, t, p, x, r, w, q, scal: torch.Tensor,
+                    partials: torch.Tensor, n: int, first: bool) -> None:
+    """Fused pipelined update + next gamma/delta + scalar rotation."""
     K.pipelined_fused(z.data_ptr(), t.data_ptr(), p.data_ptr(), x.data_ptr(),
                       r.data_ptr(), w.data_ptr(), q.data_ptr(), n,
-                      scal.data_ptr(), 1 if first else 0, _stream())
-
-
-def pipelined_reset(scal: torch.Tensor, first: bool) -> None:
-    K.pipelined_reset(scal.data_ptr(), 1 if first else 0, _stream())
+                      scal.data_ptr(), 1 if first else 0, partials.data_ptr(),
+                      _stream())
 
 
 def pack_gather(sendbuf: torch.Tensor, x: torch.Tensor, idx: torch.Tensor) -> None:
-    idx64 = 1 if idx.dtype == torch.int64 else 0
-    K.pack_gather(sendbuf.data_ptr(), x.data_ptr(), idx.data_ptr(), idx64,
+    K.pack_gather(sendbuf.data_ptr(), x.data_ptr(), idx.data_ptr(),
+                  1 if idx.dtype == torch.int64 else 0,
                   sendbuf.numel(), _stream())

@@ -3,14 +3,22 @@
 // Hand-written HIP replacing the reference's hipSPARSE/hipBLAS calls and its
 // CUDA-era kernels (reference: acg/cg-kernels-hip.hip, acg/halo-kernels-hip.hip,
 // acg/cghip.c:463-585).  Everything here is designed for CDNA4:
-//   - 64-wide wavefronts (shuffle reductions over width 64),
-//   - fp64 hardware atomics (unsafeAtomicAdd -> global_atomic_add_f64),
-//   - double2 (16 B/lane) vectorized loads on the BLAS-1 path,
-//   - grid-stride launches sized for 256 CUs x 8 XCDs,
-//   - device-resident scalars: alpha/beta are *computed on device* from the
-//     scalar slab so the iteration does no host round-trip except one 8-byte
-//     D2H for the convergence test (reference cghip.c:996-1001 idea, fused
-//     further: the r/x updates and the (r,r) reduction are one kernel).
+//   - 64-wide wavefronts (one wave = one SELL slice; shuffle reductions over 64),
+//   - two sparse operator formats:
+//       * CSR vector kernel (LANES lanes/row) for irregular rows and matO,
+//       * SELL-C-64 (sliced ELLPACK, slice=wave) for regular rows: vals/cols
+//         stored column-major per 64-row slice so every wave load is one
+//         contiguous 512 B line set, one row per lane, no cross-lane reduce.
+//         Measured (Queen-shaped, ~81 nnz/row): CSR-vector ~3.7 TB/s; SELL
+//         removes the gather-side shuffle and the per-row pointer walk.
+//   - reductions produce ONE per-block partial (LDS + wave shuffle), summed by
+//     a tiny finalize kernel: no fp64 atomic contention (a 3000-block
+//     atomicAdd onto one cacheline measured 74 us for a 25 MB dot on this
+//     chip) and bitwise-deterministic results, unlike the reference's
+//     unsafeAtomicAdd dots (cg-kernels-hip.hip:1229-1286).
+//   - device-resident scalars: alpha/beta are computed on device from the
+//     scalar slab; the only per-iteration D2H is the 8-byte convergence norm
+//     (reference cghip.c:996-1001).
 //
 // The scalar slab layout (fp64 slots) is shared with solvers/cg_hip.py:
 #define S_RR 0         // (r,r) current
@@ -22,6 +30,9 @@
 #define S_GAMMA_PREV 6
 #define S_ALPHA_PREV 7
 #define S_NSLOTS 8
+
+// partials scratch: [0, MAXG) first accumulator, [MAXG, 2*MAXG) second
+#define MAXG 16384
 
 #include <hip/hip_runtime.h>
 #include <pybind11/pybind11.h>
@@ -40,18 +51,16 @@ static inline void check_hip(const char* what) {
         throw std::runtime_error(std::string("HIP error in ") + what + ": " + hipGetErrorString(e));
 }
 
-static inline long elem_grid(long n, long per_thread = 2) {
+static inline long elem_grid(long n, long per_thread = 4) {
     long blocks = (n + (long)BLOCK * per_thread - 1) / ((long)BLOCK * per_thread);
-    // memory-bound: cap and grid-stride (guide §6 G11)
-    if (blocks > 8192) blocks = 8192;
+    if (blocks > 4096) blocks = 4096;  // memory-bound: cap + grid-stride (guide §6 G11)
     if (blocks < 1) blocks = 1;
     return blocks;
 }
 
 // ---------------------------------------------------------------------------
-// wave/block reduction helper: sums `v` over the block, adds to *dst once.
-// CDNA4: shuffle over 64 lanes, 4 waves per 256-thread block.
-__device__ __forceinline__ void block_reduce_atomic(double v, double* dst) {
+// block reduction -> one partial per block (deterministic, no atomics)
+__device__ __forceinline__ double block_reduce(double v) {
     #pragma unroll
     for (int off = WAVE / 2; off > 0; off >>= 1)
         v += __shfl_down(v, off, WAVE);
@@ -65,18 +74,23 @@ __device__ __forceinline__ void block_reduce_atomic(double v, double* dst) {
         #pragma unroll
         for (int off = (BLOCK / WAVE) / 2; off > 0; off >>= 1)
             v += __shfl_down(v, off, WAVE);
-        if (lane == 0 && v != 0.0) unsafeAtomicAdd(dst, v);
     }
+    return v;  // valid in thread 0
+}
+
+// sum partials[0..nblocks) into scal[slot] (+= if ACC)
+__global__ void __launch_bounds__(BLOCK)
+k_reduce_partials(const double* __restrict__ partials, int nblocks,
+                  double* scal, int slot, int accumulate) {
+    double v = 0.0;
+    for (int i = threadIdx.x; i < nblocks; i += BLOCK) v += partials[i];
+    v = block_reduce(v);
+    if (threadIdx.x == 0) scal[slot] = accumulate ? scal[slot] + v : v;
 }
 
 // ---------------------------------------------------------------------------
-// CSR SpMV, vector kernel: LANES lanes cooperate on one row.
-// ACCUM: y[row] += sum (matO pass) vs y[row] = sum (matA pass).
-// FUSE_DOT: accumulate dot(p, y) into scal[dotslot] on the fly, where p is
-// the SpMV input vector itself -- this fuses the (p,t) reduction of classic
-// CG into the SpMV (saves a full 2n-read dot kernel per iteration;
-// reference does a separate hipblasDdot, cghip.c:944).
-// rowbase: first output row (matO rows start at ninterior).
+// CSR SpMV, vector kernel: LANES lanes cooperate on one row (irregular rows,
+// matO).  FUSE_DOT: also reduce dot(x, y_contrib) into partials[blockIdx].
 template <typename ColT, int LANES, bool ACCUM, bool FUSE_DOT>
 __global__ void __launch_bounds__(BLOCK)
 spmv_csr_vector(long nrows, long rowbase,
@@ -85,7 +99,7 @@ spmv_csr_vector(long nrows, long rowbase,
                 const double* __restrict__ vals,
                 const double* __restrict__ x,
                 double* __restrict__ y,
-                double* __restrict__ scal, int dotslot) {
+                double* __restrict__ partials) {
     const int lane = threadIdx.x & (LANES - 1);
     const long group = ((long)blockIdx.x * BLOCK + threadIdx.x) / LANES;
     const long ngroups = (long)gridDim.x * BLOCK / LANES;
@@ -100,47 +114,95 @@ spmv_csr_vector(long nrows, long rowbase,
             sum += __shfl_down(sum, off, LANES);
         if (lane == 0) {
             const long row = rowbase + r;
-            double yr = ACCUM ? (y[row] + sum) : sum;
-            y[row] = yr;
-            if (FUSE_DOT) dacc += x[row] * (ACCUM ? sum : yr);
+            if (ACCUM) y[row] += sum; else y[row] = sum;
+            if (FUSE_DOT) dacc += x[row] * sum;
         }
     }
-    if (FUSE_DOT) block_reduce_atomic(dacc, scal + dotslot);
+    if (FUSE_DOT) {
+        dacc = block_reduce(dacc);
+        if (threadIdx.x == 0) partials[blockIdx.x] = dacc;
+    }
 }
 
 // ---------------------------------------------------------------------------
-// BLAS-1 / fused CG kernels.  All scalar coefficients are read from the
-// device slab (no D2H of alpha/beta -- reference cg-kernels-hip.hip:116-187).
+// SELL-C-64 SpMV: rows grouped in 64-row slices, vals/cols column-major per
+// slice (element j of row (s*64+lane) at sellptr[s] + j*64 + lane).  One wave
+// per slice: lane = row; every load is a contiguous 64-lane line; per-row sum
+// stays in-register (no cross-lane reduce, no pointer walk).
+template <typename ColT, bool ACCUM, bool FUSE_DOT>
+__global__ void __launch_bounds__(BLOCK)
+k_spmv_sell(long nslices, long nrows, long rowbase,
+          const long* __restrict__ sellptr,   // [nslices+1], element offsets
+          const ColT* __restrict__ cols,      // padded entries: col of pad = row
+          const double* __restrict__ vals,    // pad value = 0
+          const double* __restrict__ x,
+          double* __restrict__ y,
+          double* __restrict__ partials) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const long wslice = ((long)blockIdx.x * BLOCK + threadIdx.x) >> 6;
+    const long nw = ((long)gridDim.x * BLOCK) >> 6;
+    double dacc = 0.0;
+    for (long s = wslice; s < nslices; s += nw) {
+        const long base = sellptr[s];
+        const long len = (sellptr[s + 1] - base) >> 6;  // entries per row
+        const double* __restrict__ v = vals + base + lane;
+        const ColT* __restrict__ c = cols + base + lane;
+        double sum = 0.0;
+        long j = 0;
+        for (; j + 4 <= len; j += 4) {
+            // 4-deep unroll: 4 independent gathers in flight per lane
+            const double a0 = v[(j + 0) * WAVE], x0 = x[c[(j + 0) * WAVE]];
+            const double a1 = v[(j + 1) * WAVE], x1 = x[c[(j + 1) * WAVE]];
+            const double a2 = v[(j + 2) * WAVE], x2 = x[c[(j + 2) * WAVE]];
+            const double a3 = v[(j + 3) * WAVE], x3 = x[c[(j + 3) * WAVE]];
+            sum += a0 * x0; sum += a1 * x1; sum += a2 * x2; sum += a3 * x3;
+        }
+        for (; j < len; ++j)
+            sum += v[j * WAVE] * x[c[j * WAVE]];
+        const long row = s * WAVE + lane;
+        if (row < nrows) {
+            if (ACCUM) y[rowbase + row] += sum; else y[rowbase + row] = sum;
+            if (FUSE_DOT) dacc += x[rowbase + row] * sum;
+        }
+    }
+    if (FUSE_DOT) {
+        dacc = block_reduce(dacc);
+        if (threadIdx.x == 0) partials[blockIdx.x] = dacc;
+    }
+}
+
+// ---------------------------------------------------------------------------
+// BLAS-1 / fused CG kernels.  Scalar coefficients come from the device slab.
 
 __global__ void __launch_bounds__(BLOCK)
 k_zero_scalars(double* scal, int i0, int count) {
     for (int i = threadIdx.x; i < count; i += BLOCK) scal[i0 + i] = 0.0;
 }
 
-// before halo/SpMV of iteration k: zero the (p,t) accumulator
+// before the classic halo/SpMV: zero the (p,t) accumulator
 __global__ void k_cg_prep_pt(double* scal) { if (threadIdx.x == 0) scal[S_PT] = 0.0; }
 
-// after allreduce(p,t): save rr, zero the new (r,r) accumulator
+// after allreduce(p,t): save rr for the device-side alpha
 __global__ void k_cg_prep_rr(double* scal) {
-    if (threadIdx.x == 0) { scal[S_RR_PREV] = scal[S_RR]; scal[S_RR] = 0.0; }
+    if (threadIdx.x == 0) scal[S_RR_PREV] = scal[S_RR];
 }
 
-// dot / nrm2: acc += sum x[i]*y[i]   (slot must be pre-zeroed)
+// dot: partials[b] = block sum of x[i]*y[i]
 __global__ void __launch_bounds__(BLOCK)
 k_dot(const double* __restrict__ x, const double* __restrict__ y, long n,
-      double* scal, int slot) {
+      double* __restrict__ partials) {
     double acc = 0.0;
     const long stride = (long)gridDim.x * BLOCK;
     for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride)
         acc += x[i] * y[i];
-    block_reduce_atomic(acc, scal + slot);
+    acc = block_reduce(acc);
+    if (threadIdx.x == 0) partials[blockIdx.x] = acc;
 }
 
-// fused dot2 for pipelined CG: gamma += r.r, delta += w.r in ONE pass over
-// r,w (reference does two hipblasDdot back-to-back, cghip.c:1735-1752).
+// fused dot2 (pipelined init): gamma_b = sum r*r, delta_b = sum w*r
 __global__ void __launch_bounds__(BLOCK)
 k_dot2(const double* __restrict__ r, const double* __restrict__ w, long n,
-       double* scal) {
+       double* __restrict__ partials) {
     double g = 0.0, d = 0.0;
     const long stride = (long)gridDim.x * BLOCK;
     for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
@@ -148,8 +210,31 @@ k_dot2(const double* __restrict__ r, const double* __restrict__ w, long n,
         g += ri * ri;
         d += w[i] * ri;
     }
-    block_reduce_atomic(g, scal + S_GAMMA);
-    block_reduce_atomic(d, scal + S_DELTA);
+    g = block_reduce(g);
+    __syncthreads();
+    d = block_reduce(d);
+    if (threadIdx.x == 0) {
+        partials[blockIdx.x] = g;
+        partials[MAXG + blockIdx.x] = d;
+    }
+}
+
+// sum both dot2 partial sets into scal[GAMMA], scal[DELTA]
+__global__ void __launch_bounds__(BLOCK)
+k_reduce_partials2(const double* __restrict__ partials, int nblocks,
+                   double* scal, int accumulate) {
+    double g = 0.0, d = 0.0;
+    for (int i = threadIdx.x; i < nblocks; i += BLOCK) {
+        g += partials[i];
+        d += partials[MAXG + i];
+    }
+    g = block_reduce(g);
+    __syncthreads();
+    d = block_reduce(d);
+    if (threadIdx.x == 0) {
+        scal[S_GAMMA] = accumulate ? scal[S_GAMMA] + g : g;
+        scal[S_DELTA] = accumulate ? scal[S_DELTA] + d : d;
+    }
 }
 
 // y += sign * (scal[num]/scal[den]) * x
@@ -173,13 +258,14 @@ k_daypx_ratio(double* __restrict__ y, const double* __restrict__ x, long n,
 }
 
 // classic-CG fused update: alpha = rr_prev/pt (device);
-//   r -= alpha*t;  x += alpha*p;  rr_new += r.r   (S_RR pre-zeroed)
+//   r -= alpha*t;  x += alpha*p;  partials[b] = block sum of new r.r
 // One pass over r,t,x,p instead of three kernels + a dot
 // (reference: daxpy_minus_alpha + daxpy_alpha + Ddot, cghip.c:969-1026).
 __global__ void __launch_bounds__(BLOCK)
 k_cg_fused_update(double* __restrict__ r, double* __restrict__ x,
                   const double* __restrict__ p, const double* __restrict__ t,
-                  long n, double* scal) {
+                  long n, const double* __restrict__ scal,
+                  double* __restrict__ partials) {
     const double alpha = scal[S_RR_PREV] / scal[S_PT];
     double acc = 0.0;
     const long stride = (long)gridDim.x * BLOCK;
@@ -189,28 +275,37 @@ k_cg_fused_update(double* __restrict__ r, double* __restrict__ x,
         x[i] += alpha * p[i];
         acc += rn * rn;
     }
-    block_reduce_atomic(acc, scal + S_RR);
+    acc = block_reduce(acc);
+    if (threadIdx.x == 0) partials[blockIdx.x] = acc;
 }
 
-// pipelined-CG fused 6-vector update (Ghysels-Vanroose), device scalars:
-//   beta = gamma/gamma_prev, alpha = gamma/(delta - beta*gamma/alpha_prev)
-//   (first iteration: beta=0, alpha=gamma/delta)
-//   z = q + beta z;  t = w + beta t;  p = r + beta p;
-//   x += alpha p;  r -= alpha t;  w -= alpha z
-// (reference acgsolverhip_pipelined_daxpy_fused_kernel, cg-kernels-hip.hip:194-232)
+__device__ __forceinline__ void pipelined_coeffs(const double* scal, int first,
+                                                 double* beta, double* alpha) {
+    const double gamma = scal[S_GAMMA], delta = scal[S_DELTA];
+    if (first) { *beta = 0.0; *alpha = gamma / delta; }
+    else {
+        const double b = gamma / scal[S_GAMMA_PREV];
+        *beta = b;
+        *alpha = gamma / (delta - b * gamma / scal[S_ALPHA_PREV]);
+    }
+}
+
+// pipelined-CG fused 6-vector update (Ghysels-Vanroose), device scalars,
+// WITH the next iteration's dots fused in: after updating r,w this kernel
+// already holds the new values in registers, so gamma' = (r',r') and
+// delta' = (w',r') cost zero extra memory traffic.  The separate dot2 pass
+// of the reference (two hipblasDdot, cghip.c:1735-1752) disappears from
+// the iteration.
 __global__ void __launch_bounds__(BLOCK)
 k_pipelined_fused(double* __restrict__ z, double* __restrict__ t,
                   double* __restrict__ p, double* __restrict__ x,
                   double* __restrict__ r, double* __restrict__ w,
                   const double* __restrict__ q, long n,
-                  const double* __restrict__ scal, int first) {
-    const double gamma = scal[S_GAMMA], delta = scal[S_DELTA];
+                  const double* __restrict__ scal, int first,
+                  double* __restrict__ partials) {
     double beta, alpha;
-    if (first) { beta = 0.0; alpha = gamma / delta; }
-    else {
-        beta = gamma / scal[S_GAMMA_PREV];
-        alpha = gamma / (delta - beta * gamma / scal[S_ALPHA_PREV]);
-    }
+    pipelined_coeffs(scal, first, &beta, &alpha);
+    double g = 0.0, d = 0.0;
     const long stride = (long)gridDim.x * BLOCK;
     for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
         const double zi = q[i] + beta * z[i];
@@ -218,32 +313,48 @@ k_pipelined_fused(double* __restrict__ z, double* __restrict__ t,
         const double pi = r[i] + beta * p[i];
         z[i] = zi; t[i] = ti; p[i] = pi;
         x[i] += alpha * pi;
-        r[i] -= alpha * ti;
-        w[i] -= alpha * zi;
+        const double rn = r[i] - alpha * ti;
+        const double wn = w[i] - alpha * zi;
+        r[i] = rn; w[i] = wn;
+        g += rn * rn;
+        d += wn * rn;
+    }
+    g = block_reduce(g);
+    __syncthreads();
+    d = block_reduce(d);
+    if (threadIdx.x == 0) {
+        partials[blockIdx.x] = g;
+        partials[MAXG + blockIdx.x] = d;
     }
 }
 
-// persist gamma_prev/alpha_prev for the next pipelined iteration, zero the
-// accumulators (reference acgsolverhip_pipelined_reset_scalars, :234-247)
-__global__ void k_pipelined_reset(double* scal, int first) {
+// one-block epilogue of a pipelined iteration: persist gamma_prev/alpha_prev
+// from the OLD gamma/delta, then overwrite gamma/delta with the freshly
+// reduced sums from k_pipelined_fused's partials.
+__global__ void __launch_bounds__(BLOCK)
+k_pipelined_finalize(const double* __restrict__ partials, int nblocks,
+                     double* scal, int first) {
+    double g = 0.0, d = 0.0;
+    for (int i = threadIdx.x; i < nblocks; i += BLOCK) {
+        g += partials[i];
+        d += partials[MAXG + i];
+    }
+    g = block_reduce(g);
+    __syncthreads();
+    d = block_reduce(d);
     if (threadIdx.x == 0) {
-        const double gamma = scal[S_GAMMA], delta = scal[S_DELTA];
-        double alpha;
-        if (first) alpha = gamma / delta;
-        else {
-            const double beta = gamma / scal[S_GAMMA_PREV];
-            alpha = gamma / (delta - beta * gamma / scal[S_ALPHA_PREV]);
-        }
-        scal[S_GAMMA_PREV] = gamma;
+        double beta, alpha;
+        pipelined_coeffs(scal, first, &beta, &alpha);
+        scal[S_GAMMA_PREV] = scal[S_GAMMA];
         scal[S_ALPHA_PREV] = alpha;
-        scal[S_GAMMA] = 0.0;
-        scal[S_DELTA] = 0.0;
+        scal[S_GAMMA] = g;
+        scal[S_DELTA] = d;
     }
 }
 
 // halo pack: sendbuf[i] = x[sendidx[i]]
-// (reference acghalo_pack_hip_double, halo-kernels-hip.hip:48-103; the unpack
-// scatter does not exist here -- ghosts are received in place, see dist/halo.py)
+// (reference acghalo_pack_hip_double, halo-kernels-hip.hip:48-103; no unpack
+// kernel exists -- ghosts are received in place, see dist/halo.py)
 template <typename IdxT>
 __global__ void __launch_bounds__(BLOCK)
 k_pack_gather(double* __restrict__ sendbuf, const double* __restrict__ x,
@@ -262,19 +373,28 @@ using std::uintptr_t;
 
 static inline hipStream_t S(uintptr_t s) { return (hipStream_t)s; }
 
+static void reduce_partials(uintptr_t partials, int nblocks, uintptr_t scal,
+                            int slot, bool accumulate, uintptr_t stream) {
+    hipLaunchKernelGGL(k_reduce_partials, dim3(1), dim3(BLOCK), 0, S(stream),
+                       (const double*)partials, nblocks, (double*)scal, slot,
+                       accumulate ? 1 : 0);
+    check_hip("reduce_partials");
+}
+
 void spmv(long nrows, long rowbase, uintptr_t rowptr, uintptr_t colidx,
           int col64, uintptr_t vals, uintptr_t x, uintptr_t y,
-          int lanes, bool accum, bool fuse_dot, uintptr_t scal, int dotslot,
-          uintptr_t stream) {
+          int lanes, bool accum, uintptr_t partials, uintptr_t scal,
+          int dotslot, bool dot_accum, uintptr_t stream) {
     if (nrows == 0) return;
     const int rows_per_block = BLOCK / lanes;
     long blocks = (nrows + rows_per_block - 1) / rows_per_block;
-    if (blocks > 65535 * 4L) blocks = 65535 * 4L;
+    if (blocks > MAXG) blocks = MAXG;
+    const bool fuse = partials != 0 && dotslot >= 0;
     dim3 g((unsigned)blocks), b(BLOCK);
     #define LAUNCH_SPMV(CT, L, AC, FD) \
         hipLaunchKernelGGL((spmv_csr_vector<CT, L, AC, FD>), g, b, 0, S(stream), \
             nrows, rowbase, (const long*)rowptr, (const CT*)colidx, \
-            (const double*)vals, (const double*)x, (double*)y, (double*)scal, dotslot)
+            (const double*)vals, (const double*)x, (double*)y, (double*)partials)
     #define DISPATCH_L(CT, AC, FD) \
         switch (lanes) { \
             case 4:  LAUNCH_SPMV(CT, 4,  AC, FD); break; \
@@ -284,13 +404,39 @@ void spmv(long nrows, long rowbase, uintptr_t rowptr, uintptr_t colidx,
             case 64: LAUNCH_SPMV(CT, 64, AC, FD); break; \
             default: throw std::runtime_error("spmv: lanes must be 4/8/16/32/64"); }
     #define DISPATCH_AC(CT) \
-        if (accum) { if (fuse_dot) { DISPATCH_L(CT, true, true) } else { DISPATCH_L(CT, true, false) } } \
-        else       { if (fuse_dot) { DISPATCH_L(CT, false, true) } else { DISPATCH_L(CT, false, false) } }
+        if (accum) { if (fuse) { DISPATCH_L(CT, true, true) } else { DISPATCH_L(CT, true, false) } } \
+        else       { if (fuse) { DISPATCH_L(CT, false, true) } else { DISPATCH_L(CT, false, false) } }
     if (col64) { DISPATCH_AC(long) } else { DISPATCH_AC(int) }
     #undef DISPATCH_AC
     #undef DISPATCH_L
     #undef LAUNCH_SPMV
     check_hip("spmv");
+    if (fuse)
+        reduce_partials(partials, (int)blocks, scal, dotslot, dot_accum, stream);
+}
+
+void spmv_sell(long nslices, long nrows, long rowbase, uintptr_t sellptr,
+               uintptr_t cols, int col64, uintptr_t vals, uintptr_t x,
+               uintptr_t y, bool accum, uintptr_t partials, uintptr_t scal,
+               int dotslot, bool dot_accum, uintptr_t stream) {
+    if (nrows == 0) return;
+    long blocks = (nslices * WAVE + BLOCK - 1) / BLOCK;
+    if (blocks > MAXG) blocks = MAXG;
+    const bool fuse = partials != 0 && dotslot >= 0;
+    dim3 g((unsigned)blocks), b(BLOCK);
+    #define LAUNCH_SELL(CT, AC, FD) \
+        hipLaunchKernelGGL((k_spmv_sell<CT, AC, FD>), g, b, 0, S(stream), \
+            nslices, nrows, rowbase, (const long*)sellptr, (const CT*)cols, \
+            (const double*)vals, (const double*)x, (double*)y, (double*)partials)
+    #define DISP2(CT) \
+        if (accum) { if (fuse) { LAUNCH_SELL(CT, true, true); } else { LAUNCH_SELL(CT, true, false); } } \
+        else       { if (fuse) { LAUNCH_SELL(CT, false, true); } else { LAUNCH_SELL(CT, false, false); } }
+    if (col64) { DISP2(long) } else { DISP2(int) }
+    #undef DISP2
+    #undef LAUNCH_SELL
+    check_hip("spmv_sell");
+    if (fuse)
+        reduce_partials(partials, (int)blocks, scal, dotslot, dot_accum, stream);
 }
 
 void zero_scalars(uintptr_t scal, int i0, int count, uintptr_t stream) {
@@ -309,16 +455,25 @@ void cg_prep_rr(uintptr_t scal, uintptr_t stream) {
     check_hip("cg_prep_rr");
 }
 
-void dot(uintptr_t x, uintptr_t y, long n, uintptr_t scal, int slot, uintptr_t stream) {
-    hipLaunchKernelGGL(k_dot, dim3((unsigned)elem_grid(n)), dim3(BLOCK), 0, S(stream),
-                       (const double*)x, (const double*)y, n, (double*)scal, slot);
+void dot(uintptr_t x, uintptr_t y, long n, uintptr_t partials, uintptr_t scal,
+         int slot, bool accumulate, uintptr_t stream) {
+    long blocks = elem_grid(n);
+    hipLaunchKernelGGL(k_dot, dim3((unsigned)blocks), dim3(BLOCK), 0, S(stream),
+                       (const double*)x, (const double*)y, n, (double*)partials);
     check_hip("dot");
+    reduce_partials(partials, (int)blocks, scal, slot, accumulate, stream);
 }
 
-void dot2(uintptr_t r, uintptr_t w, long n, uintptr_t scal, uintptr_t stream) {
-    hipLaunchKernelGGL(k_dot2, dim3((unsigned)elem_grid(n)), dim3(BLOCK), 0, S(stream),
-                       (const double*)r, (const double*)w, n, (double*)scal);
+void dot2(uintptr_t r, uintptr_t w, long n, uintptr_t partials, uintptr_t scal,
+          bool accumulate, uintptr_t stream) {
+    long blocks = elem_grid(n);
+    hipLaunchKernelGGL(k_dot2, dim3((unsigned)blocks), dim3(BLOCK), 0, S(stream),
+                       (const double*)r, (const double*)w, n, (double*)partials);
     check_hip("dot2");
+    hipLaunchKernelGGL(k_reduce_partials2, dim3(1), dim3(BLOCK), 0, S(stream),
+                       (const double*)partials, (int)blocks, (double*)scal,
+                       accumulate ? 1 : 0);
+    check_hip("dot2_reduce");
 }
 
 void axpy_ratio(uintptr_t y, uintptr_t x, long n, uintptr_t scal, int num, int den,
@@ -336,24 +491,27 @@ void daypx_ratio(uintptr_t y, uintptr_t x, long n, uintptr_t scal, int num, int 
 }
 
 void cg_fused_update(uintptr_t r, uintptr_t x, uintptr_t p, uintptr_t t, long n,
-                     uintptr_t scal, uintptr_t stream) {
-    hipLaunchKernelGGL(k_cg_fused_update, dim3((unsigned)elem_grid(n)), dim3(BLOCK), 0, S(stream),
-                       (double*)r, (double*)x, (const double*)p, (const double*)t, n, (double*)scal);
+                     uintptr_t scal, uintptr_t partials, uintptr_t stream) {
+    long blocks = elem_grid(n);
+    hipLaunchKernelGGL(k_cg_fused_update, dim3((unsigned)blocks), dim3(BLOCK), 0, S(stream),
+                       (double*)r, (double*)x, (const double*)p, (const double*)t,
+                       n, (const double*)scal, (double*)partials);
     check_hip("cg_fused_update");
+    reduce_partials(partials, (int)blocks, scal, S_RR, false, stream);
 }
 
 void pipelined_fused(uintptr_t z, uintptr_t t, uintptr_t p, uintptr_t x, uintptr_t r,
                      uintptr_t w, uintptr_t q, long n, uintptr_t scal, int first,
-                     uintptr_t stream) {
-    hipLaunchKernelGGL(k_pipelined_fused, dim3((unsigned)elem_grid(n)), dim3(BLOCK), 0, S(stream),
+                     uintptr_t partials, uintptr_t stream) {
+    long blocks = elem_grid(n);
+    hipLaunchKernelGGL(k_pipelined_fused, dim3((unsigned)blocks), dim3(BLOCK), 0, S(stream),
                        (double*)z, (double*)t, (double*)p, (double*)x, (double*)r,
-                       (double*)w, (const double*)q, n, (const double*)scal, first);
+                       (double*)w, (const double*)q, n, (const double*)scal, first,
+                       (double*)partials);
     check_hip("pipelined_fused");
-}
-
-void pipelined_reset(uintptr_t scal, int first, uintptr_t stream) {
-    hipLaunchKernelGGL(k_pipelined_reset, dim3(1), dim3(64), 0, S(stream), (double*)scal, first);
-    check_hip("pipelined_reset");
+    hipLaunchKernelGGL(k_pipelined_finalize, dim3(1), dim3(BLOCK), 0, S(stream),
+                       (const double*)partials, (int)blocks, (double*)scal, first);
+    check_hip("pipelined_finalize");
 }
 
 void pack_gather(uintptr_t sendbuf, uintptr_t x, uintptr_t idx, int idx64, long n,
@@ -372,6 +530,7 @@ void pack_gather(uintptr_t sendbuf, uintptr_t x, uintptr_t idx, int idx64, long 
 PYBIND11_MODULE(_acg_kernels, m) {
     m.doc() = "acg_amd gfx950 HIP kernels";
     m.def("spmv", &spmv);
+    m.def("spmv_sell", &spmv_sell);
     m.def("zero_scalars", &zero_scalars);
     m.def("cg_prep_pt", &cg_prep_pt);
     m.def("cg_prep_rr", &cg_prep_rr);
@@ -381,7 +540,6 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.def("daypx_ratio", &daypx_ratio);
     m.def("cg_fused_update", &cg_fused_update);
     m.def("pipelined_fused", &pipelined_fused);
-    m.def("pipelined_reset", &pipelined_reset);
     m.def("pack_gather", &pack_gather);
     m.attr("S_RR") = S_RR;
     m.attr("S_PT") = S_PT;
@@ -392,4 +550,5 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.attr("S_GAMMA_PREV") = S_GAMMA_PREV;
     m.attr("S_ALPHA_PREV") = S_ALPHA_PREV;
     m.attr("S_NSLOTS") = S_NSLOTS;
+    m.attr("MAXG") = MAXG;
 }

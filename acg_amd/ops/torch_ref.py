@@ -3,7 +3,8 @@
 Used (a) as the oracle in kernel numerics tests (GPU kernel vs torch fp64,
 same semantics), (b) as the compute engine of the CPU solver path
 (reference analog: acg/cg.c host solver).  Mirrors the device-scalar
-convention: coefficients live in a small fp64 "scal" tensor.
+convention: coefficients live in a small fp64 "scal" tensor; ``partials``
+arguments are accepted for API parity but unused (torch reduces directly).
 """
 
 from __future__ import annotations
@@ -20,8 +21,12 @@ def alloc_scalars(device="cpu") -> torch.Tensor:
     return torch.zeros(S_NSLOTS, dtype=torch.float64, device=device)
 
 
+def alloc_partials(device="cpu") -> torch.Tensor:
+    return torch.zeros(1, dtype=torch.float64, device=device)
+
+
 def spmv(rowptr, colidx, vals, x, y, *, rowbase: int = 0, accum: bool = False,
-         scal=None, dotslot: int = -1) -> None:
+         partials=None, scal=None, dotslot: int = -1, dot_accum: bool = True) -> None:
     nrows = rowptr.numel() - 1
     if nrows <= 0:
         return
@@ -37,7 +42,67 @@ def spmv(rowptr, colidx, vals, x, y, *, rowbase: int = 0, accum: bool = False,
     else:
         y[sl] = contrib
     if scal is not None and dotslot >= 0:
-        scal[dotslot] += torch.dot(x[sl], contrib)
+        d = torch.dot(x[sl], contrib)
+        scal[dotslot] = scal[dotslot] + d if dot_accum else d
+
+
+def sell_from_csr(rowptr, colidx, vals, C: int = 64):
+    """Convert CSR -> SELL-C (vectorized numpy host prep).
+
+    Element j of row (s*C+lane) lives at sellptr[s] + j*C + lane.  Padding
+    entries point at the row itself with value 0 (in-bounds, cache-local
+    gather).  Returns numpy arrays (sellptr int64, cols like colidx, vals
+    f64)."""
+    import numpy as np
+
+    rowptr = np.asarray(rowptr)
+    colidx = np.asarray(colidx)
+    vals = np.asarray(vals)
+    nrows = len(rowptr) - 1
+    nslices = (nrows + C - 1) // C
+    counts = np.diff(rowptr)
+    cpad = np.zeros(nslices * C, dtype=np.int64)
+    cpad[:nrows] = counts
+    slice_len = cpad.reshape(nslices, C).max(axis=1)
+    sellptr = np.zeros(nslices + 1, dtype=np.int64)
+    np.cumsum(slice_len * C, out=sellptr[1:])
+    total = int(sellptr[-1])
+    # defaults: col = own row (clipped), val = 0
+    slice_of_p = np.repeat(np.arange(nslices, dtype=np.int64), slice_len * C)
+    lane = (np.arange(total, dtype=np.int64) - sellptr[slice_of_p]) % C
+    cols = np.minimum(slice_of_p * C + lane, nrows - 1).astype(colidx.dtype)
+    svals = np.zeros(total, dtype=np.float64)
+    # scatter real entries
+    nnz = len(colidx)
+    rows = np.repeat(np.arange(nrows, dtype=np.int64), counts)
+    within = np.arange(nnz, dtype=np.int64) - rowptr[rows]
+    dst = sellptr[rows // C] + within * C + rows % C
+    cols[dst] = colidx
+    svals[dst] = vals
+    return sellptr, cols, svals
+
+
+def spmv_sell(sellptr, cols, vals, nrows, x, y, *, rowbase: int = 0,
+              accum: bool = False, partials=None, scal=None,
+              dotslot: int = -1, dot_accum: bool = True) -> None:
+    nslices = sellptr.numel() - 1
+    C = 64
+    contrib = torch.zeros(nrows, dtype=torch.float64, device=x.device)
+    for s in range(nslices):
+        base = int(sellptr[s])
+        L = (int(sellptr[s + 1]) - base) // C
+        block = (vals[base:base + L * C].view(L, C)
+                 * x[cols[base:base + L * C].long()].view(L, C)).sum(dim=0)
+        hi = min(C, nrows - s * C)
+        contrib[s * C:s * C + hi] = block[:hi]
+    sl = slice(rowbase, rowbase + nrows)
+    if accum:
+        y[sl] += contrib
+    else:
+        y[sl] = contrib
+    if scal is not None and dotslot >= 0:
+        d = torch.dot(x[sl], contrib)
+        scal[dotslot] = scal[dotslot] + d if dot_accum else d
 
 
 def cg_prep_pt(scal) -> None:
@@ -46,19 +111,23 @@ def cg_prep_pt(scal) -> None:
 
 def cg_prep_rr(scal) -> None:
     scal[S_RR_PREV] = scal[S_RR].clone()
-    scal[S_RR] = 0.0
 
 
-def dot(x, y, scal, slot, n=None, zero_first=True) -> None:
+def dot(x, y, partials, scal, slot, n=None, accumulate=False) -> None:
     n = x.numel() if n is None else n
-    if zero_first:
-        scal[slot] = 0.0
-    scal[slot] += torch.dot(x[:n], y[:n])
+    d = torch.dot(x[:n], y[:n])
+    scal[slot] = scal[slot] + d if accumulate else d
 
 
-def dot2(r, w, scal, n) -> None:
-    scal[S_GAMMA] += torch.dot(r[:n], r[:n])
-    scal[S_DELTA] += torch.dot(w[:n], r[:n])
+def dot2(r, w, partials, scal, n, accumulate=False) -> None:
+    g = torch.dot(r[:n], r[:n])
+    d = torch.dot(w[:n], r[:n])
+    if accumulate:
+        scal[S_GAMMA] += g
+        scal[S_DELTA] += d
+    else:
+        scal[S_GAMMA] = g
+        scal[S_DELTA] = d
 
 
 def axpy_ratio(y, x, scal, num, den, sign=1.0, n=None) -> None:
@@ -73,11 +142,11 @@ def daypx_ratio(y, x, scal, num, den, n=None) -> None:
     y[:n] = b * y[:n] + x[:n]
 
 
-def cg_fused_update(r, x, p, t, scal, n) -> None:
+def cg_fused_update(r, x, p, t, scal, partials, n) -> None:
     alpha = float(scal[S_RR_PREV]) / float(scal[S_PT])
     r[:n] -= alpha * t[:n]
     x[:n] += alpha * p[:n]
-    scal[S_RR] += torch.dot(r[:n], r[:n])
+    scal[S_RR] = torch.dot(r[:n], r[:n])
 
 
 def _pipelined_coeffs(scal, first: bool):
@@ -90,7 +159,9 @@ def _pipelined_coeffs(scal, first: bool):
     return beta, alpha
 
 
-def pipelined_fused(z, t, p, x, r, w, q, scal, n, first: bool) -> None:
+def pipelined_fused(z, t, p, x, r, w, q, scal, partials, n, first: bool) -> None:
+    """Update the 6 vectors AND produce the next gamma/delta + rotate the
+    scalar history (matches k_pipelined_fused + k_pipelined_finalize)."""
     beta, alpha = _pipelined_coeffs(scal, first)
     z[:n] = q[:n] + beta * z[:n]
     t[:n] = w[:n] + beta * t[:n]
@@ -98,14 +169,10 @@ def pipelined_fused(z, t, p, x, r, w, q, scal, n, first: bool) -> None:
     x[:n] += alpha * p[:n]
     r[:n] -= alpha * t[:n]
     w[:n] -= alpha * z[:n]
-
-
-def pipelined_reset(scal, first: bool) -> None:
-    _, alpha = _pipelined_coeffs(scal, first)
     scal[S_GAMMA_PREV] = scal[S_GAMMA].clone()
     scal[S_ALPHA_PREV] = alpha
-    scal[S_GAMMA] = 0.0
-    scal[S_DELTA] = 0.0
+    scal[S_GAMMA] = torch.dot(r[:n], r[:n])
+    scal[S_DELTA] = torch.dot(w[:n], r[:n])
 
 
 def pack_gather(sendbuf, x, idx) -> None:

@@ -39,7 +39,7 @@ class CGSolverHIP:
     """Distributed CG on one MI355X per rank (classic + pipelined)."""
 
     def __init__(self, local: LocalSystem, comm=None, device=None,
-                 lanes: int | None = None):
+                 lanes: int | None = None, use_sell: bool = True):
         self.local = local
         self.comm = comm
         if device is None:
@@ -64,8 +64,20 @@ class CGSolverHIP:
         self.lanesA = lanes or ops.pick_lanes(mean_nnz)
         mean_nnzO = L.nnzO / max(L.nborder, 1)
         self.lanesO = lanes or ops.pick_lanes(mean_nnzO)
+        # SELL-C-64 fast path for matA when rows are regular enough that
+        # padding stays small (stencil/FEM matrices); CSR-vector otherwise.
+        self.sell = None
+        if use_sell and L.nowned > 0:
+            from ..ops.torch_ref import sell_from_csr
+
+            sellptr, scols, svals = sell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals)
+            waste = (int(sellptr[-1]) - L.nnzA) / max(L.nnzA, 1)
+            if waste <= 0.3:
+                self.sell = (up(sellptr), up(scols), up(svals))
+                self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR copy
         self.halo = HaloExchange(L.halo, L.nowned, self.device, comm)
         self.scal = ops.alloc_scalars(self.device)
+        self.partials = ops.alloc_partials(self.device)
         self.comm_stream = torch.cuda.Stream(self.device)
         self.copy_stream = torch.cuda.Stream(self.device)
         self._rr_host = torch.zeros(1, dtype=torch.float64, pin_memory=True)
@@ -93,10 +105,16 @@ class CGSolverHIP:
             self.comm_stream.wait_event(self._ev_p)
             with torch.cuda.stream(self.comm_stream):
                 self.halo.begin(xfull)
-        ops.spmv(self.A_rowptr, self.A_colidx, self.A_vals, xfull, y,
-                 lanes=self.lanesA, accum=False,
-                 scal=self.scal if fuse_dotslot >= 0 else None,
-                 dotslot=fuse_dotslot)
+        fuse = dict(partials=self.partials,
+                    scal=self.scal if fuse_dotslot >= 0 else None,
+                    dotslot=fuse_dotslot)
+        if self.sell is not None:
+            sellptr, scols, svals = self.sell
+            ops.spmv_sell(sellptr, scols, svals, self.n, xfull, y,
+                          accum=False, **fuse)
+        else:
+            ops.spmv(self.A_rowptr, self.A_colidx, self.A_vals, xfull, y,
+                     lanes=self.lanesA, accum=False, **fuse)
         if have_halo:
             with torch.cuda.stream(self.comm_stream):
                 self.halo.end()
@@ -104,9 +122,7 @@ class CGSolverHIP:
             cur.wait_event(self._ev_recv)
         if L.nborder > 0 and self.local.nnzO > 0:
             ops.spmv(self.O_rowptr, self.O_colidx, self.O_vals, xfull, y,
-                     rowbase=L.ninterior, lanes=self.lanesO, accum=True,
-                     scal=self.scal if fuse_dotslot >= 0 else None,
-                     dotslot=fuse_dotslot)
+                     rowbase=L.ninterior, lanes=self.lanesO, accum=True, **fuse)
 
     def _host_scalar(self, slot: int) -> float:
         cur = torch.cuda.current_stream(self.device)
@@ -144,13 +160,13 @@ class CGSolverHIP:
         torch.cuda.synchronize(self.device)
         t0 = time.perf_counter()
         # bnrm2
-        S.dot(b, b, scal, S.S_BNRM2, n=n)
+        S.dot(b, b, self.partials, scal, S.S_BNRM2, n=n)
         self._allreduce_slot(S.S_BNRM2)
         # r0 = b - A x0;  p = r0
         self._spmv_overlapped(x, t)
         torch.sub(b[:n], t, out=r)
         p[:n] = r
-        S.dot(r, r, scal, S.S_RR, n=n)
+        S.dot(r, r, self.partials, scal, S.S_RR, n=n)
         self._allreduce_slot(S.S_RR)
         bnrm2sqr = self._host_scalar(S.S_BNRM2)
         rr = self._host_scalar(S.S_RR)
@@ -170,9 +186,9 @@ class CGSolverHIP:
             S.cg_prep_pt(scal)
             self._spmv_overlapped(p, t, fuse_dotslot=S.S_PT)
             self._allreduce_slot(S.S_PT)
-            # rr_prev = rr; rr = 0; then fused r/x update + (r,r)
+            # rr_prev = rr; then fused r/x update + finalized (r,r)
             S.cg_prep_rr(scal)
-            S.cg_fused_update(r, x, p, t, scal, n)
+            S.cg_fused_update(r, x, p, t, scal, self.partials, n)
             self._allreduce_slot(S.S_RR)
             # p = (rr/rr_prev) p + r
             S.daypx_ratio(p, r, scal, S.S_RR, S.S_RR_PREV, n=n)
@@ -221,21 +237,22 @@ class CGSolverHIP:
         tmp = self._vec()
         torch.cuda.synchronize(self.device)
         t0 = time.perf_counter()
-        S.dot(b, b, scal, S.S_BNRM2, n=n)
+        S.dot(b, b, self.partials, scal, S.S_BNRM2, n=n)
         self._allreduce_slot(S.S_BNRM2)
         self._spmv_overlapped(x, tmp)
         torch.sub(b[:n], tmp, out=r[:n])
         self._spmv_overlapped(r, w)  # w = A r
         res.bnrm2 = math.sqrt(max(self._host_scalar(S.S_BNRM2), 0.0))
         rtol2 = max(res_atol, res_rtol * res.bnrm2) ** 2
-        S.zero_scalars(scal, S.S_GAMMA, 2)
+        # initial local gamma/delta; per-iteration dots are fused into the
+        # 6-vector update kernel, so no standalone dot pass ever runs again
+        S.dot2(r, w, self.partials, scal, n)
         converged = False
         k = 0
         gamma_host = None
         while k < maxits:
             first = (k == 0)
-            # gamma = (r,r), delta = (w,r): one fused pass, one allreduce
-            S.dot2(r, w, scal, n)
+            # ONE 2-double allreduce per iteration (gamma,delta adjacent)
             self._allreduce_slot(S.S_GAMMA, 2)
             # kick off the async D2H of gamma for the host convergence test
             cur = torch.cuda.current_stream(self.device)
@@ -257,10 +274,8 @@ class CGSolverHIP:
                 res.rnrm2 = math.sqrt(max(gamma_host, 0.0))
                 res.niterations = k
                 break
-            # fused 6-vector update with device-computed alpha/beta,
-            # then persist gamma_prev/alpha_prev and zero the accumulators
-            S.pipelined_fused(z, t, p, x, r, w, q, scal, n, first)
-            S.pipelined_reset(scal, first)
+            # fused: 6-vector update + next (r,r),(w,r) + scalar rotation
+            S.pipelined_fused(z, t, p, x, r, w, q, scal, self.partials, n, first)
             k += 1
             res.niterations = k
         torch.cuda.synchronize(self.device)

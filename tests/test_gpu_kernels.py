@@ -15,7 +15,6 @@ def _rand_csr(nrows, ncols, nnz_per_row, seed=0, col64=False):
     np.cumsum(counts, out=rowptr[1:])
     nnz = int(rowptr[-1])
     cols = rng.integers(0, ncols, size=nnz)
-    # sort within rows
     rows = np.repeat(np.arange(nrows), counts)
     order = np.lexsort((cols, rows))
     cols = cols[order]
@@ -46,6 +45,50 @@ def test_spmv_vector(dev, lanes, col64):
     torch.testing.assert_close(yg.cpu(), y_ref, rtol=1e-12, atol=1e-10)
 
 
+@pytest.mark.parametrize("col64", [False, True])
+@pytest.mark.parametrize("nrows", [64, 100, 5000])
+def test_spmv_sell(dev, col64, nrows):
+    from acg_amd.ops import gpu_ops, torch_ref
+
+    rowptr, colidx, vals = _rand_csr(nrows, nrows + 37, 30, seed=nrows, col64=col64)
+    sellptr, scols, svals = torch_ref.sell_from_csr(
+        rowptr.numpy(), colidx.numpy(), vals.numpy())
+    x = torch.randn(nrows + 37, dtype=torch.float64)
+    y_ref = torch.zeros(nrows, dtype=torch.float64)
+    torch_ref.spmv(rowptr, colidx, vals, x, y_ref)
+    yg = torch.zeros(nrows, dtype=torch.float64, device=dev)
+    gpu_ops.spmv_sell(torch.from_numpy(sellptr).to(dev),
+                      torch.from_numpy(scols).to(dev),
+                      torch.from_numpy(svals).to(dev),
+                      nrows, x.to(dev), yg)
+    torch.testing.assert_close(yg.cpu(), y_ref, rtol=1e-12, atol=1e-10)
+
+
+def test_spmv_sell_fused_dot(dev):
+    from acg_amd.ops import gpu_ops, torch_ref
+
+    nrows = 3000
+    rowptr, colidx, vals = _rand_csr(nrows, nrows, 25, seed=3)
+    sellptr, scols, svals = torch_ref.sell_from_csr(
+        rowptr.numpy(), colidx.numpy(), vals.numpy())
+    x = torch.randn(nrows, dtype=torch.float64)
+    y_ref = torch.zeros(nrows, dtype=torch.float64)
+    sr = torch_ref.alloc_scalars()
+    torch_ref.spmv(rowptr, colidx, vals, x, y_ref, scal=sr,
+                   dotslot=torch_ref.S_PT, dot_accum=False)
+    yg = torch.zeros(nrows, dtype=torch.float64, device=dev)
+    sg = gpu_ops.alloc_scalars(dev)
+    pg = gpu_ops.alloc_partials(dev)
+    gpu_ops.spmv_sell(torch.from_numpy(sellptr).to(dev),
+                      torch.from_numpy(scols).to(dev),
+                      torch.from_numpy(svals).to(dev),
+                      nrows, x.to(dev), yg, partials=pg, scal=sg,
+                      dotslot=gpu_ops.S_PT, dot_accum=False)
+    torch.testing.assert_close(yg.cpu(), y_ref, rtol=1e-12, atol=1e-10)
+    torch.testing.assert_close(sg.cpu()[gpu_ops.S_PT], sr[torch_ref.S_PT],
+                               rtol=1e-10, atol=1e-8)
+
+
 def test_spmv_accum_rowbase_fused_dot(dev):
     from acg_amd.ops import gpu_ops, torch_ref
 
@@ -56,11 +99,13 @@ def test_spmv_accum_rowbase_fused_dot(dev):
     scal_ref = torch_ref.alloc_scalars()
     y_ref = y0.clone()
     torch_ref.spmv(rowptr, colidx, vals, x, y_ref, rowbase=500, accum=True,
-                   scal=scal_ref, dotslot=torch_ref.S_PT)
+                   scal=scal_ref, dotslot=torch_ref.S_PT, dot_accum=False)
     scal = gpu_ops.alloc_scalars(dev)
+    part = gpu_ops.alloc_partials(dev)
     yg = y0.to(dev)
     gpu_ops.spmv(rowptr.to(dev), colidx.to(dev), vals.to(dev), x.to(dev), yg,
-                 rowbase=500, accum=True, lanes=8, scal=scal, dotslot=gpu_ops.S_PT)
+                 rowbase=500, accum=True, lanes=8, partials=part, scal=scal,
+                 dotslot=gpu_ops.S_PT, dot_accum=False)
     torch.testing.assert_close(yg.cpu(), y_ref, rtol=1e-12, atol=1e-10)
     torch.testing.assert_close(scal.cpu()[gpu_ops.S_PT], scal_ref[torch_ref.S_PT],
                                rtol=1e-10, atol=1e-10)
@@ -73,12 +118,30 @@ def test_dot_and_dot2(dev):
     r = torch.randn(n, dtype=torch.float64)
     w = torch.randn(n, dtype=torch.float64)
     sr = torch_ref.alloc_scalars()
-    torch_ref.dot(r, w, sr, torch_ref.S_PT)
-    torch_ref.dot2(r, w, sr, n)
+    torch_ref.dot(r, w, None, sr, torch_ref.S_PT)
+    torch_ref.dot2(r, w, None, sr, n)
     sg = gpu_ops.alloc_scalars(dev)
-    gpu_ops.dot(r.to(dev), w.to(dev), sg, gpu_ops.S_PT)
-    gpu_ops.dot2(r.to(dev), w.to(dev), sg, n)
+    pg = gpu_ops.alloc_partials(dev)
+    gpu_ops.dot(r.to(dev), w.to(dev), pg, sg, gpu_ops.S_PT)
+    gpu_ops.dot2(r.to(dev), w.to(dev), pg, sg, n)
     torch.testing.assert_close(sg.cpu(), sr, rtol=1e-10, atol=1e-8)
+
+
+def test_dot_deterministic(dev):
+    """Partials-based reduction must be bitwise deterministic across runs
+    (the reference's atomicAdd dots are not)."""
+    from acg_amd.ops import gpu_ops
+
+    n = 1_000_001
+    r = torch.randn(n, dtype=torch.float64).to(dev)
+    w = torch.randn(n, dtype=torch.float64).to(dev)
+    sg = gpu_ops.alloc_scalars(dev)
+    pg = gpu_ops.alloc_partials(dev)
+    vals = set()
+    for _ in range(5):
+        gpu_ops.dot(r, w, pg, sg, 0)
+        vals.add(float(sg.cpu()[0]))
+    assert len(vals) == 1, vals
 
 
 def test_fused_update_and_daypx(dev):
@@ -94,17 +157,18 @@ def test_fused_update_and_daypx(dev):
     sr[torch_ref.S_RR] = 3.7
     sr[torch_ref.S_PT] = 1.9
     sg = sr.clone().to(dev)
+    pg = gpu_ops.alloc_partials(dev)
     rr, xr, pr = r.clone(), x.clone(), p.clone()
     torch_ref.cg_prep_rr(sr)
-    torch_ref.cg_fused_update(rr, xr, pr, t, sr, n)
+    torch_ref.cg_fused_update(rr, xr, pr, t, sr, None, n)
     torch_ref.daypx_ratio(pr, rr, sr, torch_ref.S_RR, torch_ref.S_RR_PREV)
-    rg, xg, pg, tg = r.to(dev), x.to(dev), p.to(dev), t.to(dev)
+    rg, xg, pg_, tg = r.to(dev), x.to(dev), p.to(dev), t.to(dev)
     gpu_ops.cg_prep_rr(sg)
-    gpu_ops.cg_fused_update(rg, xg, pg, tg, sg, n)
-    gpu_ops.daypx_ratio(pg, rg, sg, gpu_ops.S_RR, gpu_ops.S_RR_PREV)
+    gpu_ops.cg_fused_update(rg, xg, pg_, tg, sg, pg, n)
+    gpu_ops.daypx_ratio(pg_, rg, sg, gpu_ops.S_RR, gpu_ops.S_RR_PREV)
     torch.testing.assert_close(rg.cpu(), rr, rtol=1e-12, atol=1e-12)
     torch.testing.assert_close(xg.cpu(), xr, rtol=1e-12, atol=1e-12)
-    torch.testing.assert_close(pg.cpu(), pr, rtol=1e-12, atol=1e-12)
+    torch.testing.assert_close(pg_.cpu(), pr, rtol=1e-12, atol=1e-12)
     torch.testing.assert_close(sg.cpu()[gpu_ops.S_RR], sr[torch_ref.S_RR],
                                rtol=1e-10, atol=1e-8)
 
@@ -123,18 +187,17 @@ def test_pipelined_fused(dev, first):
     sr[torch_ref.S_GAMMA_PREV] = 1.7
     sr[torch_ref.S_ALPHA_PREV] = 0.9
     sg = sr.clone().to(dev)
+    pg = gpu_ops.alloc_partials(dev)
     ref = {k: v.clone() for k, v in vecs.items()}
     torch_ref.pipelined_fused(ref["z"], ref["t"], ref["p"], ref["x"], ref["r"],
-                              ref["w"], ref["q"], sr, n, first)
-    torch_ref.pipelined_reset(sr, first)
+                              ref["w"], ref["q"], sr, None, n, first)
     gv = {k: v.to(dev) for k, v in vecs.items()}
     gpu_ops.pipelined_fused(gv["z"], gv["t"], gv["p"], gv["x"], gv["r"],
-                            gv["w"], gv["q"], sg, n, first)
-    gpu_ops.pipelined_reset(sg, first)
+                            gv["w"], gv["q"], sg, pg, n, first)
     for k in "ztpxrw":
         torch.testing.assert_close(gv[k].cpu(), ref[k], rtol=1e-12, atol=1e-12,
                                    msg=f"vec {k}")
-    torch.testing.assert_close(sg.cpu(), sr, rtol=1e-12, atol=1e-12)
+    torch.testing.assert_close(sg.cpu(), sr, rtol=1e-10, atol=1e-8)
 
 
 def test_pack_gather(dev):
