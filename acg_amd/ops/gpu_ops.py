@@ -78,22 +78,31 @@ def spmv(rowptr: torch.Tensor, colidx: torch.Tensor, vals: torch.Tensor,
            scal.data_ptr() if fuse else 0, dotslot, dot_accum, _stream())
 
 
+# SELL kernel variant bits (see kernels.hip): +1 non-temporal vals/cols,
+# +2 XCD-aware block swizzle, +4 unroll-8.
+SELL_NT, SELL_SWZ, SELL_U8 = 1, 2, 4
+DEFAULT_SELL_VARIANT = SELL_NT | SELL_SWZ
+
+
 def spmv_sell(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
               nrows: int, x: torch.Tensor, y: torch.Tensor, *,
               rowbase: int = 0, accum: bool = False,
               partials: torch.Tensor | None = None,
               scal: torch.Tensor | None = None, dotslot: int = -1,
-              dot_accum: bool = True) -> None:
+              dot_accum: bool = True, variant: int | None = None) -> None:
     """SELL-C-64 SpMV (regular-row fast path)."""
     nslices = sellptr.numel() - 1
     if nslices <= 0:
         return
+    if variant is None:
+        variant = DEFAULT_SELL_VARIANT
     fuse = scal is not None and dotslot >= 0
     K.spmv_sell(nslices, nrows, rowbase, sellptr.data_ptr(), cols.data_ptr(),
                 1 if cols.dtype == torch.int64 else 0,
                 vals.data_ptr(), x.data_ptr(), y.data_ptr(), accum,
                 partials.data_ptr() if fuse else 0,
-                scal.data_ptr() if fuse else 0, dotslot, dot_accum, _stream())
+                scal.data_ptr() if fuse else 0, dotslot, dot_accum,
+                variant, _stream())
 
 
 def zero_scalars(scal: torch.Tensor, i0: int = 0, count: int | None = None) -> None:

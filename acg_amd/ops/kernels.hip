@@ -129,7 +129,21 @@ spmv_csr_vector(long nrows, long rowbase,
 // slice (element j of row (s*64+lane) at sellptr[s] + j*64 + lane).  One wave
 // per slice: lane = row; every load is a contiguous 64-lane line; per-row sum
 // stays in-register (no cross-lane reduce, no pointer walk).
-template <typename ColT, bool ACCUM, bool FUSE_DOT>
+// Variant knobs (bitmask "variant" on the host API, A/B-tested on MI355X):
+//   NT : non-temporal loads for vals/cols (streamed exactly once per SpMV;
+//        no-allocate keeps the resident x vector from being evicted from
+//        L2 by the 4 GB vals/cols stream)
+//   SWZ: XCD-aware slice assignment.  The dispatcher places block b on XCD
+//        b%8 (guide §1); mapping contiguous slice ranges to one XCD makes
+//        each XCD's x working set ~1/8 of x (~4 MB at Queen scale = its L2).
+#define SELL_NT 1
+#define SELL_SWZ 2
+#define SELL_U8 4
+
+template <typename T>
+__device__ __forceinline__ T ld_nt(const T* p) { return __builtin_nontemporal_load(p); }
+
+template <typename ColT, bool ACCUM, bool FUSE_DOT, bool NT, bool SWZ, int UNROLL>
 __global__ void __launch_bounds__(BLOCK)
 k_spmv_sell(long nslices, long nrows, long rowbase,
           const long* __restrict__ sellptr,   // [nslices+1], element offsets
@@ -139,7 +153,15 @@ k_spmv_sell(long nslices, long nrows, long rowbase,
           double* __restrict__ y,
           double* __restrict__ partials) {
     const int lane = threadIdx.x & (WAVE - 1);
-    const long wslice = ((long)blockIdx.x * BLOCK + threadIdx.x) >> 6;
+    long blk = blockIdx.x;
+    if (SWZ) {
+        // bijective XCD remap of the BLOCK index (dispatcher places block b
+        // on XCD b%8): XCD k then owns a contiguous slice range, so its x
+        // working set is ~1/8 of x and fits the per-XCD 4 MB L2.
+        const long nb = gridDim.x, q = nb / 8, rr = nb % 8, xcd = blk % 8;
+        blk = (xcd < rr ? xcd * (q + 1) : rr * (q + 1) + (xcd - rr) * q) + blk / 8;
+    }
+    const long wslice = (blk * BLOCK + threadIdx.x) >> 6;
     const long nw = ((long)gridDim.x * BLOCK) >> 6;
     double dacc = 0.0;
     for (long s = wslice; s < nslices; s += nw) {
@@ -149,16 +171,22 @@ k_spmv_sell(long nslices, long nrows, long rowbase,
         const ColT* __restrict__ c = cols + base + lane;
         double sum = 0.0;
         long j = 0;
-        for (; j + 4 <= len; j += 4) {
-            // 4-deep unroll: 4 independent gathers in flight per lane
-            const double a0 = v[(j + 0) * WAVE], x0 = x[c[(j + 0) * WAVE]];
-            const double a1 = v[(j + 1) * WAVE], x1 = x[c[(j + 1) * WAVE]];
-            const double a2 = v[(j + 2) * WAVE], x2 = x[c[(j + 2) * WAVE]];
-            const double a3 = v[(j + 3) * WAVE], x3 = x[c[(j + 3) * WAVE]];
-            sum += a0 * x0; sum += a1 * x1; sum += a2 * x2; sum += a3 * x3;
+        for (; j + UNROLL <= len; j += UNROLL) {
+            double a[UNROLL], xx[UNROLL];
+            #pragma unroll
+            for (int u = 0; u < UNROLL; ++u) {
+                a[u] = NT ? ld_nt(v + (j + u) * WAVE) : v[(j + u) * WAVE];
+                const ColT ci = NT ? ld_nt(c + (j + u) * WAVE) : c[(j + u) * WAVE];
+                xx[u] = x[ci];
+            }
+            #pragma unroll
+            for (int u = 0; u < UNROLL; ++u) sum += a[u] * xx[u];
         }
-        for (; j < len; ++j)
-            sum += v[j * WAVE] * x[c[j * WAVE]];
+        for (; j < len; ++j) {
+            const double a = NT ? ld_nt(v + j * WAVE) : v[j * WAVE];
+            const ColT ci = NT ? ld_nt(c + j * WAVE) : c[j * WAVE];
+            sum += a * x[ci];
+        }
         const long row = s * WAVE + lane;
         if (row < nrows) {
             if (ACCUM) y[rowbase + row] += sum; else y[rowbase + row] = sum;
@@ -418,21 +446,32 @@ void spmv(long nrows, long rowbase, uintptr_t rowptr, uintptr_t colidx,
 void spmv_sell(long nslices, long nrows, long rowbase, uintptr_t sellptr,
                uintptr_t cols, int col64, uintptr_t vals, uintptr_t x,
                uintptr_t y, bool accum, uintptr_t partials, uintptr_t scal,
-               int dotslot, bool dot_accum, uintptr_t stream) {
+               int dotslot, bool dot_accum, int variant, uintptr_t stream) {
     if (nrows == 0) return;
     long blocks = (nslices * WAVE + BLOCK - 1) / BLOCK;
     if (blocks > MAXG) blocks = MAXG;
     const bool fuse = partials != 0 && dotslot >= 0;
     dim3 g((unsigned)blocks), b(BLOCK);
-    #define LAUNCH_SELL(CT, AC, FD) \
-        hipLaunchKernelGGL((k_spmv_sell<CT, AC, FD>), g, b, 0, S(stream), \
+    #define LAUNCH_SELL(CT, AC, FD, NT, SWZ, U) \
+        hipLaunchKernelGGL((k_spmv_sell<CT, AC, FD, NT, SWZ, U>), g, b, 0, S(stream), \
             nslices, nrows, rowbase, (const long*)sellptr, (const CT*)cols, \
             (const double*)vals, (const double*)x, (double*)y, (double*)partials)
+    #define DISP_V(CT, AC, FD) \
+        switch (variant & (SELL_NT | SELL_SWZ | SELL_U8)) { \
+            case 0:                               LAUNCH_SELL(CT, AC, FD, false, false, 4); break; \
+            case SELL_NT:                         LAUNCH_SELL(CT, AC, FD, true,  false, 4); break; \
+            case SELL_SWZ:                        LAUNCH_SELL(CT, AC, FD, false, true,  4); break; \
+            case SELL_NT | SELL_SWZ:              LAUNCH_SELL(CT, AC, FD, true,  true,  4); break; \
+            case SELL_U8:                         LAUNCH_SELL(CT, AC, FD, false, false, 8); break; \
+            case SELL_U8 | SELL_NT:               LAUNCH_SELL(CT, AC, FD, true,  false, 8); break; \
+            case SELL_U8 | SELL_SWZ:              LAUNCH_SELL(CT, AC, FD, false, true,  8); break; \
+            default:                              LAUNCH_SELL(CT, AC, FD, true,  true,  8); break; }
     #define DISP2(CT) \
-        if (accum) { if (fuse) { LAUNCH_SELL(CT, true, true); } else { LAUNCH_SELL(CT, true, false); } } \
-        else       { if (fuse) { LAUNCH_SELL(CT, false, true); } else { LAUNCH_SELL(CT, false, false); } }
+        if (accum) { if (fuse) { DISP_V(CT, true, true) } else { DISP_V(CT, true, false) } } \
+        else       { if (fuse) { DISP_V(CT, false, true) } else { DISP_V(CT, false, false) } }
     if (col64) { DISP2(long) } else { DISP2(int) }
     #undef DISP2
+    #undef DISP_V
     #undef LAUNCH_SELL
     check_hip("spmv_sell");
     if (fuse)
