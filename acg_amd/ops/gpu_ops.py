@@ -81,7 +81,10 @@ def spmv(rowptr: torch.Tensor, colidx: torch.Tensor, vals: torch.Tensor,
 # SELL kernel variant bits (see kernels.hip): +1 non-temporal vals/cols,
 # +2 XCD-aware block swizzle, +4 unroll-8.
 SELL_NT, SELL_SWZ, SELL_U8 = 1, 2, 4
-DEFAULT_SELL_VARIANT = SELL_NT | SELL_SWZ
+# A/B on MI355X (tools/spmv_bench.py, Queen-shaped 326M nnz): base 5.53,
+# NT 5.66, NT+U8 5.78 TB/s (92% of the ~6.3 TB/s achievable); SWZ slightly
+# negative (x is L3-resident, natural slice order already L2-local).
+DEFAULT_SELL_VARIANT = SELL_NT | SELL_U8
 
 
 def spmv_sell(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
