@@ -1,0 +1,281 @@
+"""acg-compatible CLI driver (reference: hip/acg-hip.c).
+
+Usage:  python -m acg_amd.cli [OPTION..] A.mtx [b.mtx] [x0.mtx]
+
+Mirrors the reference option surface (acg-hip.c:312-374) and its pipeline
+(§3.1 of SURVEY.md): read -> partition -> scatter -> solve -> report ->
+write solution to stdout.  Multi-rank runs come from torchrun (one process
+per GPU, RCCL) instead of mpirun; --comm gloo supports CPU multi-process.
+
+Solvers: acg (GPU classic CG), acg-pipelined (GPU pipelined),
+cpu / cpu-pipelined (host torch), scipy / scipy-pipelined (independent
+oracle, the role PETSc KSPCG plays in the reference -- PETSc is not in
+this image).
+"""
+
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+import time
+
+import numpy as np
+
+from . import __version__
+from .utils.errors import AcgError, ErrCode, collective_raise
+from .utils.numfmt import parse_numfmt
+
+
+def make_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(
+        prog="acg-amd",
+        description="MI355X-native distributed CG solver (aCG-compatible driver)")
+    p.add_argument("A", help="matrix Market file for the SPD matrix A")
+    p.add_argument("b", nargs="?", default=None, help="optional right-hand side")
+    p.add_argument("x0", nargs="?", default=None, help="optional initial guess")
+    p.add_argument("-z", "--gzip", action="store_true", help="gzip-compressed input")
+    p.add_argument("--binary", action="store_true",
+                   help="matrix file is in binary Matrix Market format (mtx2bin)")
+    p.add_argument("--idxsize", type=int, choices=(32, 64), default=64,
+                   help="index width of binary files (reference acgidx_t)")
+    p.add_argument("--partition", metavar="FILE", default=None,
+                   help="precomputed partition vector (mtx integer array)")
+    p.add_argument("--partition-method", choices=("block", "rgb"), default="block")
+    p.add_argument("--seed", type=int, default=0, help="partitioner seed")
+    p.add_argument("--solver", default=None,
+                   choices=("acg", "acg-pipelined", "acg-device", "cpu",
+                            "cpu-pipelined", "scipy", "scipy-pipelined"),
+                   help="default: acg on GPU, cpu otherwise")
+    p.add_argument("--max-iterations", type=int, default=100)
+    p.add_argument("--diff-atol", type=float, default=0.0)
+    p.add_argument("--diff-rtol", type=float, default=0.0)
+    p.add_argument("--residual-atol", type=float, default=0.0)
+    p.add_argument("--residual-rtol", type=float, default=1e-9)
+    p.add_argument("--epsilon", type=float, default=0.0,
+                   help="diagonal shift added to A")
+    p.add_argument("--warmup", type=int, default=0,
+                   help="untimed warmup iterations before the timed solve")
+    p.add_argument("--comm", choices=("none", "rccl", "gloo"), default=None,
+                   help="default: rccl when WORLD_SIZE>1 and GPUs exist")
+    p.add_argument("--manufactured-solution", action="store_true",
+                   help="b := A x* for random x*; report error norms")
+    p.add_argument("--numfmt", default=None, help="printf format for output values")
+    p.add_argument("--output-comm-matrix", action="store_true",
+                   help="write the rank x rank halo send-count matrix to stderr")
+    p.add_argument("-q", "--quiet", action="store_true",
+                   help="do not write the solution vector to stdout")
+    p.add_argument("-v", "--verbose", action="count", default=0)
+    p.add_argument("--version", action="version",
+                   version=f"acg-amd {__version__} (gfx950 HIP kernels, RCCL)")
+    return p
+
+
+def main(argv=None) -> int:
+    args = make_parser().parse_args(argv)
+    import torch
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    has_gpu = torch.cuda.is_available()
+    commkind = args.comm
+    if commkind is None:
+        commkind = ("rccl" if has_gpu else "gloo") if world > 1 else "none"
+    solver_name = args.solver or ("acg" if has_gpu else "cpu")
+    gpu_solver = solver_name.startswith("acg")
+    if gpu_solver and not has_gpu:
+        print("error: GPU solver requested but no GPU is available", file=sys.stderr)
+        return 1
+
+    from .dist.comm import Comm
+
+    device = None
+    if has_gpu:
+        local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+        device = torch.device("cuda", local_rank % max(torch.cuda.device_count(), 1))
+        torch.cuda.set_device(device)
+    comm = Comm(commkind, device=device if commkind == "rccl" else None) \
+        if commkind != "none" else None
+    nparts = comm.size if comm else 1
+    verbose = args.verbose and rank == 0
+
+    def log(msg):
+        if verbose:
+            print(msg, file=sys.stderr, flush=True)
+
+    numfmt = parse_numfmt(args.numfmt) if args.numfmt else None
+
+    # ---- root: read + partition + extract (reference acg-hip.c:1267-1652)
+    systems = None
+    b_pieces = None
+    n_global = None
+    err = None
+    try:
+        if rank == 0:
+            from .core.symcsr import SymCSRMatrix
+            from .io.mtx import read_mtx
+            from .part import extract_subdomains, partition_rows, read_partition_file
+
+            t0 = time.perf_counter()
+            m = read_mtx(args.A, gzipped=args.gzip, binary=args.binary,
+                         idxsize=args.idxsize)
+            log(f"read {args.A}: {m.nrows}x{m.ncols}, {m.nnz} stored entries "
+                f"({time.perf_counter() - t0:.2f}s)")
+            A = SymCSRMatrix.from_mtxfile(m)
+            if args.partition:
+                part = read_partition_file(args.partition, A.n)
+                if int(part.max()) >= nparts:
+                    raise AcgError(ErrCode.INVALID_VALUE,
+                                   f"partition file has {int(part.max()) + 1} parts, "
+                                   f"running with {nparts} ranks")
+            else:
+                part = partition_rows(A, nparts, seed=args.seed,
+                                      method=args.partition_method)
+            t0 = time.perf_counter()
+            systems = extract_subdomains(A, part, nparts, eps=args.epsilon)
+            log(f"partitioned into {nparts} subdomains "
+                f"({time.perf_counter() - t0:.2f}s)")
+            n_global = A.n
+
+            # RHS (reference acg-hip.c:1786-2087)
+            if args.manufactured_solution:
+                rng = np.random.default_rng(args.seed)
+                xsol = rng.standard_normal(A.n)
+                xsol /= np.linalg.norm(xsol)
+                b_global = A.dsymv(xsol)
+            elif args.b:
+                mb = read_mtx(args.b, gzipped=args.gzip)
+                b_global = np.asarray(mb.a, dtype=np.float64)
+                if len(b_global) != A.n:
+                    raise AcgError(ErrCode.INVALID_VALUE, "b length mismatch")
+                xsol = None
+            else:
+                b_global = np.ones(A.n, dtype=np.float64)
+                xsol = None
+            x0_global = None
+            if args.x0:
+                mx = read_mtx(args.x0, gzipped=args.gzip)
+                x0_global = np.asarray(mx.a, dtype=np.float64)
+            b_pieces = [(b_global[S.owned_global],
+                         None if x0_global is None else x0_global[S.owned_global])
+                        for S in systems]
+    except Exception as e:  # collective error agreement (acgerrmpi)
+        err = e
+    collective_raise(comm, err)
+
+    # ---- scatter (reference acgsymcsrmatrix_scatter, acg-hip.c:1752)
+    if comm:
+        S = comm.scatter_object(systems)
+        b_local, x0_local = comm.scatter_object(b_pieces)
+        n_global = comm.bcast_object(n_global)
+    else:
+        S = systems[0]
+        b_local, x0_local = b_pieces[0]
+
+    if args.output_comm_matrix and comm:
+        counts = comm.gather_object(
+            {int(q): int(c) for q, c in zip(S.halo.recipients, S.halo.sendcounts)})
+        if rank == 0:
+            print("%%MatrixMarket matrix coordinate integer general", file=sys.stderr)
+            entries = [(p, q, c) for p, row in enumerate(counts)
+                       for q, c in row.items()]
+            print(f"{nparts} {nparts} {len(entries)}", file=sys.stderr)
+            for p, q, c in entries:
+                print(f"{p + 1} {q + 1} {c}", file=sys.stderr)
+
+    import torch as _t
+
+    b = _t.from_numpy(np.ascontiguousarray(b_local))
+    x = _t.zeros(S.nowned + S.nghost, dtype=_t.float64)
+    if x0_local is not None:
+        x[:S.nowned] = _t.from_numpy(np.ascontiguousarray(x0_local))
+
+    # ---- solve (reference acg-hip.c:2192-2247)
+    err = None
+    res = None
+    try:
+        if solver_name in ("scipy", "scipy-pipelined"):
+            from .solvers.oracle import solve_scipy
+
+            res, xnp = solve_scipy(S, comm, b.numpy(), x[:S.nowned].numpy(),
+                                   maxits=args.max_iterations,
+                                   res_rtol=args.residual_rtol,
+                                   res_atol=args.residual_atol,
+                                   pipelined=solver_name.endswith("pipelined"))
+            x[:S.nowned] = _t.from_numpy(xnp)
+        elif gpu_solver:
+            from .solvers.hip import CGSolverHIP
+
+            solver = CGSolverHIP(S, comm=comm, device=device)
+            b = b.to(device)
+            x = x.to(device)
+            if args.warmup:
+                meth = solver.solve_pipelined if solver_name == "acg-pipelined" \
+                    else solver.solve
+                meth(b, x.clone(), maxits=args.warmup, res_rtol=0.0)
+            if solver_name == "acg-pipelined":
+                res = solver.solve_pipelined(b, x, maxits=args.max_iterations,
+                                             res_atol=args.residual_atol,
+                                             res_rtol=args.residual_rtol)
+            elif solver_name == "acg-device":
+                res = solver.solve_device(b, x, maxits=args.max_iterations,
+                                          res_atol=args.residual_atol,
+                                          res_rtol=args.residual_rtol)
+            else:
+                res = solver.solve(b, x, maxits=args.max_iterations,
+                                   res_atol=args.residual_atol,
+                                   res_rtol=args.residual_rtol)
+        else:
+            from .solvers.cpu import CGSolverCPU
+
+            solver = CGSolverCPU(S, comm=comm)
+            if solver_name == "cpu-pipelined":
+                res = solver.solve_pipelined(b, x, maxits=args.max_iterations,
+                                             res_atol=args.residual_atol,
+                                             res_rtol=args.residual_rtol)
+            else:
+                res = solver.solve(b, x, maxits=args.max_iterations,
+                                   res_atol=args.residual_atol,
+                                   res_rtol=args.residual_rtol,
+                                   diff_atol=args.diff_atol,
+                                   diff_rtol=args.diff_rtol)
+    except Exception as e:
+        err = e
+    collective_raise(comm, err)
+
+    # ---- report (reference acgsolverhip_fwritempi, acg-hip.c:2247)
+    if rank == 0:
+        print(res.summary(), file=sys.stderr)
+
+    x_host = x[:S.nowned].cpu().numpy()
+
+    # manufactured-solution error norms (reference acg-hip.c:2354-2362)
+    if args.manufactured_solution:
+        if comm:
+            xg = comm.gather_vector(x_host, S.owned_global, n_global)
+        else:
+            xg = np.empty(n_global)
+            xg[S.owned_global] = x_host
+        if rank == 0:
+            enorm = np.linalg.norm(xg - xsol)
+            print(f"manufactured solution: ||x-x*|| = {enorm:.6e} "
+                  f"(||x*|| = 1)", file=sys.stderr)
+
+    # solution to stdout as mtx array (reference acg-hip.c:2364-2403)
+    if not args.quiet:
+        if comm:
+            xg = comm.gather_vector(x_host, S.owned_global, n_global)
+        else:
+            xg = np.empty(n_global)
+            xg[S.owned_global] = x_host
+        if rank == 0:
+            from .io.mtx import vector_to_mtx, write_mtx
+
+            write_mtx(sys.stdout, vector_to_mtx(xg), numfmt=numfmt)
+    if comm:
+        comm.finalize()
+    return 0 if (res is None or res.converged or args.residual_rtol == 0) else 2
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -166,6 +166,21 @@ def pipelined_fused(z, t, p, x, r, w, q, scal: torch.Tensor,
                       _stream())
 
 
+def cg_device(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
+              nrows: int, b: torch.Tensor, x: torch.Tensor, r: torch.Tensor,
+              p: torch.Tensor, t: torch.Tensor, scal: torch.Tensor,
+              partials: torch.Tensor, out2: torch.Tensor, maxits: int,
+              res_atol: float, res_rtol: float) -> int:
+    """Monolithic device-side CG: one cooperative launch runs the whole
+    solve.  Returns the grid size used."""
+    nslices = sellptr.numel() - 1
+    return K.cg_device(nslices, nrows, sellptr.data_ptr(), cols.data_ptr(),
+                       1 if cols.dtype == torch.int64 else 0, vals.data_ptr(),
+                       b.data_ptr(), x.data_ptr(), r.data_ptr(), p.data_ptr(),
+                       t.data_ptr(), scal.data_ptr(), partials.data_ptr(),
+                       out2.data_ptr(), maxits, res_atol, res_rtol, _stream())
+
+
 def pack_gather(sendbuf: torch.Tensor, x: torch.Tensor, idx: torch.Tensor) -> None:
     K.pack_gather(sendbuf.data_ptr(), x.data_ptr(), idx.data_ptr(),
                   1 if idx.dtype == torch.int64 else 0,

@@ -380,6 +380,130 @@ k_pipelined_finalize(const double* __restrict__ partials, int nblocks,
     }
 }
 
+// ---------------------------------------------------------------------------
+// Monolithic device-side CG: the ENTIRE solver loop in one cooperative
+// launch -- zero per-iteration launch/sync overhead (reference
+// acgsolverhip_cg_kernel, cg-kernels-hip.hip:1386-1747; single-GPU, like
+// the reference's HIP build).  MI355X design: SELL-C-64 SpMV, grid-wide
+// barriers via cooperative groups, per-block partials reduced by block 0
+// (deterministic).  Residency: the launcher sizes the grid from the
+// occupancy query, everything inside is grid-stride.
+#include <hip/hip_cooperative_groups.h>
+
+// results: scal[S_BNRM2]=(b,b), scal[S_RR_PREV]=(r0,r0), scal[S_RR]=final
+// (r,r); out2[0]=niterations, out2[1]=converged.
+template <typename ColT>
+__global__ void __launch_bounds__(BLOCK)
+k_cg_device(long nslices, long nrows,
+            const long* __restrict__ sellptr, const ColT* __restrict__ cols,
+            const double* __restrict__ vals,
+            const double* __restrict__ b, double* __restrict__ x,
+            double* __restrict__ r, double* __restrict__ p,
+            double* __restrict__ t,
+            double* __restrict__ scal, double* __restrict__ partials,
+            int* __restrict__ out2,
+            int maxits, double res_atol, double res_rtol) {
+    namespace cg = cooperative_groups;
+    cg::grid_group grid = cg::this_grid();
+    const long tid = (long)blockIdx.x * BLOCK + threadIdx.x;
+    const long nth = (long)gridDim.x * BLOCK;
+    const int lane = threadIdx.x & (WAVE - 1);
+    const long wslice0 = tid >> 6;
+    const long nw = nth >> 6;
+
+    // grid-wide sum helper: every block contributes partials[bid]; block 0
+    // reduces into scal[slot]; two grid barriers bracket it.
+    auto grid_sum = [&](double v, int slot) {
+        v = block_reduce(v);
+        if (threadIdx.x == 0) partials[blockIdx.x] = v;
+        grid.sync();
+        if (blockIdx.x == 0) {
+            double s = 0.0;
+            for (int i = threadIdx.x; i < (int)gridDim.x; i += BLOCK)
+                s += partials[i];
+            s = block_reduce(s);
+            if (threadIdx.x == 0) scal[slot] = s;
+        }
+        grid.sync();
+    };
+    auto spmv = [&](const double* __restrict__ xin, double* __restrict__ yout,
+                    bool fuse) -> double {
+        double dacc = 0.0;
+        for (long s = wslice0; s < nslices; s += nw) {
+            const long base = sellptr[s];
+            const long len = (sellptr[s + 1] - base) >> 6;
+            const double* __restrict__ v = vals + base + lane;
+            const ColT* __restrict__ c = cols + base + lane;
+            double sum = 0.0;
+            long j = 0;
+            for (; j + 4 <= len; j += 4) {
+                double a0 = ld_nt(v + (j + 0) * WAVE), x0 = xin[c[(j + 0) * WAVE]];
+                double a1 = ld_nt(v + (j + 1) * WAVE), x1 = xin[c[(j + 1) * WAVE]];
+                double a2 = ld_nt(v + (j + 2) * WAVE), x2 = xin[c[(j + 2) * WAVE]];
+                double a3 = ld_nt(v + (j + 3) * WAVE), x3 = xin[c[(j + 3) * WAVE]];
+                sum += a0 * x0; sum += a1 * x1; sum += a2 * x2; sum += a3 * x3;
+            }
+            for (; j < len; ++j) sum += ld_nt(v + j * WAVE) * xin[c[j * WAVE]];
+            const long row = s * WAVE + lane;
+            if (row < nrows) {
+                yout[row] = sum;
+                if (fuse) dacc += xin[row] * sum;
+            }
+        }
+        return dacc;
+    };
+
+    // bnrm2; r0 = b - A x0; p = r0; rr0
+    double acc = 0.0;
+    for (long i = tid; i < nrows; i += nth) acc += b[i] * b[i];
+    grid_sum(acc, S_BNRM2);
+    spmv(x, t, false);
+    grid.sync();
+    acc = 0.0;
+    for (long i = tid; i < nrows; i += nth) {
+        const double ri = b[i] - t[i];
+        r[i] = ri;
+        p[i] = ri;
+        acc += ri * ri;
+    }
+    grid_sum(acc, S_RR);
+    const double bnrm2sqr = scal[S_BNRM2];
+    double rr = scal[S_RR];
+    if (tid == 0) scal[S_RR_PREV] = rr;  // report (r0,r0)
+    const double rt = res_rtol * sqrt(bnrm2sqr) > res_atol
+        ? res_rtol * sqrt(bnrm2sqr) : res_atol;
+    const double rtol2 = rt * rt;
+    int k = 0, converged = (rtol2 > 0.0 && rr <= rtol2) ? 1 : 0;
+    while (!converged && k < maxits) {
+        // t = A p (fused (p,t))
+        grid.sync();  // p is consistent (written by all blocks last iter)
+        const double pt_part = spmv(p, t, true);
+        grid_sum(pt_part, S_PT);
+        const double alpha = rr / scal[S_PT];
+        acc = 0.0;
+        for (long i = tid; i < nrows; i += nth) {
+            const double rn = r[i] - alpha * t[i];
+            r[i] = rn;
+            x[i] += alpha * p[i];
+            acc += rn * rn;
+        }
+        grid_sum(acc, S_RR);
+        const double rr_new = scal[S_RR];
+        const double beta = rr_new / rr;
+        for (long i = tid; i < nrows; i += nth)
+            p[i] = beta * p[i] + r[i];
+        rr = rr_new;
+        ++k;
+        if (rtol2 > 0.0 && rr <= rtol2) converged = 1;
+    }
+    if (tid == 0) {
+        scal[S_RR] = rr;
+        out2[0] = k;
+        out2[1] = converged;
+    }
+    (void)bnrm2sqr;
+}
+
 // halo pack: sendbuf[i] = x[sendidx[i]]
 // (reference acghalo_pack_hip_double, halo-kernels-hip.hip:48-103; no unpack
 // kernel exists -- ghosts are received in place, see dist/halo.py)
@@ -553,6 +677,49 @@ void pipelined_fused(uintptr_t z, uintptr_t t, uintptr_t p, uintptr_t x, uintptr
     check_hip("pipelined_finalize");
 }
 
+// cooperative launch of the monolithic device CG (reference
+// acgsolverhip_solve_device geometry logic, cg-kernels-hip.hip:1892-1909:
+// grid = CUs x occupancy, everything grid-stride inside)
+int cg_device(long nslices, long nrows, uintptr_t sellptr, uintptr_t cols,
+              int col64, uintptr_t vals, uintptr_t b, uintptr_t x, uintptr_t r,
+              uintptr_t p, uintptr_t t, uintptr_t scal, uintptr_t partials,
+              uintptr_t out2, int maxits, double res_atol, double res_rtol,
+              uintptr_t stream) {
+    int dev = 0;
+    hipGetDevice(&dev);
+    hipDeviceProp_t props;
+    hipGetDeviceProperties(&props, dev);
+    const void* kern = col64 ? (const void*)&k_cg_device<long>
+                             : (const void*)&k_cg_device<int>;
+    int blocks_per_cu = 0;
+    hipError_t oe = hipOccupancyMaxActiveBlocksPerMultiprocessor(
+        &blocks_per_cu, kern, BLOCK, 0);
+    if (oe != hipSuccess || blocks_per_cu < 1) blocks_per_cu = 1;
+    long grid = (long)props.multiProcessorCount * blocks_per_cu;
+    long need = (nslices * WAVE + BLOCK - 1) / BLOCK;
+    if (grid > need) grid = need;
+    if (grid > MAXG) grid = MAXG;
+    // the occupancy API can over-report by one block/CU (guide §1); the
+    // cooperative launch checks residency -- shrink and retry on rejection.
+    for (;;) {
+        void* args[] = {&nslices, &nrows, (void*)&sellptr, (void*)&cols,
+                        (void*)&vals, (void*)&b, (void*)&x, (void*)&r,
+                        (void*)&p, (void*)&t, (void*)&scal, (void*)&partials,
+                        (void*)&out2, &maxits, &res_atol, &res_rtol};
+        hipError_t e = hipLaunchCooperativeKernel(
+            kern, dim3((unsigned)grid), dim3(BLOCK), args, 0, S(stream));
+        if (e == hipSuccess) break;
+        if (e == hipErrorCooperativeLaunchTooLarge && grid > props.multiProcessorCount) {
+            (void)hipGetLastError();
+            grid -= props.multiProcessorCount;
+            continue;
+        }
+        throw std::runtime_error(std::string("cg_device launch: ") + hipGetErrorString(e));
+    }
+    check_hip("cg_device");
+    return (int)grid;
+}
+
 void pack_gather(uintptr_t sendbuf, uintptr_t x, uintptr_t idx, int idx64, long n,
                  uintptr_t stream) {
     if (n == 0) return;
@@ -580,6 +747,7 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.def("cg_fused_update", &cg_fused_update);
     m.def("pipelined_fused", &pipelined_fused);
     m.def("pack_gather", &pack_gather);
+    m.def("cg_device", &cg_device);
     m.attr("S_RR") = S_RR;
     m.attr("S_PT") = S_PT;
     m.attr("S_RR_PREV") = S_RR_PREV;

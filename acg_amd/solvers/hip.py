@@ -213,6 +213,50 @@ class CGSolverHIP:
         self.niterations_total += res.niterations
         return res
 
+    # -- monolithic device-side CG ---------------------------------------
+
+    def solve_device(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
+                     res_atol: float = 0.0, res_rtol: float = 1e-9) -> SolveResult:
+        """Whole CG solve in ONE cooperative kernel launch (reference
+        acgsolverhip_solve_device, §3.4; single-GPU only, as in the
+        reference's HIP build -- cg-kernels-hip.hip:1832).  Zero
+        per-iteration launch overhead: wins in the launch-bound regime
+        (small/medium systems)."""
+        if self.comm is not None and self.comm.size > 1:
+            from ..utils.errors import AcgError, ErrCode
+
+            raise AcgError(ErrCode.NOT_SUPPORTED,
+                           "device-side CG is single-GPU (reference parity)")
+        if self.sell is None:
+            from ..utils.errors import AcgError, ErrCode
+
+            raise AcgError(ErrCode.NOT_SUPPORTED,
+                           "device-side CG requires the SELL operator format")
+        res = SolveResult(solver="cg-hip-device", maxits=maxits,
+                          res_atol=res_atol, res_rtol=res_rtol, nranks=1)
+        n = self.n
+        r = self._vec()
+        t = self._vec()
+        p = self._vec(nghost=True)
+        out2 = torch.zeros(2, dtype=torch.int32, device=self.device)
+        sellptr, scols, svals = self.sell
+        torch.cuda.synchronize(self.device)
+        t0 = time.perf_counter()
+        ops.cg_device(sellptr, scols, svals, n, b, x, r, p, t, self.scal,
+                      self.partials, out2, maxits, res_atol, res_rtol)
+        torch.cuda.synchronize(self.device)
+        res.tsolve = time.perf_counter() - t0
+        S = ops
+        res.bnrm2 = math.sqrt(max(float(self.scal[S.S_BNRM2]), 0.0))
+        res.r0nrm2 = math.sqrt(max(float(self.scal[S.S_RR_PREV]), 0.0))
+        res.rnrm2 = math.sqrt(max(float(self.scal[S.S_RR]), 0.0))
+        res.niterations = int(out2[0])
+        res.converged = bool(int(out2[1]))
+        nnz_full = self.local.nnzA + self.local.nnzO
+        res.nflops = res.niterations * cg_flops_per_iter(nnz_full, n)
+        self.niterations_total += res.niterations
+        return res
+
     # -- pipelined CG -----------------------------------------------------
 
     def solve_pipelined(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,

@@ -75,6 +75,24 @@ def test_hip_pipelined_matches_classic(problem):
     torch.testing.assert_close(x1[:S.nowned], x2[:S.nowned], rtol=1e-6, atol=1e-8)
 
 
+def test_hip_device_monolithic_cg(problem):
+    """Whole-solve cooperative kernel vs the host-driven classic solver."""
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A, S = problem
+    rng = np.random.default_rng(4)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+    gpu = CGSolverHIP(S, device="cuda:0")
+    assert gpu.sell is not None
+    x1 = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
+    r1 = gpu.solve(b, x1, maxits=400, res_rtol=1e-10)
+    x2 = torch.zeros_like(x1)
+    r2 = gpu.solve_device(b, x2, maxits=400, res_rtol=1e-10)
+    assert r1.converged and r2.converged, (r1.summary(), r2.summary())
+    assert abs(r1.niterations - r2.niterations) <= 2
+    torch.testing.assert_close(x1[:S.nowned], x2[:S.nowned], rtol=1e-7, atol=1e-9)
+
+
 def test_slab_generated_gpu_solve():
     """Flagship path: slab-generated Queen-like system, single GPU."""
     from acg_amd.gen import queen_like_spec, stencil_local_slab
