@@ -63,6 +63,8 @@ def make_parser() -> argparse.ArgumentParser:
     p.add_argument("--numfmt", default=None, help="printf format for output values")
     p.add_argument("--output-comm-matrix", action="store_true",
                    help="write the rank x rank halo send-count matrix to stderr")
+    p.add_argument("--profile", action="store_true",
+                   help="per-op hipEvent timing (reference ACG_ENABLE_PROFILING)")
     p.add_argument("-q", "--quiet", action="store_true",
                    help="do not write the solution vector to stdout")
     p.add_argument("-v", "--verbose", action="count", default=0)
@@ -206,7 +208,8 @@ def main(argv=None) -> int:
         elif gpu_solver:
             from .solvers.hip import CGSolverHIP
 
-            solver = CGSolverHIP(S, comm=comm, device=device)
+            solver = CGSolverHIP(S, comm=comm, device=device,
+                                 profile=args.profile)
             b = b.to(device)
             x = x.to(device)
             if args.warmup:
@@ -244,8 +247,9 @@ def main(argv=None) -> int:
     collective_raise(comm, err)
 
     # ---- report (reference acgsolverhip_fwritempi, acg-hip.c:2247)
-    if rank == 0:
-        print(res.summary(), file=sys.stderr)
+    from .solvers.profiling import write_stats
+
+    write_stats(res, S, comm, file=sys.stderr)
 
     x_host = x[:S.nowned].cpu().numpy()
 

@@ -39,7 +39,8 @@ class CGSolverHIP:
     """Distributed CG on one MI355X per rank (classic + pipelined)."""
 
     def __init__(self, local: LocalSystem, comm=None, device=None,
-                 lanes: int | None = None, use_sell: bool = True):
+                 lanes: int | None = None, use_sell: bool = True,
+                 profile: bool = False):
         self.local = local
         self.comm = comm
         if device is None:
@@ -78,6 +79,9 @@ class CGSolverHIP:
         self.halo = HaloExchange(L.halo, L.nowned, self.device, comm)
         self.scal = ops.alloc_scalars(self.device)
         self.partials = ops.alloc_partials(self.device)
+        from .profiling import EventProfiler
+
+        self.prof = EventProfiler(profile)
         self.comm_stream = torch.cuda.Stream(self.device)
         self.copy_stream = torch.cuda.Stream(self.device)
         self._rr_host = torch.zeros(1, dtype=torch.float64, pin_memory=True)
@@ -108,21 +112,23 @@ class CGSolverHIP:
         fuse = dict(partials=self.partials,
                     scal=self.scal if fuse_dotslot >= 0 else None,
                     dotslot=fuse_dotslot)
-        if self.sell is not None:
-            sellptr, scols, svals = self.sell
-            ops.spmv_sell(sellptr, scols, svals, self.n, xfull, y,
-                          accum=False, **fuse)
-        else:
-            ops.spmv(self.A_rowptr, self.A_colidx, self.A_vals, xfull, y,
-                     lanes=self.lanesA, accum=False, **fuse)
+        with self.prof.span("spmvA"):
+            if self.sell is not None:
+                sellptr, scols, svals = self.sell
+                ops.spmv_sell(sellptr, scols, svals, self.n, xfull, y,
+                              accum=False, **fuse)
+            else:
+                ops.spmv(self.A_rowptr, self.A_colidx, self.A_vals, xfull, y,
+                         lanes=self.lanesA, accum=False, **fuse)
         if have_halo:
             with torch.cuda.stream(self.comm_stream):
                 self.halo.end()
                 self._ev_recv.record(self.comm_stream)
             cur.wait_event(self._ev_recv)
         if L.nborder > 0 and self.local.nnzO > 0:
-            ops.spmv(self.O_rowptr, self.O_colidx, self.O_vals, xfull, y,
-                     rowbase=L.ninterior, lanes=self.lanesO, accum=True, **fuse)
+            with self.prof.span("spmvO"):
+                ops.spmv(self.O_rowptr, self.O_colidx, self.O_vals, xfull, y,
+                         rowbase=L.ninterior, lanes=self.lanesO, accum=True, **fuse)
 
     def _host_scalar(self, slot: int) -> float:
         cur = torch.cuda.current_stream(self.device)
@@ -185,13 +191,17 @@ class CGSolverHIP:
             # the (p,t) reduction fused into both SpMV passes
             S.cg_prep_pt(scal)
             self._spmv_overlapped(p, t, fuse_dotslot=S.S_PT)
-            self._allreduce_slot(S.S_PT)
+            with self.prof.span("allreduce"):
+                self._allreduce_slot(S.S_PT)
             # rr_prev = rr; then fused r/x update + finalized (r,r)
             S.cg_prep_rr(scal)
-            S.cg_fused_update(r, x, p, t, scal, self.partials, n)
-            self._allreduce_slot(S.S_RR)
+            with self.prof.span("update_classic"):
+                S.cg_fused_update(r, x, p, t, scal, self.partials, n)
+            with self.prof.span("allreduce"):
+                self._allreduce_slot(S.S_RR)
             # p = (rr/rr_prev) p + r
-            S.daypx_ratio(p, r, scal, S.S_RR, S.S_RR_PREV, n=n)
+            with self.prof.span("daypx"):
+                S.daypx_ratio(p, r, scal, S.S_RR, S.S_RR_PREV, n=n)
             k += 1
             res.niterations = k
             if k % check_every == 0 or k == maxits:
@@ -211,6 +221,11 @@ class CGSolverHIP:
         res.halo_bytes_sent = self.halo.bytes_sent
         res.halo_msgs_sent = self.halo.nmsgs_sent
         self.niterations_total += res.niterations
+        if self.prof.enabled:
+            from .profiling import annotate_op_stats
+
+            annotate_op_stats(res, self.local, self.prof.collect())
+            self.prof.reset()
         return res
 
     # -- monolithic device-side CG ---------------------------------------
@@ -297,7 +312,8 @@ class CGSolverHIP:
         while k < maxits:
             first = (k == 0)
             # ONE 2-double allreduce per iteration (gamma,delta adjacent)
-            self._allreduce_slot(S.S_GAMMA, 2)
+            with self.prof.span("allreduce"):
+                self._allreduce_slot(S.S_GAMMA, 2)
             # kick off the async D2H of gamma for the host convergence test
             cur = torch.cuda.current_stream(self.device)
             self._ev_rr.record(cur)
@@ -319,7 +335,8 @@ class CGSolverHIP:
                 res.niterations = k
                 break
             # fused: 6-vector update + next (r,r),(w,r) + scalar rotation
-            S.pipelined_fused(z, t, p, x, r, w, q, scal, self.partials, n, first)
+            with self.prof.span("update"):
+                S.pipelined_fused(z, t, p, x, r, w, q, scal, self.partials, n, first)
             k += 1
             res.niterations = k
         torch.cuda.synchronize(self.device)
@@ -332,4 +349,9 @@ class CGSolverHIP:
         res.halo_bytes_sent = self.halo.bytes_sent
         res.halo_msgs_sent = self.halo.nmsgs_sent
         self.niterations_total += res.niterations
+        if self.prof.enabled:
+            from .profiling import annotate_op_stats
+
+            annotate_op_stats(res, self.local, self.prof.collect())
+            self.prof.reset()
         return res

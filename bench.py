@@ -48,7 +48,9 @@ def main() -> int:
 
     device = torch.device("cuda", local_rank % max(torch.cuda.device_count(), 1))
     torch.cuda.set_device(device)
-    comm = Comm("rccl", device=device) if world > 1 else None
+    # init RCCL whenever torchrun launched us (even world=1: exercises the
+    # same bootstrap the driver's multi-GPU runs use)
+    comm = Comm("rccl", device=device) if "RANK" in os.environ else None
 
     G = args.grid
     spec = queen_like_spec(args.dof)
