@@ -53,29 +53,38 @@ class CGSolverHIP:
             t = torch.from_numpy(np.ascontiguousarray(a))
             return t.to(self.device, non_blocking=True)
 
-        self.A_rowptr = up(L.A_rowptr)
-        self.A_colidx = up(L.A_colidx)
-        self.A_vals = up(L.A_vals)
-        self.O_rowptr = up(L.O_rowptr)
-        self.O_colidx = up(L.O_colidx)
-        self.O_vals = up(L.O_vals)
         self.n = L.nowned
         self.nlocal = L.nowned + L.nghost
-        mean_nnz = L.nnzA / max(L.nowned, 1)
-        self.lanesA = lanes or ops.pick_lanes(mean_nnz)
-        mean_nnzO = L.nnzO / max(L.nborder, 1)
-        self.lanesO = lanes or ops.pick_lanes(mean_nnzO)
-        # SELL-C-64 fast path for matA when rows are regular enough that
-        # padding stays small (stencil/FEM matrices); CSR-vector otherwise.
         self.sell = None
-        if use_sell and L.nowned > 0:
-            from ..ops.torch_ref import sell_from_csr
+        self.sellO = None
+        self.A_rowptr = self.A_colidx = self.A_vals = None
+        self.O_rowptr = self.O_colidx = self.O_vals = None
+        self.lanesA = self.lanesO = lanes or 16
+        if hasattr(L, "A_sell"):
+            # device-generated system (gen.device_slab): SELL already in HBM
+            self.sell = L.A_sell
+            self.sellO = L.O_sell
+        else:
+            self.A_rowptr = up(L.A_rowptr)
+            self.A_colidx = up(L.A_colidx)
+            self.A_vals = up(L.A_vals)
+            self.O_rowptr = up(L.O_rowptr)
+            self.O_colidx = up(L.O_colidx)
+            self.O_vals = up(L.O_vals)
+            mean_nnz = L.nnzA / max(L.nowned, 1)
+            self.lanesA = lanes or ops.pick_lanes(mean_nnz)
+            mean_nnzO = L.nnzO / max(L.nborder, 1)
+            self.lanesO = lanes or ops.pick_lanes(mean_nnzO)
+            # SELL-C-64 fast path for matA when rows are regular enough that
+            # padding stays small (stencil/FEM); CSR-vector otherwise.
+            if use_sell and L.nowned > 0:
+                from ..ops.torch_ref import sell_from_csr
 
-            sellptr, scols, svals = sell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals)
-            waste = (int(sellptr[-1]) - L.nnzA) / max(L.nnzA, 1)
-            if waste <= 0.3:
-                self.sell = (up(sellptr), up(scols), up(svals))
-                self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR copy
+                sellptr, scols, svals = sell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals)
+                waste = (int(sellptr[-1]) - L.nnzA) / max(L.nnzA, 1)
+                if waste <= 0.3:
+                    self.sell = (up(sellptr), up(scols), up(svals))
+                    self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR
         self.halo = HaloExchange(L.halo, L.nowned, self.device, comm)
         self.scal = ops.alloc_scalars(self.device)
         self.partials = ops.alloc_partials(self.device)
@@ -127,8 +136,14 @@ class CGSolverHIP:
             cur.wait_event(self._ev_recv)
         if L.nborder > 0 and self.local.nnzO > 0:
             with self.prof.span("spmvO"):
-                ops.spmv(self.O_rowptr, self.O_colidx, self.O_vals, xfull, y,
-                         rowbase=L.ninterior, lanes=self.lanesO, accum=True, **fuse)
+                if self.sellO is not None:
+                    optr, ocols, ovals = self.sellO
+                    ops.spmv_sell(optr, ocols, ovals, L.nborder, xfull, y,
+                                  rowbase=L.ninterior, accum=True, **fuse)
+                else:
+                    ops.spmv(self.O_rowptr, self.O_colidx, self.O_vals, xfull, y,
+                             rowbase=L.ninterior, lanes=self.lanesO, accum=True,
+                             **fuse)
 
     def _host_scalar(self, slot: int) -> float:
         cur = torch.cuda.current_stream(self.device)

@@ -79,7 +79,8 @@ def annotate_op_stats(res: SolveResult, local, ops_times: dict) -> None:
     n = local.nowned
     it = max(res.niterations, 1)
     nnzA, nnzO = local.nnzA, local.nnzO
-    colb = local.A_colidx.dtype.itemsize if hasattr(local.A_colidx, "dtype") else 4
+    _ac = getattr(local, "A_colidx", None)
+    colb = _ac.dtype.itemsize if _ac is not None and hasattr(_ac, "dtype") else 4
     model = {
         "spmvA": (2.0 * nnzA, nnzA * (8 + colb) + 16.0 * n),
         "spmvO": (2.0 * nnzO, nnzO * (8 + colb) + 16.0 * local.nborder),

@@ -30,9 +30,15 @@ def main() -> int:
     ap.add_argument("--steps", type=int, default=200)
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--solver", choices=["pipelined", "classic"], default="pipelined")
-    ap.add_argument("--grid", type=int, default=111,
-                    help="grid edge G; rows = 3*G^3 (111 -> 4.10M rows, Queen_4147 scale)")
-    ap.add_argument("--dof", type=int, default=3)
+    ap.add_argument("--config", choices=["queen", "poisson7"], default="queen",
+                    help="queen: 27-pt dof-3 Queen_4147-shaped (BASELINE configs 2-4); "
+                         "poisson7: 7-pt 3D Poisson (BASELINE config 5 sizing)")
+    ap.add_argument("--grid", type=int, default=None,
+                    help="grid edge G (queen default 111 -> 4.10M rows; "
+                         "poisson7 default 512)")
+    ap.add_argument("--dof", type=int, default=None)
+    ap.add_argument("--gen", choices=["device", "host"], default="device",
+                    help="matrix generation: on-GPU SELL (default) or host numpy")
     ap.add_argument("--lanes", type=int, default=None,
                     help="override SpMV lanes-per-row (4/8/16/32/64)")
     args = ap.parse_args()
@@ -43,7 +49,7 @@ def main() -> int:
     ngpus = max(world, 1)
 
     from acg_amd.dist.comm import Comm
-    from acg_amd.gen import queen_like_spec, stencil_local_slab
+    from acg_amd.gen import STENCIL_7PT_3D, queen_like_spec, stencil_local_slab
     from acg_amd.solvers.hip import CGSolverHIP
 
     device = torch.device("cuda", local_rank % max(torch.cuda.device_count(), 1))
@@ -52,9 +58,23 @@ def main() -> int:
     # same bootstrap the driver's multi-GPU runs use)
     comm = Comm("rccl", device=device) if "RANK" in os.environ else None
 
-    G = args.grid
-    spec = queen_like_spec(args.dof)
-    S = stencil_local_slab(G, G, G, spec, rank, ngpus)
+    if args.config == "queen":
+        dof = args.dof or 3
+        G = args.grid or 111
+        spec = queen_like_spec(dof)
+        model = f"queen4147-like-27pt-dof{dof}-G{G}"
+    else:
+        dof = args.dof or 1
+        G = args.grid or 512
+        spec = dict(STENCIL_7PT_3D)
+        spec["dof"] = dof
+        model = f"poisson3d-7pt-G{G}"
+    if args.gen == "device":
+        from acg_amd.gen.device_slab import device_stencil_slab
+
+        S = device_stencil_slab(G, G, G, spec, rank, ngpus, device)
+    else:
+        S = stencil_local_slab(G, G, G, spec, rank, ngpus)
     solver = CGSolverHIP(S, comm=comm, device=device, lanes=args.lanes)
 
     rloc = np.random.default_rng(10_000 + rank)
@@ -86,7 +106,7 @@ def main() -> int:
         elapsed = float(et.item())
 
     assert res.niterations == args.steps, (res.niterations, args.steps)
-    nrows_global = args.dof * G * G * G
+    nrows_global = dof * G * G * G
     nnz_local = S.nnzA + S.nnzO
     if comm is not None:
         import torch.distributed as dist
@@ -101,7 +121,9 @@ def main() -> int:
     ms_per_step = 1000.0 * elapsed / args.steps
     if rank == 0:
         out = {
-            "metric": "CG iter/s (whole node), Queen_4147-shaped fp64",
+            "metric": ("CG iter/s (whole node), Queen_4147-shaped fp64"
+                       if args.config == "queen" else
+                       "CG iter/s (whole node), 7-pt 3D Poisson fp64"),
             "value": iters_per_s,
             "unit": "iter/s",
             "n_gpus": ngpus,
@@ -112,9 +134,11 @@ def main() -> int:
             "scaling": "strong",
             "vs_baseline": None,
             "dtype": "fp64",
-            "data": "synthetic SPD (27-pt dof-3 stencil, Queen_4147 shape; random RHS)",
+            "data": ("synthetic SPD (27-pt dof-3 stencil, Queen_4147 shape; random RHS)"
+                     if args.config == "queen" else
+                     "synthetic SPD (7-pt 3D Poisson; random RHS)"),
             "config": {
-                "model": f"queen4147-like-27pt-dof{args.dof}-G{G}",
+                "model": model,
                 "rows": nrows_global,
                 "nnz": nnz_global,
                 "solver": f"cg-{args.solver}",
