@@ -500,7 +500,7 @@ void stencil_rowlen(long nrows_nodes, long row0_node, int gx, int gy, int gz,
                     uintptr_t pb, uintptr_t offs, int ksten, int filter_ghost,
                     uintptr_t rowlen, uintptr_t stream) {
     hipLaunchKernelGGL(k_stencil_rowlen, dim3((unsigned)elem_grid(nrows_nodes)),
-                       dim3(BLOCK), 0, S(stream),
+                       dim3(BLOCK), 0, (hipStream_t)stream,
                        nrows_nodes, row0_node, gx, gy, gz, dof, nown_nodes,
                        (const int*)zs_of_plane, (const long*)pb,
                        (const double*)offs, ksten, filter_ghost, (long*)rowlen);
@@ -514,7 +514,7 @@ void stencil_fill(long nrows_nodes, long row0_node, int gx, int gy, int gz,
                   uintptr_t vals, uintptr_t stream) {
     const long nrows = nrows_nodes * dof;
     hipLaunchKernelGGL(k_stencil_fill, dim3((unsigned)elem_grid(nrows)),
-                       dim3(BLOCK), 0, S(stream),
+                       dim3(BLOCK), 0, (hipStream_t)stream,
                        nrows_nodes, row0_node, gx, gy, gz, dof, nown_nodes,
                        (const int*)zs_of_plane, (const long*)pb,
                        (const double*)offs, ksten, (const double*)blocks,
@@ -891,6 +891,8 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.def("pipelined_fused", &pipelined_fused);
     m.def("pack_gather", &pack_gather);
     m.def("cg_device", &cg_device);
+    m.def("stencil_rowlen", &stencil_rowlen);
+    m.def("stencil_fill", &stencil_fill);
     m.attr("S_RR") = S_RR;
     m.attr("S_PT") = S_PT;
     m.attr("S_RR_PREV") = S_RR_PREV;
