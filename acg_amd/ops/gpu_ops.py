@@ -157,6 +157,29 @@ def cg_fused_update(r: torch.Tensor, x: torch.Tensor, p: torch.Tensor,
                       n, scal.data_ptr(), partials.data_ptr(), _stream())
 
 
+def sell_pipe(sellptr, cols, vals, nrows_pass: int, rowbase: int,
+              border_base: int, w_old, qpart, z, t, p, x, r, w_new,
+              scal, first: bool, partials, partials_off: int,
+              mato: bool) -> int:
+    """Megafused pipelined iteration pass (SpMV + 6-vector update + dots).
+    Returns the number of partial blocks written."""
+    nslices = sellptr.numel() - 1
+    assert cols.dtype == torch.int32
+    return K.sell_pipe(nslices, nrows_pass, rowbase, border_base,
+                       sellptr.data_ptr(), cols.data_ptr(), vals.data_ptr(),
+                       w_old.data_ptr(),
+                       qpart.data_ptr() if qpart is not None else 0,
+                       z.data_ptr(), t.data_ptr(), p.data_ptr(), x.data_ptr(),
+                       r.data_ptr(), w_new.data_ptr(), scal.data_ptr(),
+                       1 if first else 0, partials.data_ptr(), partials_off,
+                       mato, _stream())
+
+
+def pipelined_finalize(partials, nblocks: int, scal, first: bool) -> None:
+    K.pipelined_finalize(partials.data_ptr(), nblocks, scal.data_ptr(),
+                         1 if first else 0, _stream())
+
+
 def pipelined_fused(z, t, p, x, r, w, q, scal: torch.Tensor,
                     partials: torch.Tensor, n: int, first: bool) -> None:
     """Fused pipelined update + next gamma/delta + scalar rotation."""

@@ -85,6 +85,15 @@ class CGSolverHIP:
                 if waste <= 0.3:
                     self.sell = (up(sellptr), up(scols), up(svals))
                     self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR
+                    if L.nnzO > 0:
+                        optr, ocols, ovals = sell_from_csr(L.O_rowptr, L.O_colidx,
+                                                           L.O_vals)
+                        self.sellO = (up(optr), up(ocols), up(ovals))
+        # megafused pipelined iteration needs SELL everywhere + int32 cols
+        self.can_megafuse = (
+            self.sell is not None and self.sell[1].dtype == torch.int32
+            and (L.nnzO == 0 or (self.sellO is not None
+                                 and self.sellO[1].dtype == torch.int32)))
         self.halo = HaloExchange(L.halo, L.nowned, self.device, comm)
         self.scal = ops.alloc_scalars(self.device)
         self.partials = ops.alloc_partials(self.device)
@@ -311,14 +320,22 @@ class CGSolverHIP:
                           nranks=self.comm.size if self.comm else 1)
         n = self.n
         S = ops
+        L = self.local
         scal = self.scal
+        mega = self.can_megafuse
         r = self._vec(nghost=True)
         w = self._vec(nghost=True)
-        q = self._vec()
         z = self._vec()
         t = self._vec()
         p = self._vec()
         tmp = self._vec()
+        # megafused path double-buffers w (SpMV gathers w_old while the
+        # fused epilogue writes w_new) and stages border q in qpart;
+        # the separate q vector exists only on the fallback path.
+        w2 = self._vec(nghost=True) if mega else None
+        qpart = (torch.zeros(max(L.nborder, 1), dtype=torch.float64,
+                             device=self.device) if mega else None)
+        q = None if mega else self._vec()
         torch.cuda.synchronize(self.device)
         t0 = time.perf_counter()
         S.dot(b, b, self.partials, scal, S.S_BNRM2, n=n)
@@ -335,7 +352,37 @@ class CGSolverHIP:
         gamma_host = None
         serial = self.comm is None or self.comm.size == 1
         graph = None
+        graphs = [None, None]  # megafused: even/odd w ping-pong
         graph_ok = use_graph and serial and not self.prof.enabled
+
+        def mega_body(wa, wb, first):
+            """One megafused iteration: halo(wa) || matA pass (SpMV + update
+            of interior rows), then matO pass finishing border rows."""
+            have_halo = not serial
+            cur = torch.cuda.current_stream(self.device)
+            if have_halo:
+                self._ev_p.record(cur)
+                self.comm_stream.wait_event(self._ev_p)
+                with torch.cuda.stream(self.comm_stream):
+                    self.halo.begin(wa)
+            sp, sc, sv = self.sell
+            border_base = L.ninterior if L.nnzO > 0 else n
+            nbA = S.sell_pipe(sp, sc, sv, n, 0, border_base, wa, qpart,
+                              z, t, p, x, r, wb, scal, first,
+                              self.partials, 0, mato=False)
+            nb = nbA
+            if have_halo:
+                with torch.cuda.stream(self.comm_stream):
+                    self.halo.end()
+                    self._ev_recv.record(self.comm_stream)
+                cur.wait_event(self._ev_recv)
+            if L.nnzO > 0:
+                op_, oc, ov = self.sellO
+                nbO = S.sell_pipe(op_, oc, ov, L.nborder, L.ninterior,
+                                  L.ninterior, wa, qpart, z, t, p, x, r, wb,
+                                  scal, first, self.partials, nbA, mato=True)
+                nb += nbO
+            S.pipelined_finalize(self.partials, nb, scal, first)
         # lag-1 convergence pipeline: gamma_k is copied to the host as soon
         # as its allreduce lands, but the host *reads* it one iteration
         # later -- the test still runs for every iteration, the host just
@@ -388,8 +435,19 @@ class CGSolverHIP:
             if k >= LAG and check(k - LAG):
                 break
             issue_gamma_copy(k)
-            # iteration body: halo+SpMV(q = A w) then the fused update
-            if graph is not None:
+            if mega:
+                wa, wb = (w, w2) if k % 2 == 0 else (w2, w)
+                g = graphs[k % 2]
+                if g is not None:
+                    g.replay()
+                else:
+                    mega_body(wa, wb, first)
+                    if graph_ok and k in (2, 3):
+                        gg = torch.cuda.CUDAGraph()
+                        with torch.cuda.graph(gg):
+                            mega_body(wa, wb, False)
+                        graphs[k % 2] = gg
+            elif graph is not None:
                 graph.replay()
             else:
                 self._spmv_overlapped(w, q)
