@@ -89,11 +89,18 @@ class CGSolverHIP:
                         optr, ocols, ovals = sell_from_csr(L.O_rowptr, L.O_colidx,
                                                            L.O_vals)
                         self.sellO = (up(optr), up(ocols), up(ovals))
-        # megafused pipelined iteration needs SELL everywhere + int32 cols
+        # megafused pipelined iteration needs SELL everywhere + int32 cols.
+        # Measured on MI355X: for wide rows (~80 nnz, Queen-shaped) the fused
+        # epilogue's 6 vector streams cost the SpMV more x-gather locality
+        # than the saved q round-trip is worth (815us fused vs 805us split);
+        # for narrow rows (7-pt Poisson) the q elimination is ~8% of the
+        # iteration's traffic.  Auto-enable below ~16 nnz/row.
         self.can_megafuse = (
             self.sell is not None and self.sell[1].dtype == torch.int32
             and (L.nnzO == 0 or (self.sellO is not None
                                  and self.sellO[1].dtype == torch.int32)))
+        self.megafuse_auto = (self.can_megafuse
+                              and L.nnzA / max(L.nowned, 1) <= 16.0)
         self.halo = HaloExchange(L.halo, L.nowned, self.device, comm)
         self.scal = ops.alloc_scalars(self.device)
         self.partials = ops.alloc_partials(self.device)
@@ -300,7 +307,8 @@ class CGSolverHIP:
 
     def solve_pipelined(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
                         res_atol: float = 0.0, res_rtol: float = 1e-9,
-                        check_every: int = 1, use_graph: bool = True) -> SolveResult:
+                        check_every: int = 1, use_graph: bool = True,
+                        megafuse: bool | None = None) -> SolveResult:
         """Pipelined (Ghysels-Vanroose) CG: ONE 2-double allreduce per
         iteration, overlapped with the halo + SpMV of q = A w
         (reference acgsolverhip_solve_pipelined, cghip.c:1187-1933).
@@ -322,7 +330,7 @@ class CGSolverHIP:
         S = ops
         L = self.local
         scal = self.scal
-        mega = self.can_megafuse
+        mega = self.megafuse_auto if megafuse is None else (megafuse and self.can_megafuse)
         r = self._vec(nghost=True)
         w = self._vec(nghost=True)
         z = self._vec()

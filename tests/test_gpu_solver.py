@@ -167,11 +167,11 @@ def test_pipelined_megafused_matches_fallback(problem):
     gpu = CGSolverHIP(S, device="cuda:0")
     assert gpu.can_megafuse
     x1 = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
-    gpu.can_megafuse = False
-    r1 = gpu.solve_pipelined(b, x1, maxits=300, res_rtol=1e-10, use_graph=False)
-    gpu.can_megafuse = True
+    r1 = gpu.solve_pipelined(b, x1, maxits=300, res_rtol=1e-10, use_graph=False,
+                             megafuse=False)
     x2 = torch.zeros_like(x1)
-    r2 = gpu.solve_pipelined(b, x2, maxits=300, res_rtol=1e-10, use_graph=False)
+    r2 = gpu.solve_pipelined(b, x2, maxits=300, res_rtol=1e-10, use_graph=False,
+                             megafuse=True)
     assert r1.converged and r2.converged
     assert r1.niterations == r2.niterations, (r1.niterations, r2.niterations)
     torch.testing.assert_close(x1[:S.nowned], x2[:S.nowned], rtol=1e-6, atol=1e-8)
