@@ -108,6 +108,7 @@ class CGSolverHIP:
 
         self.prof = EventProfiler(profile)
         self.comm_stream = torch.cuda.Stream(self.device)
+        self.allred_stream = torch.cuda.Stream(self.device)
         self.copy_stream = torch.cuda.Stream(self.device)
         self._rr_host = torch.zeros(1, dtype=torch.float64, pin_memory=True)
         self._ev_p = torch.cuda.Event()
@@ -432,17 +433,42 @@ class CGSolverHIP:
                 return True
             return False
 
+        ev_gd = torch.cuda.Event()
+        ev_ar = torch.cuda.Event()
         k = 0
         while k < maxits:
             first = (k == 0)
-            # ONE 2-double allreduce per iteration (gamma,delta adjacent)
-            with self.prof.span("allreduce"):
-                self._allreduce_slot(S.S_GAMMA, 2)
+            # ONE 2-double allreduce per iteration (gamma,delta adjacent).
+            # Multi-GPU non-megafused: the allreduce rides a dedicated side
+            # stream so SpMV(q = A w) -- which does not read the scalars --
+            # overlaps it; only the fused update waits (this is the point
+            # of Ghysels-Vanroose pipelining: reference cghip.c:1750-1811).
+            overlap_ar = (not serial) and not mega
+            if overlap_ar:
+                cur = torch.cuda.current_stream(self.device)
+                ev_gd.record(cur)
+                self.allred_stream.wait_event(ev_gd)
+                with torch.cuda.stream(self.allred_stream):
+                    with self.prof.span("allreduce", self.allred_stream):
+                        self._allreduce_slot(S.S_GAMMA, 2)
+                    ev_ar.record(self.allred_stream)
+            else:
+                with self.prof.span("allreduce"):
+                    self._allreduce_slot(S.S_GAMMA, 2)
             # lagged host test of the previous iteration's gamma (read
             # BEFORE issuing this iteration's copy: LAG+1 buffers rotate)
             if k >= LAG and check(k - LAG):
                 break
-            issue_gamma_copy(k)
+            if overlap_ar:
+                # D2H of gamma chains off the allreduce, not the main stream
+                j = k % (LAG + 1)
+                self.copy_stream.wait_event(ev_ar)
+                with torch.cuda.stream(self.copy_stream):
+                    hostbuf[j].copy_(scal[S.S_GAMMA:S.S_GAMMA + 1],
+                                     non_blocking=True)
+                    evdone[j].record(self.copy_stream)
+            else:
+                issue_gamma_copy(k)
             if mega:
                 wa, wb = (w, w2) if k % 2 == 0 else (w2, w)
                 g = graphs[k % 2]
@@ -459,6 +485,9 @@ class CGSolverHIP:
                 graph.replay()
             else:
                 self._spmv_overlapped(w, q)
+                if overlap_ar:
+                    # the fused update reads gamma/delta: wait the allreduce
+                    torch.cuda.current_stream(self.device).wait_event(ev_ar)
                 with self.prof.span("update"):
                     S.pipelined_fused(z, t, p, x, r, w, q, scal, self.partials,
                                       n, first)
