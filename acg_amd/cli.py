@@ -41,6 +41,8 @@ def make_parser() -> argparse.ArgumentParser:
                    help="index width of binary files (reference acgidx_t)")
     p.add_argument("--partition", metavar="FILE", default=None,
                    help="precomputed partition vector (mtx integer array)")
+    p.add_argument("--binary-partition", action="store_true",
+                   help="partition file is in binary Matrix Market format")
     p.add_argument("--partition-method", choices=("block", "rgb"), default="block")
     p.add_argument("--seed", type=int, default=0, help="partitioner seed")
     p.add_argument("--solver", default=None,
@@ -125,7 +127,9 @@ def main(argv=None) -> int:
                 f"({time.perf_counter() - t0:.2f}s)")
             A = SymCSRMatrix.from_mtxfile(m)
             if args.partition:
-                part = read_partition_file(args.partition, A.n)
+                part = read_partition_file(args.partition, A.n,
+                                           binary=args.binary_partition,
+                                           idxsize=args.idxsize)
                 if int(part.max()) >= nparts:
                     raise AcgError(ErrCode.INVALID_VALUE,
                                    f"partition file has {int(part.max()) + 1} parts, "

@@ -153,6 +153,16 @@ def read_mtx(path, gzipped: bool = False, binary: bool = False, idxsize: int = 6
         obj, fmt, fld, sym, nrows, ncols, nnz, comments = _read_header(f)
         m = MtxFile(obj, fmt, fld, sym, nrows, ncols, nnz, comments=comments)
         if binary:
+            if fmt == "array" and fld in ("real", "integer"):
+                # binary array: raw values after the size line (integer
+                # values stored as acgidx_t, real as float64)
+                vdt = np.float64 if fld == "real" else (np.int32 if idxsize == 32 else np.int64)
+                raw = f.read(nnz * np.dtype(vdt).itemsize)
+                if len(raw) < nnz * np.dtype(vdt).itemsize:
+                    raise AcgError(ErrCode.EOF, "binary array body truncated")
+                m.a = np.frombuffer(raw, dtype=vdt, count=nnz).astype(
+                    np.float64 if fld == "real" else np.int64)
+                return m
             if not (obj == "matrix" and fmt == "coordinate" and fld in ("real", "integer")):
                 raise AcgError(ErrCode.NOT_SUPPORTED, "binary supports matrix/coordinate real|integer")
             idt = np.int32 if idxsize == 32 else np.int64
@@ -242,13 +252,18 @@ def write_mtx(path_or_file, m: MtxFile, binary: bool = False, idxsize: int = 64,
                 _w(f"{m.nrows}\n")
             else:
                 _w(f"{m.nrows} {m.ncols}\n")
-            fmtv = numfmt.format if numfmt else (lambda v: repr(float(v)))
-            if m.field_ == "integer":
-                for v in m.a:
-                    _w(f"{int(v)}\n")
+            if binary:
+                vdt = np.float64 if m.field_ == "real" else \
+                    (np.int32 if idxsize == 32 else np.int64)
+                f.write(np.asarray(m.a, dtype=vdt).tobytes())
             else:
-                for v in m.a:
-                    _w(f"{fmtv(v)}\n")
+                fmtv = numfmt.format if numfmt else (lambda v: repr(float(v)))
+                if m.field_ == "integer":
+                    for v in m.a:
+                        _w(f"{int(v)}\n")
+                else:
+                    for v in m.a:
+                        _w(f"{fmtv(v)}\n")
         else:
             raise AcgError(ErrCode.NOT_SUPPORTED, f"{m.object}/{m.format}")
     finally:

@@ -121,15 +121,16 @@ def partition_rows(A, nparts: int, seed: int = 0, method: str = "auto") -> np.nd
     raise AcgError(ErrCode.NOT_SUPPORTED, f"partition method {method!r}")
 
 
-def read_partition_file(path, n: int | None = None) -> np.ndarray:
+def read_partition_file(path, n: int | None = None, binary: bool = False,
+                        gzipped: bool = False, idxsize: int = 64) -> np.ndarray:
     """Read a partition vector (mtx integer array, 1-based parts).
 
-    Reference: --partition handling at hip/acg-hip.c:1513-1641 and the
-    mtxpartition tool output format.
+    Reference: --partition / --binary-partition handling at
+    hip/acg-hip.c:1513-1641 and the mtxpartition tool output format.
     """
     from ..io.mtx import read_mtx
 
-    m = read_mtx(path)
+    m = read_mtx(path, binary=binary, gzipped=gzipped, idxsize=idxsize)
     part = np.asarray(m.a, dtype=np.int64)
     if n is not None and len(part) != n:
         raise AcgError(ErrCode.INVALID_VALUE,
@@ -137,7 +138,8 @@ def read_partition_file(path, n: int | None = None) -> np.ndarray:
     return (part - 1).astype(np.int32)
 
 
-def write_partition_file(path, part: np.ndarray) -> None:
+def write_partition_file(path, part: np.ndarray, binary: bool = False,
+                         idxsize: int = 64) -> None:
     """Write a 1-based partition vector as mtx integer array (mtxpartition)."""
     from ..io.mtx import MtxFile, write_mtx
 
@@ -145,4 +147,4 @@ def write_partition_file(path, part: np.ndarray) -> None:
     m = MtxFile(object="matrix", format="array", field_="integer",
                 symmetry="general", nrows=len(part), ncols=1, nnz=len(part),
                 a=part.astype(np.int64) + 1)
-    write_mtx(path, m)
+    write_mtx(path, m, binary=binary, idxsize=idxsize)

@@ -106,6 +106,21 @@ class LocalSystem:
     def nnzO(self) -> int:
         return int(self.O_rowptr[-1]) if len(self.O_rowptr) else 0
 
+    def dump(self, file=None) -> None:
+        """Debug dump of the subdomain structure (reference acggraph_fwrite,
+        graph.c:404 and acghalo_fwrite, halo.c:356)."""
+        import sys
+
+        f = file or sys.stderr
+        h = self.halo
+        print(f"LocalSystem(rank={self.rank}/{self.nparts}, n_global={self.n_global})",
+              file=f)
+        print(f"  rows: owned={self.nowned} (interior={self.ninterior}, "
+              f"border={self.nborder}), ghost={self.nghost}", file=f)
+        print(f"  matA: nnz={self.nnzA}; matO: nnz={self.nnzO}", file=f)
+        print(f"  halo: recv from {list(h.senders)} counts {list(h.recvcounts)}; "
+              f"send to {list(h.recipients)} counts {list(h.sendcounts)}", file=f)
+
 
 def _col_dtype(ncols: int):
     return np.int32 if ncols < 2**31 else np.int64

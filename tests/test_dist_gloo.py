@@ -100,6 +100,37 @@ def _body_cg_pipelined(comm):
     return (S.owned_global, x[:S.nowned].numpy(), res.niterations)
 
 
+def _body_cg_ws4(comm):
+    """4-rank pipelined CG on a generic (non-slab) partition via the full
+    extract_subdomains path — covers multi-neighbour halos."""
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.solvers.cpu import CGSolverCPU
+
+    A = stencil_global(6, 6, 6, STENCIL_27PT_3D)
+    part = partition_rows(A, comm.size, method="rgb", seed=1)
+    S = extract_subdomains(A, part, comm.size)[comm.rank]
+    rng = np.random.default_rng(7)
+    b_global = rng.standard_normal(A.n)
+    b = torch.from_numpy(b_global[S.owned_global])
+    solver = CGSolverCPU(S, comm=comm)
+    x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64)
+    res = solver.solve_pipelined(b, x, maxits=800, res_rtol=1e-10)
+    assert res.converged, res.summary()
+    return (S.owned_global, x[:S.nowned].numpy(), res.niterations)
+
+
+def test_distributed_cg_ws4_rgb():
+    results = _run_dist("_body_cg_ws4", world=4, port=29604)
+    A = stencil_global(6, 6, 6, STENCIL_27PT_3D)
+    rng = np.random.default_rng(7)
+    b_global = rng.standard_normal(A.n)
+    import scipy.sparse.linalg as spla
+
+    x_ref = spla.spsolve(A.to_scipy_full().tocsc(), b_global)
+    for rank, (owned_global, xloc, nit) in results.items():
+        np.testing.assert_allclose(xloc, x_ref[owned_global], rtol=1e-6, atol=1e-8)
+
+
 def test_halo_exchange_gloo_ws2():
     _run_dist("_body_halo", world=2, port=29601)
 
