@@ -76,12 +76,22 @@ class SymCSRMatrix:
         ``symmetric_input=True`` means the COO lists each off-diagonal pair
         once (standard MTX ``symmetric``); entries are canonicalised to the
         upper triangle (row <= col), duplicates summed.
+
+        Uses the native C++/OpenMP assembly (acg_amd.host) when built;
+        falls back to vectorised numpy.
         """
         i = np.asarray(rowidx, dtype=np.int64)
         j = np.asarray(colidx, dtype=np.int64)
         v = np.asarray(vals, dtype=np.float64)
         if i.shape != j.shape or i.shape != v.shape:
             raise AcgError(ErrCode.INVALID_VALUE, "COO array length mismatch")
+        try:
+            from ..host import _acg_host as H
+
+            rowptr, cols, vv = H.coo_to_sym_csr(n, i, j, v)
+            return cls(n, rowptr, cols, vv)
+        except ImportError:
+            pass
         # canonicalise to upper triangle
         lo = i > j
         iu = np.where(lo, j, i)
@@ -123,8 +133,19 @@ class SymCSRMatrix:
         """Expand the packed upper triangle to a full CSR operator.
 
         ``eps`` is added to every diagonal entry (reference --epsilon
-        diagonal shift, symcsrmatrix.c:760-845).
+        diagonal shift, symcsrmatrix.c:760-845).  Native C++/OpenMP
+        expansion (O(nnz), parallel) when built; numpy lexsort fallback.
         """
+        try:
+            from ..host import _acg_host as H
+
+            col32 = self.n < 2**31
+            rowptr, cols, vv = H.sym_expand_full(self.n, self.rowptr,
+                                                 self.colidx, self.vals,
+                                                 eps, col32)
+            return FullCSR(self.n, self.n, rowptr, cols, vv)
+        except ImportError:
+            pass
         rows_u = self._rows()
         cols_u = self.colidx
         vals_u = self.vals

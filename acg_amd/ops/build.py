@@ -16,6 +16,7 @@ from pathlib import Path
 
 HERE = Path(__file__).resolve().parent
 SRC = HERE / "kernels.hip"
+HOST_SRC = HERE.parent / "host" / "host.cpp"
 ARCH = os.environ.get("ACG_AMD_ARCH", "gfx950")
 
 
@@ -32,26 +33,35 @@ def needs_build() -> bool:
     return (not so.exists()) or so.stat().st_mtime < SRC.stat().st_mtime
 
 
-def build(verbose: bool = True, force: bool = False) -> Path:
-    """Compile kernels.hip -> _acg_kernels.so (gfx950)."""
-    so = so_path()
-    if not force and not needs_build():
-        return so
-    import pybind11
+def host_so_path() -> Path:
+    return HOST_SRC.parent / f"_acg_host{_ext_suffix()}"
 
-    cmd = [
-        "hipcc", f"--offload-arch={ARCH}", "-O3", "-std=c++17",
-        "-shared", "-fPIC",
-        f"-I{pybind11.get_include()}",
-        f"-I{sysconfig.get_paths()['include']}",
-        str(SRC), "-o", str(so),
-    ]
+
+def _run(cmd, verbose):
     if verbose:
         print("[acg_amd.ops.build]", " ".join(cmd), file=sys.stderr)
     r = subprocess.run(cmd, capture_output=True, text=True)
     if r.returncode != 0:
         raise RuntimeError(
-            f"hipcc failed ({r.returncode}):\n{r.stdout}\n{r.stderr}")
+            f"{cmd[0]} failed ({r.returncode}):\n{r.stdout}\n{r.stderr}")
+
+
+def build(verbose: bool = True, force: bool = False) -> Path:
+    """Compile kernels.hip -> _acg_kernels.so (gfx950) and the C++/OpenMP
+    host extension _acg_host.so."""
+    import pybind11
+
+    inc = [f"-I{pybind11.get_include()}", f"-I{sysconfig.get_paths()['include']}"]
+    so = so_path()
+    if force or needs_build():
+        _run(["hipcc", f"--offload-arch={ARCH}", "-O3", "-std=c++17",
+              "-shared", "-fPIC", *inc, str(SRC), "-o", str(so)], verbose)
+    hso = host_so_path()
+    if force or (not hso.exists()) or hso.stat().st_mtime < HOST_SRC.stat().st_mtime:
+        # -march=x86-64-v3 (not native): the built .so travels to GPU boxes
+        # with possibly different host CPUs
+        _run(["g++", "-O3", "-std=c++17", "-fopenmp", "-shared", "-fPIC",
+              "-march=x86-64-v3", *inc, str(HOST_SRC), "-o", str(hso)], verbose)
     return so
 
 
