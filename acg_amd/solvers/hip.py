@@ -217,8 +217,18 @@ class CGSolverHIP:
             res.tsolve = time.perf_counter() - t0
             return res
         converged = False
-        k = 0
-        while k < maxits:
+        serial = self.comm is None or self.comm.size == 1
+        graph = None
+        graph_ok = serial and not self.prof.enabled
+        # lag-1 convergence pipeline + hipGraph replay, mirroring
+        # solve_pipelined (the host test runs for every iteration; the host
+        # reads the value one iteration late so the GPU never stalls)
+        LAG = 1
+        hostbuf = [torch.zeros(1, dtype=torch.float64, pin_memory=True)
+                   for _ in range(LAG + 1)]
+        evdone = [torch.cuda.Event() for _ in range(LAG + 1)]
+
+        def body(first_capture=False):
             # zero the fused (p,t) accumulator, then halo+split SpMV with
             # the (p,t) reduction fused into both SpMV passes
             S.cg_prep_pt(scal)
@@ -234,14 +244,47 @@ class CGSolverHIP:
             # p = (rr/rr_prev) p + r
             with self.prof.span("daypx"):
                 S.daypx_ratio(p, r, scal, S.S_RR, S.S_RR_PREV, n=n)
+
+        def read_rr(j):
+            evdone[j % (LAG + 1)].synchronize()
+            return float(hostbuf[j % (LAG + 1)][0])
+
+        def check(j):
+            nonlocal rr, converged
+            rr = read_rr(j)
+            if not math.isfinite(rr):
+                raise FloatingPointError(f"CG diverged: rr={rr} at it {j + 1}")
+            if rtol2 > 0 and rr <= rtol2:
+                converged = True
+                res.niterations = j + 1
+                return True
+            return False
+
+        k = 0
+        while k < maxits:
+            if k >= LAG and check(k - LAG):
+                break
+            if graph is not None:
+                graph.replay()
+            else:
+                body()
+                if graph_ok and k == 1:
+                    graph = torch.cuda.CUDAGraph()
+                    with torch.cuda.graph(graph):
+                        body()
+            # issue the lagged rr D2H (after this iteration's update)
+            cur = torch.cuda.current_stream(self.device)
+            self._ev_rr.record(cur)
+            self.copy_stream.wait_event(self._ev_rr)
+            j = k % (LAG + 1)
+            with torch.cuda.stream(self.copy_stream):
+                hostbuf[j].copy_(scal[S.S_RR:S.S_RR + 1], non_blocking=True)
+                evdone[j].record(self.copy_stream)
             k += 1
             res.niterations = k
-            if k % check_every == 0 or k == maxits:
-                rr = self._host_scalar(S.S_RR)
-                if not math.isfinite(rr):
-                    raise FloatingPointError(f"CG diverged: rr={rr} at it {k}")
-                if rtol2 > 0 and rr <= rtol2:
-                    converged = True
+        if not converged:
+            for j in range(max(maxits - LAG, 0), maxits):
+                if check(j):
                     break
         torch.cuda.synchronize(self.device)
         res.tsolve = time.perf_counter() - t0
