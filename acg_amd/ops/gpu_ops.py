@@ -181,12 +181,13 @@ def pipelined_finalize(partials, nblocks: int, scal, first: bool) -> None:
 
 
 def pipelined_fused(z, t, p, x, r, w, q, scal: torch.Tensor,
-                    partials: torch.Tensor, n: int, first: bool) -> None:
+                    partials: torch.Tensor, n: int, first: bool,
+                    nt_update: bool = True) -> None:
     """Fused pipelined update + next gamma/delta + scalar rotation."""
     K.pipelined_fused(z.data_ptr(), t.data_ptr(), p.data_ptr(), x.data_ptr(),
                       r.data_ptr(), w.data_ptr(), q.data_ptr(), n,
                       scal.data_ptr(), 1 if first else 0, partials.data_ptr(),
-                      _stream())
+                      nt_update, _stream())
 
 
 def cg_device(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,

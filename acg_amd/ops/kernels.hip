@@ -326,6 +326,7 @@ __device__ __forceinline__ void pipelined_coeffs(const double* scal, int first,
 // delta' = (w',r') cost zero extra memory traffic.  The separate dot2 pass
 // of the reference (two hipblasDdot, cghip.c:1735-1752) disappears from
 // the iteration.
+template <bool NTU>
 __global__ void __launch_bounds__(BLOCK)
 k_pipelined_fused(double* __restrict__ z, double* __restrict__ t,
                   double* __restrict__ p, double* __restrict__ x,
@@ -337,18 +338,22 @@ k_pipelined_fused(double* __restrict__ z, double* __restrict__ t,
     pipelined_coeffs(scal, first, &beta, &alpha);
     double g = 0.0, d = 0.0;
     const long stride = (long)gridDim.x * BLOCK;
-    // z,t,p,x,q are touched once per iteration: non-temporal so their
-    // ~400 MB/iter of streams do not evict w (the next SpMV's gather
-    // source) or r from L2/L3.  w/r stay cached (measured: the in-loop
-    // SpMV runs ~8% slower than isolated purely from this eviction).
+    // NTU: z,t,p,x,q are touched once per iteration -- non-temporal so
+    // their ~400 MB/iter of streams do not evict w (the next SpMV's
+    // gather source) or r from L2/L3.
     for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
-        const double zi = ld_nt(q + i) + beta * ld_nt(z + i);
-        const double ti = w[i] + beta * ld_nt(t + i);
-        const double pi = r[i] + beta * ld_nt(p + i);
-        __builtin_nontemporal_store(zi, z + i);
-        __builtin_nontemporal_store(ti, t + i);
-        __builtin_nontemporal_store(pi, p + i);
-        __builtin_nontemporal_store(ld_nt(x + i) + alpha * pi, x + i);
+        const double zi = (NTU ? ld_nt(q + i) : q[i]) + beta * (NTU ? ld_nt(z + i) : z[i]);
+        const double ti = w[i] + beta * (NTU ? ld_nt(t + i) : t[i]);
+        const double pi = r[i] + beta * (NTU ? ld_nt(p + i) : p[i]);
+        const double xn = (NTU ? ld_nt(x + i) : x[i]) + alpha * pi;
+        if (NTU) {
+            __builtin_nontemporal_store(zi, z + i);
+            __builtin_nontemporal_store(ti, t + i);
+            __builtin_nontemporal_store(pi, p + i);
+            __builtin_nontemporal_store(xn, x + i);
+        } else {
+            z[i] = zi; t[i] = ti; p[i] = pi; x[i] = xn;
+        }
         const double rn = r[i] - alpha * ti;
         const double wn = w[i] - alpha * zi;
         r[i] = rn; w[i] = wn;
@@ -930,12 +935,18 @@ void pipelined_finalize(uintptr_t partials, int nblocks, uintptr_t scal, int fir
 
 void pipelined_fused(uintptr_t z, uintptr_t t, uintptr_t p, uintptr_t x, uintptr_t r,
                      uintptr_t w, uintptr_t q, long n, uintptr_t scal, int first,
-                     uintptr_t partials, uintptr_t stream) {
+                     uintptr_t partials, bool nt_update, uintptr_t stream) {
     long blocks = elem_grid(n);
-    hipLaunchKernelGGL(k_pipelined_fused, dim3((unsigned)blocks), dim3(BLOCK), 0, S(stream),
-                       (double*)z, (double*)t, (double*)p, (double*)x, (double*)r,
-                       (double*)w, (const double*)q, n, (const double*)scal, first,
-                       (double*)partials);
+    if (nt_update)
+        hipLaunchKernelGGL(k_pipelined_fused<true>, dim3((unsigned)blocks), dim3(BLOCK), 0, S(stream),
+                           (double*)z, (double*)t, (double*)p, (double*)x, (double*)r,
+                           (double*)w, (const double*)q, n, (const double*)scal, first,
+                           (double*)partials);
+    else
+        hipLaunchKernelGGL(k_pipelined_fused<false>, dim3((unsigned)blocks), dim3(BLOCK), 0, S(stream),
+                           (double*)z, (double*)t, (double*)p, (double*)x, (double*)r,
+                           (double*)w, (const double*)q, n, (const double*)scal, first,
+                           (double*)partials);
     check_hip("pipelined_fused");
     hipLaunchKernelGGL(k_pipelined_finalize, dim3(1), dim3(BLOCK), 0, S(stream),
                        (const double*)partials, (int)blocks, (double*)scal, first);
