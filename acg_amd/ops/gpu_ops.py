@@ -92,8 +92,10 @@ def spmv_sell(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
               rowbase: int = 0, accum: bool = False,
               partials: torch.Tensor | None = None,
               scal: torch.Tensor | None = None, dotslot: int = -1,
-              dot_accum: bool = True, variant: int | None = None) -> None:
-    """SELL-C-64 SpMV (regular-row fast path)."""
+              dot_accum: bool = True, variant: int | None = None,
+              perm: torch.Tensor | None = None) -> None:
+    """SELL-C-64(-sigma) SpMV.  ``perm`` (int32) maps SELL row -> matrix
+    row for sigma-sorted irregular matrices."""
     nslices = sellptr.numel() - 1
     if nslices <= 0:
         return
@@ -105,7 +107,7 @@ def spmv_sell(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
                 vals.data_ptr(), x.data_ptr(), y.data_ptr(), accum,
                 partials.data_ptr() if fuse else 0,
                 scal.data_ptr() if fuse else 0, dotslot, dot_accum,
-                variant, _stream())
+                variant, perm.data_ptr() if perm is not None else 0, _stream())
 
 
 def zero_scalars(scal: torch.Tensor, i0: int = 0, count: int | None = None) -> None:

@@ -143,7 +143,8 @@ spmv_csr_vector(long nrows, long rowbase,
 template <typename T>
 __device__ __forceinline__ T ld_nt(const T* p) { return __builtin_nontemporal_load(p); }
 
-template <typename ColT, bool ACCUM, bool FUSE_DOT, bool NT, bool SWZ, int UNROLL>
+template <typename ColT, bool ACCUM, bool FUSE_DOT, bool NT, bool SWZ, int UNROLL,
+          bool PERM = false>
 __global__ void __launch_bounds__(BLOCK)
 k_spmv_sell(long nslices, long nrows, long rowbase,
           const long* __restrict__ sellptr,   // [nslices+1], element offsets
@@ -151,7 +152,8 @@ k_spmv_sell(long nslices, long nrows, long rowbase,
           const double* __restrict__ vals,    // pad value = 0
           const double* __restrict__ x,
           double* __restrict__ y,
-          double* __restrict__ partials) {
+          double* __restrict__ partials,
+          const int* __restrict__ perm = nullptr) {  // SELL row -> matrix row (sigma-sorted)
     const int lane = threadIdx.x & (WAVE - 1);
     long blk = blockIdx.x;
     if (SWZ) {
@@ -187,7 +189,7 @@ k_spmv_sell(long nslices, long nrows, long rowbase,
             const ColT ci = NT ? ld_nt(c + j * WAVE) : c[j * WAVE];
             sum += a * x[ci];
         }
-        const long row = s * WAVE + lane;
+        const long row = PERM ? (long)perm[s * WAVE + lane] : s * WAVE + lane;
         if (row < nrows) {
             if (ACCUM) y[rowbase + row] += sum; else y[rowbase + row] = sum;
             if (FUSE_DOT) dacc += x[rowbase + row] * sum;
@@ -810,12 +812,31 @@ void spmv(long nrows, long rowbase, uintptr_t rowptr, uintptr_t colidx,
 void spmv_sell(long nslices, long nrows, long rowbase, uintptr_t sellptr,
                uintptr_t cols, int col64, uintptr_t vals, uintptr_t x,
                uintptr_t y, bool accum, uintptr_t partials, uintptr_t scal,
-               int dotslot, bool dot_accum, int variant, uintptr_t stream) {
+               int dotslot, bool dot_accum, int variant, uintptr_t perm,
+               uintptr_t stream) {
     if (nrows == 0) return;
     long blocks = (nslices * WAVE + BLOCK - 1) / BLOCK;
     if (blocks > MAXG) blocks = MAXG;
     const bool fuse = partials != 0 && dotslot >= 0;
     dim3 g((unsigned)blocks), b(BLOCK);
+    if (perm != 0) {
+        // sigma-sorted SELL (irregular rows): fixed NT, no swizzle, unroll 4
+        #define LAUNCH_PERM(CT, AC, FD) \
+            hipLaunchKernelGGL((k_spmv_sell<CT, AC, FD, true, false, 4, true>), \
+                g, b, 0, S(stream), nslices, nrows, rowbase, \
+                (const long*)sellptr, (const CT*)cols, (const double*)vals, \
+                (const double*)x, (double*)y, (double*)partials, (const int*)perm)
+        #define DISPP(CT) \
+            if (accum) { if (fuse) { LAUNCH_PERM(CT, true, true); } else { LAUNCH_PERM(CT, true, false); } } \
+            else       { if (fuse) { LAUNCH_PERM(CT, false, true); } else { LAUNCH_PERM(CT, false, false); } }
+        if (col64) { DISPP(long) } else { DISPP(int) }
+        #undef DISPP
+        #undef LAUNCH_PERM
+        check_hip("spmv_sell_perm");
+        if (fuse)
+            reduce_partials(partials, (int)blocks, scal, dotslot, dot_accum, stream);
+        return;
+    }
     #define LAUNCH_SELL(CT, AC, FD, NT, SWZ, U) \
         hipLaunchKernelGGL((k_spmv_sell<CT, AC, FD, NT, SWZ, U>), g, b, 0, S(stream), \
             nslices, nrows, rowbase, (const long*)sellptr, (const CT*)cols, \

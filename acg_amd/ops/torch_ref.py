@@ -46,13 +46,15 @@ def spmv(rowptr, colidx, vals, x, y, *, rowbase: int = 0, accum: bool = False,
         scal[dotslot] = scal[dotslot] + d if dot_accum else d
 
 
-def sell_from_csr(rowptr, colidx, vals, C: int = 64):
-    """Convert CSR -> SELL-C (vectorized numpy host prep).
+def sell_from_csr(rowptr, colidx, vals, C: int = 64, sigma: int = 1):
+    """Convert CSR -> SELL-C-sigma (vectorized numpy host prep).
 
-    Element j of row (s*C+lane) lives at sellptr[s] + j*C + lane.  Padding
-    entries point at the row itself with value 0 (in-bounds, cache-local
-    gather).  Returns numpy arrays (sellptr int64, cols like colidx, vals
-    f64)."""
+    Element j of SELL row (s*C+lane) lives at sellptr[s] + j*C + lane.
+    ``sigma`` > 1 sorts rows by descending length within windows of
+    sigma*C rows, shrinking padding for irregular matrices; the returned
+    ``perm`` (or None for sigma=1) maps SELL row -> matrix row (pad lanes
+    hold nrows as a sentinel).  Padding entries point at a valid row with
+    value 0.  Returns (sellptr int64, cols, vals f64[, perm int32])."""
     import numpy as np
 
     rowptr = np.asarray(rowptr)
@@ -61,24 +63,48 @@ def sell_from_csr(rowptr, colidx, vals, C: int = 64):
     nrows = len(rowptr) - 1
     nslices = (nrows + C - 1) // C
     counts = np.diff(rowptr)
+    perm = None
+    rowof = np.arange(nslices * C, dtype=np.int64)  # SELL row -> matrix row
+    if sigma > 1 and nrows:
+        win = sigma * C
+        order = np.empty(nrows, dtype=np.int64)
+        for w0 in range(0, nrows, win):
+            w1 = min(w0 + win, nrows)
+            sub = np.argsort(-counts[w0:w1], kind="stable")
+            order[w0:w1] = w0 + sub
+        rowof = np.full(nslices * C, nrows, dtype=np.int64)
+        rowof[:nrows] = order
+        perm = rowof.astype(np.int32)
     cpad = np.zeros(nslices * C, dtype=np.int64)
-    cpad[:nrows] = counts
+    valid = rowof < nrows
+    cpad[valid] = counts[rowof[valid]]
     slice_len = cpad.reshape(nslices, C).max(axis=1)
     sellptr = np.zeros(nslices + 1, dtype=np.int64)
     np.cumsum(slice_len * C, out=sellptr[1:])
     total = int(sellptr[-1])
-    # defaults: col = own row (clipped), val = 0
+    # defaults: col = first valid row's own col target (clipped), val = 0
     slice_of_p = np.repeat(np.arange(nslices, dtype=np.int64), slice_len * C)
     lane = (np.arange(total, dtype=np.int64) - sellptr[slice_of_p]) % C
-    cols = np.minimum(slice_of_p * C + lane, nrows - 1).astype(colidx.dtype)
+    defrow = np.minimum(rowof[np.minimum(slice_of_p * C + lane,
+                                         nslices * C - 1)], nrows - 1)
+    cols = defrow.astype(colidx.dtype)
     svals = np.zeros(total, dtype=np.float64)
-    # scatter real entries
+    # scatter real entries: matrix row r sits at SELL row invperm[r]
     nnz = len(colidx)
-    rows = np.repeat(np.arange(nrows, dtype=np.int64), counts)
-    within = np.arange(nnz, dtype=np.int64) - rowptr[rows]
+    if perm is not None:
+        invperm = np.empty(nslices * C, dtype=np.int64)
+        invperm[rowof] = np.arange(nslices * C, dtype=np.int64)
+        sellrow = invperm[:nrows]
+    else:
+        sellrow = np.arange(nrows, dtype=np.int64)
+    rows = np.repeat(sellrow, counts)
+    within = np.arange(nnz, dtype=np.int64) - rowptr[np.repeat(
+        np.arange(nrows, dtype=np.int64), counts)]
     dst = sellptr[rows // C] + within * C + rows % C
     cols[dst] = colidx
     svals[dst] = vals
+    if perm is not None:
+        return sellptr, cols, svals, perm
     return sellptr, cols, svals
 
 

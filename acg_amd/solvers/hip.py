@@ -57,6 +57,7 @@ class CGSolverHIP:
         self.nlocal = L.nowned + L.nghost
         self.sell = None
         self.sellO = None
+        self.sell_perm = None
         self.A_rowptr = self.A_colidx = self.A_vals = None
         self.O_rowptr = self.O_colidx = self.O_vals = None
         self.lanesA = self.lanesO = lanes or 16
@@ -80,9 +81,18 @@ class CGSolverHIP:
             if use_sell and L.nowned > 0:
                 from ..ops.torch_ref import sell_from_csr
 
+                self.sell_perm = None
                 sellptr, scols, svals = sell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals)
                 waste = (int(sellptr[-1]) - L.nnzA) / max(L.nnzA, 1)
-                if waste <= 0.3:
+                if waste > 0.3:
+                    # irregular rows: sigma-sort within 16-slice windows
+                    out = sell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals,
+                                        sigma=16)
+                    sellptr, scols, svals, perm = out
+                    waste = (int(sellptr[-1]) - L.nnzA) / max(L.nnzA, 1)
+                    if waste <= 0.5:
+                        self.sell_perm = up(perm)
+                if waste <= 0.5:
                     self.sell = (up(sellptr), up(scols), up(svals))
                     self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR
                     if L.nnzO > 0:
@@ -96,7 +106,8 @@ class CGSolverHIP:
         # for narrow rows (7-pt Poisson) the q elimination is ~8% of the
         # iteration's traffic.  Auto-enable below ~16 nnz/row.
         self.can_megafuse = (
-            self.sell is not None and self.sell[1].dtype == torch.int32
+            self.sell is not None and self.sell_perm is None
+            and self.sell[1].dtype == torch.int32
             and (L.nnzO == 0 or (self.sellO is not None
                                  and self.sellO[1].dtype == torch.int32)))
         self.megafuse_auto = (self.can_megafuse
@@ -142,7 +153,7 @@ class CGSolverHIP:
             if self.sell is not None:
                 sellptr, scols, svals = self.sell
                 ops.spmv_sell(sellptr, scols, svals, self.n, xfull, y,
-                              accum=False, **fuse)
+                              accum=False, perm=self.sell_perm, **fuse)
             else:
                 ops.spmv(self.A_rowptr, self.A_colidx, self.A_vals, xfull, y,
                          lanes=self.lanesA, accum=False, **fuse)
