@@ -47,7 +47,8 @@ def test_split_spmv_multirank_emulated(nranks, gen):
                     dotslot=gpu_ops.S_PT, dot_accum=True)
         gpu_ops.zero_scalars(solver.scal, gpu_ops.S_PT, 1)
         sp, sc, sv = solver.sell
-        gpu_ops.spmv_sell(sp, sc, sv, S.nowned, xl, y, **fuse)
+        gpu_ops.spmv_sell(sp, sc, sv, S.nowned, xl, y,
+                          perm=solver.sell_perm, **fuse)
         if S.nnzO:
             if solver.sellO is not None:
                 op_, oc, ov = solver.sellO
