@@ -96,7 +96,8 @@ def device_stencil_slab(gx: int, gy: int, gz: int, spec: dict,
         K.stencil_rowlen(nodes, row0_node, gx, gy, gz, dof, nown_nodes,
                          zs_own.data_ptr(), pb.data_ptr(), offs.data_ptr(),
                          ksten, filter_ghost, rowlen_nodes.data_ptr(), stream)
-        rowlen = rowlen_nodes.repeat_interleave(dof)
+        # avoid an extra multi-GB copy for dof=1 (the 8.6e9-row Poisson)
+        rowlen = rowlen_nodes if dof == 1 else rowlen_nodes.repeat_interleave(dof)
         nnz = int(rowlen.sum())
         nrows = nodes * dof
         nslices = (nrows + 63) // 64

@@ -123,7 +123,9 @@ class LocalSystem:
 
 
 def _col_dtype(ncols: int):
-    return np.int32 if ncols < 2**31 else np.int64
+    from ..utils.config import col_dtype
+
+    return col_dtype(ncols)
 
 
 def _sort_rows_cols(rowptr, colidx, vals, nrows):

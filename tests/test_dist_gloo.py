@@ -131,6 +131,29 @@ def test_distributed_cg_ws4_rgb():
         np.testing.assert_allclose(xloc, x_ref[owned_global], rtol=1e-6, atol=1e-8)
 
 
+def _body_collective_error(comm):
+    """Reference acgerrmpi semantics (error.c:149): one rank's failure must
+    raise on ALL ranks instead of deadlocking."""
+    from acg_amd.utils.errors import AcgError, ErrCode, collective_raise
+
+    exc = None
+    if comm.rank == 1:
+        exc = AcgError(ErrCode.INVALID_VALUE, "rank 1 exploded")
+    try:
+        collective_raise(comm, exc)
+    except AcgError as e:
+        return ("raised", int(e.code))
+    return ("no-raise", 0)
+
+
+def test_collective_error_agreement():
+    results = _run_dist("_body_collective_error", world=2, port=29605)
+    for rank, (status, code) in results.items():
+        assert status == "raised", f"rank {rank} did not raise"
+        assert code == int(__import__("acg_amd.utils.errors",
+                                      fromlist=["ErrCode"]).ErrCode.INVALID_VALUE)
+
+
 def test_halo_exchange_gloo_ws2():
     _run_dist("_body_halo", world=2, port=29601)
 
