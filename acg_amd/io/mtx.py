@@ -208,6 +208,32 @@ def read_mtx(path, gzipped: bool = False, binary: bool = False, idxsize: int = 6
         return m
 
 
+def _savetxt(f, arr, fmt: str, bytes_mode: bool, chunk: int = 1 << 20) -> None:
+    """Fast text writer: numpy savetxt in chunks (avoids per-row python)."""
+    import io as _io2
+
+    for lo in range(0, len(arr), chunk):
+        buf = _io2.StringIO()
+        np.savetxt(buf, arr[lo:lo + chunk], fmt=fmt)
+        s = buf.getvalue()
+        f.write(s.encode() if bytes_mode else s)
+
+
+def _savetxt_mixed(f, i, j, v, vfmt: str, bytes_mode: bool) -> None:
+    """int int float rows (coordinate real) via chunked savetxt."""
+    import io as _io2
+
+    chunk = 1 << 20
+    for lo in range(0, len(i), chunk):
+        buf = _io2.StringIO()
+        np.savetxt(buf, np.column_stack([i[lo:lo + chunk].astype(np.float64),
+                                         j[lo:lo + chunk].astype(np.float64),
+                                         v[lo:lo + chunk]]),
+                   fmt=f"%d %d {vfmt}")
+        s = buf.getvalue()
+        f.write(s.encode() if bytes_mode else s)
+
+
 def write_mtx(path_or_file, m: MtxFile, binary: bool = False, idxsize: int = 64,
               numfmt=None, gzipped: bool = False) -> None:
     """Write a Matrix Market file (reference mtxfile_fwrite_double, mtx2bin).
@@ -237,16 +263,20 @@ def write_mtx(path_or_file, m: MtxFile, binary: bool = False, idxsize: int = 64,
                 f.write((m.colidx + 1).astype(idt).tobytes())
                 f.write(np.asarray(m.a, dtype=np.float64 if m.field_ == "real" else np.int64).tobytes())
             else:
-                fmtv = numfmt.format if numfmt else (lambda v: repr(float(v)))
+                vfmt = numfmt._py if numfmt else "%.17g"
                 if m.field_ == "pattern":
-                    for i, j in zip(m.rowidx, m.colidx):
-                        _w(f"{i + 1} {j + 1}\n")
+                    body = np.column_stack([m.rowidx + 1, m.colidx + 1])
+                    _savetxt(f, body, "%d %d", binary or gzipped)
                 elif m.field_ == "integer":
+                    body = np.column_stack([m.rowidx + 1, m.colidx + 1,
+                                            np.asarray(m.a, dtype=np.int64)])
+                    _savetxt(f, body, "%d %d %d", binary or gzipped)
+                elif numfmt is not None and numfmt._hex:
                     for i, j, v in zip(m.rowidx, m.colidx, m.a):
-                        _w(f"{i + 1} {j + 1} {int(v)}\n")
+                        _w(f"{i + 1} {j + 1} {numfmt.format(v)}\n")
                 else:
-                    for i, j, v in zip(m.rowidx, m.colidx, m.a):
-                        _w(f"{i + 1} {j + 1} {fmtv(v)}\n")
+                    _savetxt_mixed(f, m.rowidx + 1, m.colidx + 1, m.a, vfmt,
+                                   binary or gzipped)
         elif m.format == "array":
             if m.object == "vector":
                 _w(f"{m.nrows}\n")
@@ -257,13 +287,16 @@ def write_mtx(path_or_file, m: MtxFile, binary: bool = False, idxsize: int = 64,
                     (np.int32 if idxsize == 32 else np.int64)
                 f.write(np.asarray(m.a, dtype=vdt).tobytes())
             else:
-                fmtv = numfmt.format if numfmt else (lambda v: repr(float(v)))
+                vfmt = numfmt._py if numfmt else "%.17g"
                 if m.field_ == "integer":
+                    _savetxt(f, np.asarray(m.a, dtype=np.int64), "%d",
+                             binary or gzipped)
+                elif numfmt is not None and numfmt._hex:
                     for v in m.a:
-                        _w(f"{int(v)}\n")
+                        _w(f"{numfmt.format(v)}\n")
                 else:
-                    for v in m.a:
-                        _w(f"{fmtv(v)}\n")
+                    _savetxt(f, np.asarray(m.a, dtype=np.float64), vfmt,
+                             binary or gzipped)
         else:
             raise AcgError(ErrCode.NOT_SUPPORTED, f"{m.object}/{m.format}")
     finally:

@@ -30,8 +30,9 @@ def main() -> int:
     ap.add_argument("--steps", type=int, default=200)
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--solver", choices=["pipelined", "classic"], default="pipelined")
-    ap.add_argument("--config", choices=["queen", "poisson7"], default="queen",
-                    help="queen: 27-pt dof-3 Queen_4147-shaped (BASELINE configs 2-4); "
+    ap.add_argument("--config", choices=["queen", "flan", "poisson7"], default="queen",
+                    help="queen: 27-pt dof-3 Queen_4147-shaped (BASELINE configs 3-4); "
+                         "flan: Flan_1565-shaped, 1.56M rows (BASELINE config 2); "
                          "poisson7: 7-pt 3D Poisson (BASELINE config 5 sizing)")
     ap.add_argument("--grid", type=int, default=None,
                     help="grid edge G (queen default 111 -> 4.10M rows; "
@@ -63,6 +64,13 @@ def main() -> int:
         G = args.grid or 111
         spec = queen_like_spec(dof)
         model = f"queen4147-like-27pt-dof{dof}-G{G}"
+    elif args.config == "flan":
+        # Flan_1565: 1.56M rows / 117M nnz (SuiteSparse shell problem,
+        # 3 dof/node); G=80 -> 1.536M rows / 121M nnz
+        dof = args.dof or 3
+        G = args.grid or 80
+        spec = queen_like_spec(dof)
+        model = f"flan1565-like-27pt-dof{dof}-G{G}"
     else:
         dof = args.dof or 1
         G = args.grid or 512
@@ -121,9 +129,10 @@ def main() -> int:
     ms_per_step = 1000.0 * elapsed / args.steps
     if rank == 0:
         out = {
-            "metric": ("CG iter/s (whole node), Queen_4147-shaped fp64"
-                       if args.config == "queen" else
-                       "CG iter/s (whole node), 7-pt 3D Poisson fp64"),
+            "metric": {"queen": "CG iter/s (whole node), Queen_4147-shaped fp64",
+                       "flan": "CG iter/s (whole node), Flan_1565-shaped fp64",
+                       "poisson7": "CG iter/s (whole node), 7-pt 3D Poisson fp64",
+                       }[args.config],
             "value": iters_per_s,
             "unit": "iter/s",
             "n_gpus": ngpus,
@@ -134,9 +143,10 @@ def main() -> int:
             "scaling": "strong",
             "vs_baseline": None,
             "dtype": "fp64",
-            "data": ("synthetic SPD (27-pt dof-3 stencil, Queen_4147 shape; random RHS)"
-                     if args.config == "queen" else
-                     "synthetic SPD (7-pt 3D Poisson; random RHS)"),
+            "data": {"queen": "synthetic SPD (27-pt dof-3 stencil, Queen_4147 shape; random RHS)",
+                     "flan": "synthetic SPD (27-pt dof-3 stencil, Flan_1565 shape; random RHS)",
+                     "poisson7": "synthetic SPD (7-pt 3D Poisson; random RHS)",
+                     }[args.config],
             "config": {
                 "model": model,
                 "rows": nrows_global,

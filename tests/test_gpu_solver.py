@@ -191,6 +191,45 @@ def test_device_generated_solver():
     assert res.converged, res.summary()
 
 
+@pytest.mark.parametrize("maxits", [1, 2, 3, 5])
+def test_pipelined_small_maxits(problem, maxits):
+    """Graph/lag machinery must be exact at tiny iteration counts."""
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A, S = problem
+    rng = np.random.default_rng(13)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+    gpu = CGSolverHIP(S, device="cuda:0")
+    x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
+    res = gpu.solve_pipelined(b, x, maxits=maxits, res_rtol=0.0)
+    assert res.niterations == maxits
+    res2 = gpu.solve(b, torch.zeros_like(x), maxits=maxits, res_rtol=0.0)
+    assert res2.niterations == maxits
+
+
+def test_convergence_iteration_counts_match_cpu(problem):
+    """Detected convergence iteration must equal the CPU oracle's for
+    classic and pipelined, graphs on and off."""
+    from acg_amd.solvers.cpu import CGSolverCPU
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A, S = problem
+    rng = np.random.default_rng(17)
+    b_np = rng.standard_normal(S.nowned)
+    cpu = CGSolverCPU(S)
+    xc = torch.zeros(S.nowned + S.nghost, dtype=torch.float64)
+    rc = cpu.solve(torch.from_numpy(b_np.copy()), xc, maxits=500, res_rtol=1e-8)
+    gpu = CGSolverHIP(S, device="cuda:0")
+    b = torch.from_numpy(b_np).cuda()
+    for name, fn in (("classic", gpu.solve),
+                     ("pipelined", gpu.solve_pipelined)):
+        x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
+        rg = fn(b, x, maxits=500, res_rtol=1e-8)
+        assert rg.converged
+        assert abs(rg.niterations - rc.niterations) <= 2, \
+            (name, rg.niterations, rc.niterations)
+
+
 def test_slab_generated_gpu_solve():
     """Flagship path: slab-generated Queen-like system, single GPU."""
     from acg_amd.gen import queen_like_spec, stencil_local_slab
