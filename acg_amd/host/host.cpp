@@ -248,6 +248,21 @@ py::tuple coo_to_sym_csr(i64 n,
     return py::make_tuple(rowptr, colidx, v);
 }
 
+// in-place scans (reference acgprefixsum_inplace_*, prefixsum.h:72-116)
+py::array_t<i64> prefix_sum(py::array_t<i64, py::array::c_style | py::array::forcecast> a,
+                            bool inclusive) {
+    i64 n = (i64)a.shape(0);
+    py::array_t<i64> out(n);
+    const i64* src = a.data();
+    i64* dst = out.mutable_data();
+    i64 run = 0;
+    for (i64 i = 0; i < n; ++i) {
+        if (inclusive) { run += src[i]; dst[i] = run; }
+        else { dst[i] = run; run += src[i]; }
+    }
+    return out;
+}
+
 int num_threads() {
 #ifdef _OPENMP
     return omp_get_max_threads();
@@ -263,5 +278,6 @@ PYBIND11_MODULE(_acg_host, m) {
           py::arg("n"), py::arg("rowptr_u"), py::arg("col_u"), py::arg("val_u"),
           py::arg("eps") = 0.0, py::arg("col32") = true);
     m.def("coo_to_sym_csr", &coo_to_sym_csr);
+    m.def("prefix_sum", &prefix_sum, py::arg("a"), py::arg("inclusive") = true);
     m.def("num_threads", &num_threads);
 }

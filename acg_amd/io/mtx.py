@@ -181,7 +181,7 @@ def read_mtx(path, gzipped: bool = False, binary: bool = False, idxsize: int = 6
             return m
         body = f.read()
         if obj == "matrix" and fmt == "coordinate":
-            want = 2 if fld == "pattern" else 3
+            want = {"pattern": 2, "complex": 4}.get(fld, 3)
             arr = _parse_body_text(body, want)
             if arr.shape[0] < nnz:
                 raise AcgError(ErrCode.EOF, f"expected {nnz} entries, got {arr.shape[0]}")
@@ -192,6 +192,8 @@ def read_mtx(path, gzipped: bool = False, binary: bool = False, idxsize: int = 6
                 m.a = np.ones(nnz, dtype=np.float64)
             elif fld == "integer":
                 m.a = arr[:, 2].astype(np.int64)
+            elif fld == "complex":
+                m.a = arr[:, 2] + 1j * arr[:, 3]
             else:
                 m.a = np.ascontiguousarray(arr[:, 2])
         elif fmt == "array":

@@ -141,9 +141,12 @@ class CGSolverHIP:
         L = self.local
         have_halo = self.comm is not None and self.comm.size > 1
         cur = torch.cuda.current_stream(self.device)
+        halo_span = None
         if have_halo:
             self._ev_p.record(cur)
             self.comm_stream.wait_event(self._ev_p)
+            halo_span = self.prof.span("halo", self.comm_stream)
+            halo_span.__enter__()  # closed after halo.end() below
             with torch.cuda.stream(self.comm_stream):
                 self.halo.begin(xfull)
         fuse = dict(partials=self.partials,
@@ -161,6 +164,8 @@ class CGSolverHIP:
             with torch.cuda.stream(self.comm_stream):
                 self.halo.end()
                 self._ev_recv.record(self.comm_stream)
+            if halo_span is not None:
+                halo_span.__exit__(None, None, None)
             cur.wait_event(self._ev_recv)
         if L.nborder > 0 and self.local.nnzO > 0:
             with self.prof.span("spmvO"):
