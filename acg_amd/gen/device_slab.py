@@ -119,6 +119,29 @@ def device_stencil_slab(gx: int, gy: int, gz: int, spec: dict,
     A_sellptr, A_cols, A_vals, nnzA = build(0, nown_nodes, 0)
     O_sellptr, O_cols, O_vals, nnzO = build(1, nborder_nodes, ninterior_nodes)
 
+    # Block-SELL for matA when the operator has dense dof x dof blocks:
+    # one int32 index per block instead of per entry (4 -> 4/dof^2 B/nnz).
+    A_bsell = None
+    if 2 <= dof <= 4 and nown_nodes > 0:
+        blocklen = torch.empty(nown_nodes, dtype=torch.int64, device=device)
+        K.stencil_blocklen(nown_nodes, gx, gy, gz, nown_nodes,
+                           zs_own.data_ptr(), pb.data_ptr(), offs.data_ptr(),
+                           ksten, blocklen.data_ptr(), stream)
+        nbslices = (nown_nodes + 63) // 64
+        bp = torch.zeros(nbslices * 64, dtype=torch.int64, device=device)
+        bp[:nown_nodes] = blocklen
+        blk_slice = bp.view(nbslices, 64).max(dim=1).values
+        bptr = torch.zeros(nbslices + 1, dtype=torch.int64, device=device)
+        torch.cumsum(blk_slice * 64, dim=0, out=bptr[1:])
+        btotal = int(bptr[-1])
+        bcol = torch.zeros(btotal, dtype=torch.int32, device=device)
+        bvals = torch.zeros(btotal * dof * dof, dtype=torch.float64, device=device)
+        K.stencil_bfill(nown_nodes, gx, gy, gz, dof, nown_nodes,
+                        zs_own.data_ptr(), pb.data_ptr(), offs.data_ptr(),
+                        ksten, blocks.data_ptr(), bptr.data_ptr(),
+                        bcol.data_ptr(), bvals.data_ptr(), stream)
+        A_bsell = (bptr, bcol, bvals, dof)
+
     # halo pattern (analytic, host-side; identical to stencil_local_slab)
     senders, recvcounts, rdispls = [], [], []
     recipients, sendidx_parts = [], []
@@ -165,6 +188,7 @@ def device_stencil_slab(gx: int, gy: int, gz: int, spec: dict,
         nowned=nowned, ninterior=ninterior_nodes * dof,
         nborder=nborder_nodes * dof, nghost=nghost,
         A_sell=(A_sellptr, A_cols, A_vals),
+        A_bsell=A_bsell,
         O_sell=(O_sellptr, O_cols, O_vals),
         _nnzA=nnzA, _nnzO=nnzO, halo=halo, device=device,
     )

@@ -110,6 +110,22 @@ def spmv_sell(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
                 variant, perm.data_ptr() if perm is not None else 0, _stream())
 
 
+def spmv_bsell(bptr: torch.Tensor, bcol: torch.Tensor, bvals: torch.Tensor,
+               nnodes: int, dof: int, x: torch.Tensor, y: torch.Tensor, *,
+               partials: torch.Tensor | None = None,
+               scal: torch.Tensor | None = None, dotslot: int = -1,
+               dot_accum: bool = True) -> None:
+    """Block-SELL SpMV (dense dof x dof blocks; one int32 index per block)."""
+    nslices = bptr.numel() - 1
+    if nslices <= 0:
+        return
+    fuse = scal is not None and dotslot >= 0
+    K.spmv_bsell(nslices, nnodes, dof, bptr.data_ptr(), bcol.data_ptr(),
+                 bvals.data_ptr(), x.data_ptr(), y.data_ptr(),
+                 partials.data_ptr() if fuse else 0,
+                 scal.data_ptr() if fuse else 0, dotslot, dot_accum, _stream())
+
+
 def zero_scalars(scal: torch.Tensor, i0: int = 0, count: int | None = None) -> None:
     K.zero_scalars(scal.data_ptr(), i0, count if count is not None else scal.numel() - i0,
                    _stream())

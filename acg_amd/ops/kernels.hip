@@ -202,6 +202,144 @@ k_spmv_sell(long nslices, long nrows, long rowbase,
 }
 
 // ---------------------------------------------------------------------------
+// Block-SELL (BSELL): SELL over BLOCK-rows for matrices with a dense
+// dof x dof block structure (FEM/structural problems like Queen_4147:
+// 3 dof per mesh node).  One lane = one block-row (node); per block only
+// ONE int32 block-column index covers dof*dof values, cutting index
+// traffic from 4 B/nnz to 4/dof^2 B/nnz (Queen dof=3: 12 -> 8.44 B/nnz
+// total, ~30% less SpMV traffic -- the iteration is at the HBM roofline,
+// so bytes are the only lever left).
+//
+// Layout per 64-node slice s (bptr in block units):
+//   bcol[bptr[s] + j*64 + lane]                 block-col of node's j-th block
+//   bvals[bptr[s]*dof^2 + (j*dof^2 + k)*64 + lane]   value k (row-major in
+//                                               the block), all 64 lanes
+//                                               contiguous per (j,k)
+__device__ __forceinline__ long stencil_col_node(
+    int xi, int yi, int zi, int dx, int dy, int dz, int gx, int gy, int gz,
+    const long* __restrict__ pb);  // defined with the stencil generators below
+
+template <int DOF, bool FUSE_DOT>
+__global__ void __launch_bounds__(BLOCK)
+k_spmv_bsell(long nslices, long nnodes,
+             const long* __restrict__ bptr,
+             const int* __restrict__ bcol,
+             const double* __restrict__ bvals,
+             const double* __restrict__ x,
+             double* __restrict__ y,
+             double* __restrict__ partials) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const long wslice = ((long)blockIdx.x * BLOCK + threadIdx.x) >> 6;
+    const long nw = ((long)gridDim.x * BLOCK) >> 6;
+    double dacc = 0.0;
+    for (long s = wslice; s < nslices; s += nw) {
+        const long b0 = bptr[s];
+        const long blen = (bptr[s + 1] - b0) >> 6;  // blocks per node
+        const int* __restrict__ c = bcol + b0 + lane;
+        const double* __restrict__ v = bvals + b0 * (DOF * DOF) + lane;
+        double acc[DOF];
+        #pragma unroll
+        for (int r = 0; r < DOF; ++r) acc[r] = 0.0;
+        for (long j = 0; j < blen; ++j) {
+            const int cb = ld_nt(c + j * WAVE);
+            double xv[DOF];
+            #pragma unroll
+            for (int cc = 0; cc < DOF; ++cc) xv[cc] = x[(long)cb * DOF + cc];
+            #pragma unroll
+            for (int k = 0; k < DOF * DOF; ++k) {
+                const double a = ld_nt(v + (j * DOF * DOF + k) * WAVE);
+                acc[k / DOF] += a * xv[k % DOF];
+            }
+        }
+        const long node = s * WAVE + lane;
+        if (node < nnodes) {
+            #pragma unroll
+            for (int r = 0; r < DOF; ++r) {
+                y[node * DOF + r] = acc[r];
+                if (FUSE_DOT) dacc += x[node * DOF + r] * acc[r];
+            }
+        }
+    }
+    if (FUSE_DOT) {
+        dacc = block_reduce(dacc);
+        if (threadIdx.x == 0) partials[blockIdx.x] = dacc;
+    }
+}
+
+// device-side BSELL generation for the block-stencil slab (block-level
+// analog of k_stencil_rowlen / k_stencil_fill; matA only -- owned x owned)
+__global__ void __launch_bounds__(BLOCK)
+k_stencil_blocklen(long nnodes, int gx, int gy, int gz, long nown_nodes,
+                   const int* __restrict__ zs_of_plane,
+                   const long* __restrict__ pb,
+                   const double* __restrict__ offs, int ksten,
+                   long* __restrict__ blocklen) {
+    const long stride = (long)gridDim.x * BLOCK;
+    const long plane_nodes = (long)gx * gy;
+    for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nnodes; i += stride) {
+        const int pl = (int)(i / plane_nodes);
+        const long rem = i - (long)pl * plane_nodes;
+        const int xi = (int)(rem % gx), yi = (int)(rem / gx);
+        const int zi = zs_of_plane[pl];
+        long cnt = 1;  // self block
+        for (int o = 0; o < ksten; ++o) {
+            const long cn = stencil_col_node(xi, yi, zi, (int)offs[o * 4],
+                                             (int)offs[o * 4 + 1],
+                                             (int)offs[o * 4 + 2], gx, gy, gz, pb);
+            if (cn >= 0 && cn < nown_nodes) ++cnt;
+        }
+        blocklen[i] = cnt;
+    }
+}
+
+template <int DOF>
+__global__ void __launch_bounds__(BLOCK)
+k_stencil_bfill(long nnodes, int gx, int gy, int gz, long nown_nodes,
+                const int* __restrict__ zs_of_plane,
+                const long* __restrict__ pb,
+                const double* __restrict__ offs, int ksten,
+                const double* __restrict__ blocks,  // M then D
+                const long* __restrict__ bptr,
+                int* __restrict__ bcol, double* __restrict__ bvals) {
+    const long stride = (long)gridDim.x * BLOCK;
+    const long plane_nodes = (long)gx * gy;
+    const double* M = blocks;
+    const double* D = blocks + DOF * DOF;
+    for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nnodes; i += stride) {
+        const int pl = (int)(i / plane_nodes);
+        const long rem = i - (long)pl * plane_nodes;
+        const int xi = (int)(rem % gx), yi = (int)(rem / gx);
+        const int zi = zs_of_plane[pl];
+        const long s = i >> 6;
+        const int lane = (int)(i & 63);
+        const long b0 = bptr[s];
+        const long blen = (bptr[s + 1] - b0) >> 6;
+        long j = 0;
+        // self block (D)
+        bcol[b0 + j * WAVE + lane] = (int)i;
+        for (int k = 0; k < DOF * DOF; ++k)
+            bvals[b0 * DOF * DOF + (j * DOF * DOF + k) * WAVE + lane] = D[k];
+        ++j;
+        for (int o = 0; o < ksten; ++o) {
+            const long cn = stencil_col_node(xi, yi, zi, (int)offs[o * 4],
+                                             (int)offs[o * 4 + 1],
+                                             (int)offs[o * 4 + 2], gx, gy, gz, pb);
+            if (cn < 0 || cn >= nown_nodes) continue;
+            const double w = offs[o * 4 + 3];
+            bcol[b0 + j * WAVE + lane] = (int)cn;
+            for (int k = 0; k < DOF * DOF; ++k)
+                bvals[b0 * DOF * DOF + (j * DOF * DOF + k) * WAVE + lane] = w * M[k];
+            ++j;
+        }
+        for (; j < blen; ++j) {  // padding: self col, zero block
+            bcol[b0 + j * WAVE + lane] = (int)i;
+            for (int k = 0; k < DOF * DOF; ++k)
+                bvals[b0 * DOF * DOF + (j * DOF * DOF + k) * WAVE + lane] = 0.0;
+        }
+    }
+}
+
+// ---------------------------------------------------------------------------
 // BLAS-1 / fused CG kernels.  Scalar coefficients come from the device slab.
 
 __global__ void __launch_bounds__(BLOCK)
@@ -592,6 +730,68 @@ k_stencil_fill(long nrows_nodes, long row0_node, int gx, int gy, int gz,
             vals[base + j * WAVE + lane] = 0.0;
         }
     }
+}
+
+void spmv_bsell(long nslices, long nnodes, int dof, uintptr_t bptr,
+                uintptr_t bcol, uintptr_t bvals, uintptr_t x, uintptr_t y,
+                uintptr_t partials, uintptr_t scal, int dotslot,
+                bool dot_accum, uintptr_t stream) {
+    if (nnodes == 0) return;
+    long blocks = (nslices * WAVE + BLOCK - 1) / BLOCK;
+    if (blocks > MAXG) blocks = MAXG;
+    const bool fuse = partials != 0 && dotslot >= 0;
+    dim3 g((unsigned)blocks), b(BLOCK);
+    #define LBS(D, FD) \
+        hipLaunchKernelGGL((k_spmv_bsell<D, FD>), g, b, 0, (hipStream_t)stream, \
+            nslices, nnodes, (const long*)bptr, (const int*)bcol, \
+            (const double*)bvals, (const double*)x, (double*)y, (double*)partials)
+    switch (dof) {
+        case 2: if (fuse) { LBS(2, true); } else { LBS(2, false); } break;
+        case 3: if (fuse) { LBS(3, true); } else { LBS(3, false); } break;
+        case 4: if (fuse) { LBS(4, true); } else { LBS(4, false); } break;
+        default: throw std::runtime_error("spmv_bsell: dof must be 2/3/4");
+    }
+    #undef LBS
+    check_hip("spmv_bsell");
+    if (fuse) {
+        hipLaunchKernelGGL(k_reduce_partials, dim3(1), dim3(BLOCK), 0,
+                           (hipStream_t)stream, (const double*)partials,
+                           (int)blocks, (double*)scal, dotslot, dot_accum ? 1 : 0);
+        check_hip("spmv_bsell_reduce");
+    }
+}
+
+void stencil_blocklen(long nnodes, int gx, int gy, int gz, long nown_nodes,
+                      uintptr_t zs_of_plane, uintptr_t pb, uintptr_t offs,
+                      int ksten, uintptr_t blocklen, uintptr_t stream) {
+    hipLaunchKernelGGL(k_stencil_blocklen, dim3((unsigned)elem_grid(nnodes)),
+                       dim3(BLOCK), 0, (hipStream_t)stream,
+                       nnodes, gx, gy, gz, nown_nodes, (const int*)zs_of_plane,
+                       (const long*)pb, (const double*)offs, ksten,
+                       (long*)blocklen);
+    check_hip("stencil_blocklen");
+}
+
+void stencil_bfill(long nnodes, int gx, int gy, int gz, int dof,
+                   long nown_nodes, uintptr_t zs_of_plane, uintptr_t pb,
+                   uintptr_t offs, int ksten, uintptr_t blocks_md,
+                   uintptr_t bptr, uintptr_t bcol, uintptr_t bvals,
+                   uintptr_t stream) {
+    dim3 g((unsigned)elem_grid(nnodes)), b(BLOCK);
+    #define LBF(D) \
+        hipLaunchKernelGGL((k_stencil_bfill<D>), g, b, 0, (hipStream_t)stream, \
+            nnodes, gx, gy, gz, nown_nodes, (const int*)zs_of_plane, \
+            (const long*)pb, (const double*)offs, ksten, \
+            (const double*)blocks_md, (const long*)bptr, (int*)bcol, \
+            (double*)bvals)
+    switch (dof) {
+        case 2: LBF(2); break;
+        case 3: LBF(3); break;
+        case 4: LBF(4); break;
+        default: throw std::runtime_error("stencil_bfill: dof must be 2/3/4");
+    }
+    #undef LBF
+    check_hip("stencil_bfill");
 }
 
 void stencil_rowlen(long nrows_nodes, long row0_node, int gx, int gy, int gz,
@@ -1049,6 +1249,9 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.def("cg_device", &cg_device);
     m.def("stencil_rowlen", &stencil_rowlen);
     m.def("stencil_fill", &stencil_fill);
+    m.def("spmv_bsell", &spmv_bsell);
+    m.def("stencil_blocklen", &stencil_blocklen);
+    m.def("stencil_bfill", &stencil_bfill);
     m.attr("S_RR") = S_RR;
     m.attr("S_PT") = S_PT;
     m.attr("S_RR_PREV") = S_RR_PREV;

@@ -61,10 +61,12 @@ class CGSolverHIP:
         self.A_rowptr = self.A_colidx = self.A_vals = None
         self.O_rowptr = self.O_colidx = self.O_vals = None
         self.lanesA = self.lanesO = lanes or 16
+        self.bsell = None
         if hasattr(L, "A_sell"):
             # device-generated system (gen.device_slab): SELL already in HBM
             self.sell = L.A_sell
             self.sellO = L.O_sell
+            self.bsell = getattr(L, "A_bsell", None)
         else:
             self.A_rowptr = up(L.A_rowptr)
             self.A_colidx = up(L.A_colidx)
@@ -153,7 +155,11 @@ class CGSolverHIP:
                     scal=self.scal if fuse_dotslot >= 0 else None,
                     dotslot=fuse_dotslot)
         with self.prof.span("spmvA"):
-            if self.sell is not None:
+            if self.bsell is not None:
+                bptr, bcol, bvals, dof = self.bsell
+                ops.spmv_bsell(bptr, bcol, bvals, self.n // dof, dof,
+                               xfull, y, **fuse)
+            elif self.sell is not None:
                 sellptr, scols, svals = self.sell
                 ops.spmv_sell(sellptr, scols, svals, self.n, xfull, y,
                               accum=False, perm=self.sell_perm, **fuse)

@@ -177,6 +177,36 @@ def test_pipelined_megafused_matches_fallback(problem):
     torch.testing.assert_close(x1[:S.nowned], x2[:S.nowned], rtol=1e-6, atol=1e-8)
 
 
+def test_bsell_matches_scalar_sell():
+    """Block-SELL SpMV == scalar SELL SpMV on the device-generated system."""
+    from acg_amd.gen import queen_like_spec
+    from acg_amd.gen.device_slab import device_stencil_slab
+    from acg_amd.ops import gpu_ops
+
+    for g, nranks, rank in ((13, 1, 0), (9, 2, 1)):
+        S = device_stencil_slab(g, g, g, queen_like_spec(3), rank, nranks,
+                                "cuda:0")
+        assert S.A_bsell is not None
+        x = torch.randn(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
+        y1 = torch.zeros(S.nowned, dtype=torch.float64, device="cuda")
+        sp, sc, sv = S.A_sell
+        gpu_ops.spmv_sell(sp, sc, sv, S.nowned, x, y1)
+        y2 = torch.zeros(S.nowned, dtype=torch.float64, device="cuda")
+        bptr, bcol, bvals, dof = S.A_bsell
+        gpu_ops.spmv_bsell(bptr, bcol, bvals, S.nowned // dof, dof, x, y2)
+        torch.testing.assert_close(y1, y2, rtol=1e-13, atol=1e-11)
+        # fused dot agrees too
+        scal1 = gpu_ops.alloc_scalars("cuda:0")
+        scal2 = gpu_ops.alloc_scalars("cuda:0")
+        part = gpu_ops.alloc_partials("cuda:0")
+        gpu_ops.spmv_sell(sp, sc, sv, S.nowned, x, y1, partials=part,
+                          scal=scal1, dotslot=0, dot_accum=False)
+        gpu_ops.spmv_bsell(bptr, bcol, bvals, S.nowned // dof, dof, x, y2,
+                           partials=part, scal=scal2, dotslot=0,
+                           dot_accum=False)
+        torch.testing.assert_close(scal1[0], scal2[0], rtol=1e-12, atol=1e-9)
+
+
 def test_device_generated_solver():
     from acg_amd.gen import queen_like_spec
     from acg_amd.gen.device_slab import device_stencil_slab
