@@ -108,6 +108,66 @@ def sell_from_csr(rowptr, colidx, vals, C: int = 64, sigma: int = 1):
     return sellptr, cols, svals
 
 
+def bsell_from_csr(rowptr, colidx, vals, dof: int, C: int = 64):
+    """Convert CSR -> Block-SELL (dense dof x dof blocks per node pair).
+
+    Returns (bptr int64[nslices+1] in block units, bcol int32, bvals f64
+    laid out [slice][j][k][64], density) or None when the matrix shape is
+    not divisible by dof.  Blocks missing entries are zero-filled; the
+    caller decides (via density) whether the 4->4/dof^2 B/nnz index saving
+    beats the zero-fill cost."""
+    import numpy as np
+
+    rowptr = np.asarray(rowptr)
+    colidx = np.asarray(colidx, dtype=np.int64)
+    vals = np.asarray(vals)
+    nrows = len(rowptr) - 1
+    if dof < 2 or nrows % dof:
+        return None
+    nnodes = nrows // dof
+    nnz = len(colidx)
+    rows = np.repeat(np.arange(nrows, dtype=np.int64), np.diff(rowptr))
+    node = rows // dof
+    bcol_of_entry = colidx // dof
+    # unique (node, blockcol) pairs -> per-node block lists
+    key = node * (np.int64(1) << 32) | bcol_of_entry.astype(np.int64)
+    uk, inv = np.unique(key, return_inverse=True)
+    nblocks = len(uk)
+    bnode = (uk >> 32).astype(np.int64)
+    bcols = (uk & 0xFFFFFFFF).astype(np.int64)
+    density = nnz / (nblocks * dof * dof)
+    blocks_per_node = np.bincount(bnode, minlength=nnodes)
+    nslices = (nnodes + C - 1) // C
+    padlen = np.zeros(nslices * C, dtype=np.int64)
+    padlen[:nnodes] = blocks_per_node
+    slice_len = padlen.reshape(nslices, C).max(axis=1)
+    bptr = np.zeros(nslices + 1, dtype=np.int64)
+    np.cumsum(slice_len * C, out=bptr[1:])
+    btotal = int(bptr[-1])
+    # default: self block col (in-bounds), zero values
+    slice_of_p = np.repeat(np.arange(nslices, dtype=np.int64), slice_len * C)
+    lane = (np.arange(btotal, dtype=np.int64) - bptr[slice_of_p]) % C
+    bcol = np.minimum(slice_of_p * C + lane, nnodes - 1).astype(np.int32)
+    bvals = np.zeros(btotal * dof * dof, dtype=np.float64)
+    # position of block b within its node's list (blocks sorted by key =>
+    # grouped by node, ascending bcol)
+    starts = np.zeros(nnodes + 1, dtype=np.int64)
+    np.cumsum(blocks_per_node, out=starts[1:])
+    within = np.arange(nblocks, dtype=np.int64) - starts[bnode]
+    bdst = bptr[bnode // C] + within * C + bnode % C
+    bcol[bdst] = bcols.astype(np.int32)
+    # scatter entries into their block slots
+    entry_block_dst = bdst[inv]
+    k = (rows % dof) * dof + colidx % dof
+    # bvals index: (block position p, k) -> (p // C)*C*dof^2 ... layout
+    # [slice][j][k][64]: linear = (p - lane_p) * dof^2 + k*C + lane_p
+    lane_p = entry_block_dst % C
+    base_p = entry_block_dst - lane_p
+    vdst = base_p * (dof * dof) + k * C + lane_p
+    bvals[vdst] = vals
+    return bptr, bcol, bvals, density
+
+
 def spmv_sell(sellptr, cols, vals, nrows, x, y, *, rowbase: int = 0,
               accum: bool = False, partials=None, scal=None,
               dotslot: int = -1, dot_accum: bool = True) -> None:

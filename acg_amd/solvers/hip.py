@@ -96,11 +96,28 @@ class CGSolverHIP:
                         self.sell_perm = up(perm)
                 if waste <= 0.5:
                     self.sell = (up(sellptr), up(scols), up(svals))
-                    self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR
                     if L.nnzO > 0:
                         optr, ocols, ovals = sell_from_csr(L.O_rowptr, L.O_colidx,
                                                            L.O_vals)
                         self.sellO = (up(optr), up(ocols), up(ovals))
+                # Block-SELL when the matrix has dense dof x dof blocks
+                # (FEM/structural, e.g. Queen_4147's 3-dof nodes): one int32
+                # index per block.  Worth it when block density is high
+                # enough that the index saving beats the zero-fill.
+                from ..ops.torch_ref import bsell_from_csr
+
+                for dof_try in (3, 2):
+                    out = bsell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals,
+                                         dof_try)
+                    if out is None:
+                        continue
+                    bptr_h, bcol_h, bvals_h, density = out
+                    if density >= 0.75:
+                        self.bsell = (up(bptr_h), up(bcol_h), up(bvals_h),
+                                      dof_try)
+                        break
+                if self.sell is not None or self.bsell is not None:
+                    self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR
         # megafused pipelined iteration needs SELL everywhere + int32 cols.
         # Measured on MI355X: for wide rows (~80 nnz, Queen-shaped) the fused
         # epilogue's 6 vector streams cost the SpMV more x-gather locality

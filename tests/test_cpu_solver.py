@@ -142,6 +142,46 @@ def test_slab_generator_matches_global(spec, g, nranks):
     assert covered.all()
 
 
+def test_bsell_from_csr_layout():
+    """Host Block-SELL conversion: emulate the kernel's indexing in numpy
+    and compare against the CSR SpMV."""
+    from acg_amd.gen import queen_like_spec
+    from acg_amd.ops.torch_ref import bsell_from_csr
+    from acg_amd.part import extract_subdomains, partition_rows
+
+    A = stencil_global(5, 5, 5, queen_like_spec(3))
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    out = bsell_from_csr(S.A_rowptr, S.A_colidx, S.A_vals, 3)
+    assert out is not None
+    bptr, bcol, bvals, density = out
+    assert density > 0.95, density
+    rng = np.random.default_rng(0)
+    x = rng.standard_normal(S.nowned)
+    dof, C = 3, 64
+    nnodes = S.nowned // dof
+    y = np.zeros(S.nowned)
+    nslices = len(bptr) - 1
+    for s in range(nslices):
+        b0 = int(bptr[s])
+        blen = (int(bptr[s + 1]) - b0) // C
+        for lane in range(C):
+            node = s * C + lane
+            if node >= nnodes:
+                continue
+            acc = np.zeros(dof)
+            for j in range(blen):
+                cb = int(bcol[b0 + j * C + lane])
+                xv = x[cb * dof:(cb + 1) * dof]
+                for k in range(dof * dof):
+                    a = bvals[b0 * dof * dof + (j * dof * dof + k) * C + lane]
+                    acc[k // dof] += a * xv[k % dof]
+            y[node * dof:(node + 1) * dof] = acc
+    import scipy.sparse as sp
+
+    Acsr = sp.csr_matrix((S.A_vals, S.A_colidx.astype(np.int64), S.A_rowptr))
+    np.testing.assert_allclose(y, Acsr @ x, rtol=1e-12, atol=1e-12)
+
+
 def test_slab_halo_pairing():
     from acg_amd.gen import stencil_local_slab
 
