@@ -356,11 +356,11 @@ class CGSolverHIP:
 
             raise AcgError(ErrCode.NOT_SUPPORTED,
                            "device-side CG is single-GPU (reference parity)")
-        if self.sell is None:
+        if self.sell is None or self.sell_perm is not None:
             from ..utils.errors import AcgError, ErrCode
 
             raise AcgError(ErrCode.NOT_SUPPORTED,
-                           "device-side CG requires the SELL operator format")
+                           "device-side CG requires the unpermuted SELL format")
         res = SolveResult(solver="cg-hip-device", maxits=maxits,
                           res_atol=res_atol, res_rtol=res_rtol, nranks=1)
         n = self.n
