@@ -47,8 +47,11 @@ def make_parser() -> argparse.ArgumentParser:
     p.add_argument("--seed", type=int, default=0, help="partitioner seed")
     p.add_argument("--solver", default=None,
                    choices=("acg", "acg-pipelined", "acg-device", "cpu",
-                            "cpu-pipelined", "scipy", "scipy-pipelined"),
-                   help="default: acg on GPU, cpu otherwise")
+                            "cpu-pipelined", "scipy", "scipy-pipelined",
+                            "petsc", "petsc-pipelined"),
+                   help="default: acg on GPU, cpu otherwise.  petsc[-pipelined] "
+                        "are accepted for aCG compatibility and run the scipy "
+                        "oracle (PETSc is not available in this image)")
     p.add_argument("--max-iterations", type=int, default=100)
     p.add_argument("--diff-atol", type=float, default=0.0)
     p.add_argument("--diff-rtol", type=float, default=0.0)
@@ -86,6 +89,10 @@ def main(argv=None) -> int:
     if commkind is None:
         commkind = ("rccl" if has_gpu else "gloo") if world > 1 else "none"
     solver_name = args.solver or ("acg" if has_gpu else "cpu")
+    # aCG compatibility: the PETSc baseline role is played by scipy here
+    solver_name = {"petsc": "scipy",
+                   "petsc-pipelined": "scipy-pipelined"}.get(solver_name,
+                                                             solver_name)
     gpu_solver = solver_name.startswith("acg")
     if gpu_solver and not has_gpu:
         print("error: GPU solver requested but no GPU is available", file=sys.stderr)

@@ -177,6 +177,24 @@ def test_pipelined_megafused_matches_fallback(problem):
     torch.testing.assert_close(x1[:S.nowned], x2[:S.nowned], rtol=1e-6, atol=1e-8)
 
 
+def test_profile_mode(problem):
+    """hipEvent per-op profiling produces sane per-op stats on GPU."""
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A, S = problem
+    rng = np.random.default_rng(21)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+    gpu = CGSolverHIP(S, device="cuda:0", profile=True)
+    x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
+    res = gpu.solve_pipelined(b, x, maxits=50, res_rtol=0.0)
+    assert "update" in res.ops
+    up = res.ops["update"]
+    assert up.count >= 50 and up.seconds > 0
+    assert up.bytes > 0  # analytic annotation attached
+    res2 = gpu.solve(b, torch.zeros_like(x), maxits=50, res_rtol=0.0)
+    assert "spmvA" in res2.ops and res2.ops["spmvA"].seconds > 0
+
+
 def test_bsell_matches_scalar_sell():
     """Block-SELL SpMV == scalar SELL SpMV on the device-generated system."""
     from acg_amd.gen import queen_like_spec
