@@ -211,13 +211,23 @@ k_spmv_sell(long nslices, long nrows, long rowbase,
 // so bytes are the only lever left).
 //
 // Layout per 64-node slice s (bptr in block units):
-//   bcol[bptr[s] + j*64 + lane]                 block-col of node's j-th block
-//   bvals[bptr[s]*dof^2 + (j*dof^2 + k)*64 + lane]   value k (row-major in
-//                                               the block), all 64 lanes
-//                                               contiguous per (j,k)
+//   bcol[bptr[s] + j*64 + lane]              block-col of node's j-th block
+//   bvals: PAIR-major so each lane's value pair (k, k+1) is 16 B
+//   contiguous and clang emits dwordx4 loads (16 B/lane = the coalescing
+//   sweet spot): element (j,k,lane) at
+//     bptr[s]*dof^2 + j*dof^2*64 + bval_off(k, lane)
+//   with bval_off = (k/2)*128 + lane*2 + (k&1) for paired k, and the odd
+//   tail element (k = dof^2-1 when dof^2 is odd) at (dof^2-1)*64 + lane.
 __device__ __forceinline__ long stencil_col_node(
     int xi, int yi, int zi, int dx, int dy, int dz, int gx, int gy, int gz,
     const long* __restrict__ pb);  // defined with the stencil generators below
+
+template <int D2>
+__device__ __forceinline__ long bval_off(int k, int lane) {
+    constexpr int EVEN = D2 & ~1;
+    return (k < EVEN) ? (long)(k >> 1) * (2 * WAVE) + lane * 2 + (k & 1)
+                      : (long)EVEN * WAVE + lane;
+}
 
 template <int DOF, bool FUSE_DOT>
 __global__ void __launch_bounds__(BLOCK)
@@ -236,7 +246,7 @@ k_spmv_bsell(long nslices, long nnodes,
         const long b0 = bptr[s];
         const long blen = (bptr[s + 1] - b0) >> 6;  // blocks per node
         const int* __restrict__ c = bcol + b0 + lane;
-        const double* __restrict__ v = bvals + b0 * (DOF * DOF) + lane;
+        const double* __restrict__ v = bvals + b0 * (DOF * DOF);
         double acc[DOF];
         #pragma unroll
         for (int r = 0; r < DOF; ++r) acc[r] = 0.0;
@@ -245,9 +255,10 @@ k_spmv_bsell(long nslices, long nnodes,
             double xv[DOF];
             #pragma unroll
             for (int cc = 0; cc < DOF; ++cc) xv[cc] = x[(long)cb * DOF + cc];
+            const double* __restrict__ vj = v + j * (DOF * DOF) * WAVE;
             #pragma unroll
             for (int k = 0; k < DOF * DOF; ++k) {
-                const double a = ld_nt(v + (j * DOF * DOF + k) * WAVE);
+                const double a = ld_nt(vj + bval_off<DOF * DOF>(k, lane));
                 acc[k / DOF] += a * xv[k % DOF];
             }
         }
@@ -315,10 +326,11 @@ k_stencil_bfill(long nnodes, int gx, int gy, int gz, long nown_nodes,
         const long b0 = bptr[s];
         const long blen = (bptr[s + 1] - b0) >> 6;
         long j = 0;
+        double* __restrict__ bv = bvals + b0 * DOF * DOF;
         // self block (D)
         bcol[b0 + j * WAVE + lane] = (int)i;
         for (int k = 0; k < DOF * DOF; ++k)
-            bvals[b0 * DOF * DOF + (j * DOF * DOF + k) * WAVE + lane] = D[k];
+            bv[j * DOF * DOF * WAVE + bval_off<DOF * DOF>(k, lane)] = D[k];
         ++j;
         for (int o = 0; o < ksten; ++o) {
             const long cn = stencil_col_node(xi, yi, zi, (int)offs[o * 4],
@@ -328,13 +340,13 @@ k_stencil_bfill(long nnodes, int gx, int gy, int gz, long nown_nodes,
             const double w = offs[o * 4 + 3];
             bcol[b0 + j * WAVE + lane] = (int)cn;
             for (int k = 0; k < DOF * DOF; ++k)
-                bvals[b0 * DOF * DOF + (j * DOF * DOF + k) * WAVE + lane] = w * M[k];
+                bv[j * DOF * DOF * WAVE + bval_off<DOF * DOF>(k, lane)] = w * M[k];
             ++j;
         }
         for (; j < blen; ++j) {  // padding: self col, zero block
             bcol[b0 + j * WAVE + lane] = (int)i;
             for (int k = 0; k < DOF * DOF; ++k)
-                bvals[b0 * DOF * DOF + (j * DOF * DOF + k) * WAVE + lane] = 0.0;
+                bv[j * DOF * DOF * WAVE + bval_off<DOF * DOF>(k, lane)] = 0.0;
         }
     }
 }

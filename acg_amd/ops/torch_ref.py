@@ -159,11 +159,16 @@ def bsell_from_csr(rowptr, colidx, vals, dof: int, C: int = 64):
     # scatter entries into their block slots
     entry_block_dst = bdst[inv]
     k = (rows % dof) * dof + colidx % dof
-    # bvals index: (block position p, k) -> (p // C)*C*dof^2 ... layout
-    # [slice][j][k][64]: linear = (p - lane_p) * dof^2 + k*C + lane_p
+    # bvals index, PAIR-major (see kernels.hip bval_off): element (j,k,lane)
+    # at base_p*dof^2 + (k//2)*2C + lane*2 + k%2 for paired k, odd tail at
+    # (dof^2-1)*C + lane
     lane_p = entry_block_dst % C
     base_p = entry_block_dst - lane_p
-    vdst = base_p * (dof * dof) + k * C + lane_p
+    D2 = dof * dof
+    even = D2 & ~1
+    off = np.where(k < even, (k >> 1) * (2 * C) + lane_p * 2 + (k & 1),
+                   even * C + lane_p)
+    vdst = base_p * D2 + off
     bvals[vdst] = vals
     return bptr, bcol, bvals, density
 

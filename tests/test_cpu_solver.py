@@ -169,11 +169,15 @@ def test_bsell_from_csr_layout():
             if node >= nnodes:
                 continue
             acc = np.zeros(dof)
+            D2 = dof * dof
+            even = D2 & ~1
             for j in range(blen):
                 cb = int(bcol[b0 + j * C + lane])
                 xv = x[cb * dof:(cb + 1) * dof]
-                for k in range(dof * dof):
-                    a = bvals[b0 * dof * dof + (j * dof * dof + k) * C + lane]
+                for k in range(D2):
+                    off = ((k >> 1) * 2 * C + lane * 2 + (k & 1)
+                           if k < even else even * C + lane)
+                    a = bvals[b0 * D2 + j * D2 * C + off]
                     acc[k // dof] += a * xv[k % dof]
             y[node * dof:(node + 1) * dof] = acc
     import scipy.sparse as sp
