@@ -78,6 +78,19 @@ def main():
                                                    variant=v),
                f"sell variant={variant} [{'+'.join(tags) or 'base'}]")
 
+    # Block-SELL (if the operator has dof-blocks)
+    from acg_amd.ops.torch_ref import bsell_from_csr
+
+    out = bsell_from_csr(S.A_rowptr, S.A_colidx, S.A_vals, args.dof)
+    if out is not None and out[3] >= 0.75:
+        bptr = torch.from_numpy(out[0]).to(dev)
+        bcol = torch.from_numpy(out[1]).to(dev)
+        bvals = torch.from_numpy(out[2]).to(dev)
+        print(f"BSELL dof={args.dof} density={out[3]:.3f}")
+        timeit(lambda: gpu_ops.spmv_bsell(bptr, bcol, bvals, n // args.dof,
+                                          args.dof, x, y),
+               f"bsell dof={args.dof} (pair-major)")
+
 
 if __name__ == "__main__":
     main()
