@@ -211,16 +211,20 @@ def pipelined_fused(z, t, p, x, r, w, q, scal: torch.Tensor,
 def cg_device(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
               nrows: int, b: torch.Tensor, x: torch.Tensor, r: torch.Tensor,
               p: torch.Tensor, t: torch.Tensor, scal: torch.Tensor,
-              partials: torch.Tensor, out2: torch.Tensor, maxits: int,
+              partials: torch.Tensor, out2: torch.Tensor,
+              barrier_state: torch.Tensor, maxits: int,
               res_atol: float, res_rtol: float) -> int:
     """Monolithic device-side CG: one cooperative launch runs the whole
-    solve.  Returns the grid size used."""
+    solve.  ``barrier_state``: 3 zeroed uint32 words (grid-barrier counter,
+    generation, fail flag -- must be zeroed before EVERY launch).
+    Returns the grid size used."""
     nslices = sellptr.numel() - 1
     return K.cg_device(nslices, nrows, sellptr.data_ptr(), cols.data_ptr(),
                        1 if cols.dtype == torch.int64 else 0, vals.data_ptr(),
                        b.data_ptr(), x.data_ptr(), r.data_ptr(), p.data_ptr(),
                        t.data_ptr(), scal.data_ptr(), partials.data_ptr(),
-                       out2.data_ptr(), maxits, res_atol, res_rtol, _stream())
+                       out2.data_ptr(), barrier_state.data_ptr(),
+                       maxits, res_atol, res_rtol, _stream())
 
 
 def pack_gather(sendbuf: torch.Tensor, x: torch.Tensor, idx: torch.Tensor) -> None:

@@ -844,8 +844,55 @@ void stencil_fill(long nrows_nodes, long row0_node, int gx, int gy, int gz,
 // occupancy query, everything inside is grid-stride.
 #include <hip/hip_cooperative_groups.h>
 
+#define RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+typedef __attribute__((address_space(1))) unsigned int gu32;
+
+// Hand-rolled grid barrier (guide §6 G16): cooperative_groups::grid.sync()
+// measured ~150 us per sync on ROCm 7.2 (system-scope fences + L2
+// writebacks), which made the monolithic CG ~6x slower than host-driven.
+// This one is the generation-counter form of the recipe: every wave drains
+// its stores (vmcnt), one lane per block does an agent-scope release +
+// relaxed arrival; the last arriver resets the counter and bumps the
+// generation; everyone else polls the generation RELAXED (never acquire in
+// the loop), then ONE acquire drops stale L1 lines.  Spins are bounded:
+// on timeout the barrier sets fail[0] and every block exits (no hang).
+// Cooperative launch is kept purely for the residency guarantee.
+__device__ __forceinline__ bool grid_barrier(gu32* cnt, gu32* gen, gu32* fail) {
+    // every wave: drain pending payload stores before signalling
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    bool ok = true;
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // G16 pitfall 12
+        const unsigned g = __hip_atomic_load(gen, RLX_AGENT);
+        const unsigned arrived = __hip_atomic_fetch_add(cnt, 1u, RLX_AGENT) + 1u;
+        if (arrived == gridDim.x) {
+            __hip_atomic_store(cnt, 0u, RLX_AGENT);
+            __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+            __hip_atomic_store(gen, g + 1u, RLX_AGENT);
+        } else {
+            unsigned spins = 0;
+            while (__hip_atomic_load(gen, RLX_AGENT) == g) {
+                __builtin_amdgcn_s_sleep(8);
+                if (++spins > 400000000u) {  // ~bounded: never hang the GPU
+                    __hip_atomic_store(fail, 1u, RLX_AGENT);
+                    ok = false;
+                    break;
+                }
+                if (__hip_atomic_load(fail, RLX_AGENT)) { ok = false; break; }
+            }
+        }
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    }
+    __syncthreads();
+    return ok;
+}
+
 // results: scal[S_BNRM2]=(b,b), scal[S_RR_PREV]=(r0,r0), scal[S_RR]=final
-// (r,r); out2[0]=niterations, out2[1]=converged.
+// (r,r); out2[0]=niterations, out2[1]=converged (-1 = barrier timeout).
+// barrier_state: 3 zeroed u32 words {cnt, gen, fail}.
 template <typename ColT>
 __global__ void __launch_bounds__(BLOCK)
 k_cg_device(long nslices, long nrows,
@@ -855,22 +902,28 @@ k_cg_device(long nslices, long nrows,
             double* __restrict__ r, double* __restrict__ p,
             double* __restrict__ t,
             double* __restrict__ scal, double* __restrict__ partials,
-            int* __restrict__ out2,
+            int* __restrict__ out2, unsigned* __restrict__ barrier_state,
             int maxits, double res_atol, double res_rtol) {
-    namespace cg = cooperative_groups;
-    cg::grid_group grid = cg::this_grid();
     const long tid = (long)blockIdx.x * BLOCK + threadIdx.x;
     const long nth = (long)gridDim.x * BLOCK;
     const int lane = threadIdx.x & (WAVE - 1);
     const long wslice0 = tid >> 6;
     const long nw = nth >> 6;
+    gu32* bar_cnt = (gu32*)(barrier_state + 0);
+    gu32* bar_gen = (gu32*)(barrier_state + 1);
+    gu32* bar_fail = (gu32*)(barrier_state + 2);
+    bool alive = true;
+    auto gsync = [&]() {
+        if (alive && !grid_barrier(bar_cnt, bar_gen, bar_fail)) alive = false;
+        return alive;
+    };
 
     // grid-wide sum helper: every block contributes partials[bid]; block 0
     // reduces into scal[slot]; two grid barriers bracket it.
     auto grid_sum = [&](double v, int slot) {
         v = block_reduce(v);
         if (threadIdx.x == 0) partials[blockIdx.x] = v;
-        grid.sync();
+        if (!gsync()) return;
         if (blockIdx.x == 0) {
             double s = 0.0;
             for (int i = threadIdx.x; i < (int)gridDim.x; i += BLOCK)
@@ -878,7 +931,7 @@ k_cg_device(long nslices, long nrows,
             s = block_reduce(s);
             if (threadIdx.x == 0) scal[slot] = s;
         }
-        grid.sync();
+        gsync();
     };
     auto spmv = [&](const double* __restrict__ xin, double* __restrict__ yout,
                     bool fuse) -> double {
@@ -912,7 +965,7 @@ k_cg_device(long nslices, long nrows,
     for (long i = tid; i < nrows; i += nth) acc += b[i] * b[i];
     grid_sum(acc, S_BNRM2);
     spmv(x, t, false);
-    grid.sync();
+    gsync();
     acc = 0.0;
     for (long i = tid; i < nrows; i += nth) {
         const double ri = b[i] - t[i];
@@ -921,18 +974,19 @@ k_cg_device(long nslices, long nrows,
         acc += ri * ri;
     }
     grid_sum(acc, S_RR);
-    const double bnrm2sqr = scal[S_BNRM2];
-    double rr = scal[S_RR];
+    const double bnrm2sqr = alive ? scal[S_BNRM2] : 1.0;
+    double rr = alive ? scal[S_RR] : 0.0;
     if (tid == 0) scal[S_RR_PREV] = rr;  // report (r0,r0)
     const double rt = res_rtol * sqrt(bnrm2sqr) > res_atol
         ? res_rtol * sqrt(bnrm2sqr) : res_atol;
     const double rtol2 = rt * rt;
     int k = 0, converged = (rtol2 > 0.0 && rr <= rtol2) ? 1 : 0;
-    while (!converged && k < maxits) {
+    while (alive && !converged && k < maxits) {
         // t = A p (fused (p,t))
-        grid.sync();  // p is consistent (written by all blocks last iter)
+        gsync();  // p is consistent (written by all blocks last iter)
         const double pt_part = spmv(p, t, true);
         grid_sum(pt_part, S_PT);
+        if (!alive) break;
         const double alpha = rr / scal[S_PT];
         acc = 0.0;
         for (long i = tid; i < nrows; i += nth) {
@@ -942,6 +996,7 @@ k_cg_device(long nslices, long nrows,
             acc += rn * rn;
         }
         grid_sum(acc, S_RR);
+        if (!alive) break;
         const double rr_new = scal[S_RR];
         const double beta = rr_new / rr;
         for (long i = tid; i < nrows; i += nth)
@@ -953,7 +1008,7 @@ k_cg_device(long nslices, long nrows,
     if (tid == 0) {
         scal[S_RR] = rr;
         out2[0] = k;
-        out2[1] = converged;
+        out2[1] = alive ? converged : -1;
     }
     (void)bnrm2sqr;
 }
@@ -1192,7 +1247,8 @@ void pipelined_fused(uintptr_t z, uintptr_t t, uintptr_t p, uintptr_t x, uintptr
 int cg_device(long nslices, long nrows, uintptr_t sellptr, uintptr_t cols,
               int col64, uintptr_t vals, uintptr_t b, uintptr_t x, uintptr_t r,
               uintptr_t p, uintptr_t t, uintptr_t scal, uintptr_t partials,
-              uintptr_t out2, int maxits, double res_atol, double res_rtol,
+              uintptr_t out2, uintptr_t barrier_state,
+              int maxits, double res_atol, double res_rtol,
               uintptr_t stream) {
     int dev = 0;
     hipGetDevice(&dev);
@@ -1214,7 +1270,8 @@ int cg_device(long nslices, long nrows, uintptr_t sellptr, uintptr_t cols,
         void* args[] = {&nslices, &nrows, (void*)&sellptr, (void*)&cols,
                         (void*)&vals, (void*)&b, (void*)&x, (void*)&r,
                         (void*)&p, (void*)&t, (void*)&scal, (void*)&partials,
-                        (void*)&out2, &maxits, &res_atol, &res_rtol};
+                        (void*)&out2, (void*)&barrier_state,
+                        &maxits, &res_atol, &res_rtol};
         hipError_t e = hipLaunchCooperativeKernel(
             kern, dim3((unsigned)grid), dim3(BLOCK), args, 0, S(stream));
         if (e == hipSuccess) break;

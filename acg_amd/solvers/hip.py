@@ -368,11 +368,13 @@ class CGSolverHIP:
         t = self._vec()
         p = self._vec(nghost=True)
         out2 = torch.zeros(2, dtype=torch.int32, device=self.device)
+        barrier_state = torch.zeros(3, dtype=torch.int32, device=self.device)
         sellptr, scols, svals = self.sell
         torch.cuda.synchronize(self.device)
         t0 = time.perf_counter()
         ops.cg_device(sellptr, scols, svals, n, b, x, r, p, t, self.scal,
-                      self.partials, out2, maxits, res_atol, res_rtol)
+                      self.partials, out2, barrier_state, maxits, res_atol,
+                      res_rtol)
         torch.cuda.synchronize(self.device)
         res.tsolve = time.perf_counter() - t0
         S = ops
@@ -380,7 +382,10 @@ class CGSolverHIP:
         res.r0nrm2 = math.sqrt(max(float(self.scal[S.S_RR_PREV]), 0.0))
         res.rnrm2 = math.sqrt(max(float(self.scal[S.S_RR]), 0.0))
         res.niterations = int(out2[0])
-        res.converged = bool(int(out2[1]))
+        conv = int(out2[1])
+        if conv < 0:
+            raise RuntimeError("device CG grid barrier timed out (residency?)")
+        res.converged = bool(conv)
         nnz_full = self.local.nnzA + self.local.nnzO
         res.nflops = res.niterations * cg_flops_per_iter(nnz_full, n)
         self.niterations_total += res.niterations
