@@ -848,35 +848,60 @@ void stencil_fill(long nrows_nodes, long row0_node, int gx, int gy, int gz,
 typedef __attribute__((address_space(1))) unsigned int gu32;
 
 // Hand-rolled grid barrier (guide §6 G16): cooperative_groups::grid.sync()
-// measured ~150 us per sync on ROCm 7.2 (system-scope fences + L2
-// writebacks), which made the monolithic CG ~6x slower than host-driven.
-// This one is the generation-counter form of the recipe: every wave drains
-// its stores (vmcnt), one lane per block does an agent-scope release +
-// relaxed arrival; the last arriver resets the counter and bumps the
-// generation; everyone else polls the generation RELAXED (never acquire in
-// the loop), then ONE acquire drops stale L1 lines.  Spins are bounded:
-// on timeout the barrier sets fail[0] and every block exits (no hang).
-// Cooperative launch is kept purely for the residency guarantee.
-__device__ __forceinline__ bool grid_barrier(gu32* cnt, gu32* gen, gu32* fail) {
-    // every wave: drain pending payload stores before signalling
+// measured ~150 us per sync on ROCm 7.2.  v2 design, after measuring a
+// per-block release-fence + single-word-poll version still slow at >1000
+// blocks:
+//  - ALL shared payloads (vectors, partials) are written WRITE-THROUGH
+//    (sc1: relaxed agent-scope 8-byte atomic stores) so no release fence
+//    (no per-block buffer_wbl2 storm) is needed -- every wave just drains
+//    vmcnt before its block arrives (guide R1 + pitfall 14).
+//  - waiters poll a PER-XCD copy of the generation word (the dispatcher
+//    places block b on XCD b%8, so each poll line serves ~1/8 of the
+//    blocks; correctness does not depend on the placement, every copy is
+//    written).  Polling is RELAXED with s_sleep backoff; ONE acquire
+//    (L1 invalidate) per block after wake.  Spins are bounded: on timeout
+//    fail[0] is set and every block exits (no hang).
+// barrier_state layout (u32 words): [0]=cnt(even phases), [8]=cnt(odd
+// phases) (sense-reversing parity counters: the releaser resets the
+// just-used counter, which no block touches again until two barriers
+// later -- a racing early arrival for the NEXT barrier targets the OTHER
+// word, so the reset can never eat an arrival), [1]=fail,
+// [16+16*x]=generation copy for XCD x (64 B stride: distinct lines).
+#define BAR_GEN0 16
+#define BAR_GENSTRIDE 16
+
+__device__ __forceinline__ void st_sc1_f64(double* p, double v) {
+    __hip_atomic_store(reinterpret_cast<unsigned long long*>(p),
+                       (unsigned long long)__double_as_longlong(v), RLX_AGENT);
+}
+
+__device__ __forceinline__ bool grid_barrier(unsigned* state) {
+    gu32* fail = (gu32*)(state + 1);
+    // every wave: drain pending sc1 payload stores before signalling (also
+    // completes this block's own reset/gen stores from the previous
+    // barrier before its new arrival -- the parity-counter safety hinges
+    // on this drain)
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
     bool ok = true;
     if (threadIdx.x == 0) {
-        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // G16 pitfall 12
-        const unsigned g = __hip_atomic_load(gen, RLX_AGENT);
+        const int myxcd = (int)(blockIdx.x & 7);
+        gu32* mygen = (gu32*)(state + BAR_GEN0 + BAR_GENSTRIDE * myxcd);
+        const unsigned g = __hip_atomic_load(mygen, RLX_AGENT);
+        gu32* cnt = (gu32*)(state + ((g & 1u) ? 8 : 0));
         const unsigned arrived = __hip_atomic_fetch_add(cnt, 1u, RLX_AGENT) + 1u;
         if (arrived == gridDim.x) {
             __hip_atomic_store(cnt, 0u, RLX_AGENT);
-            __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
-            asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-            __hip_atomic_store(gen, g + 1u, RLX_AGENT);
+            #pragma unroll
+            for (int xx = 0; xx < 8; ++xx)
+                __hip_atomic_store((gu32*)(state + BAR_GEN0 + BAR_GENSTRIDE * xx),
+                                   g + 1u, RLX_AGENT);
         } else {
             unsigned spins = 0;
-            while (__hip_atomic_load(gen, RLX_AGENT) == g) {
-                __builtin_amdgcn_s_sleep(8);
-                if (++spins > 400000000u) {  // ~bounded: never hang the GPU
+            while (__hip_atomic_load(mygen, RLX_AGENT) == g) {
+                if (spins < 32) __builtin_amdgcn_s_sleep(2);
+                else __builtin_amdgcn_s_sleep(64);
+                if (++spins > 100000000u) {  // bounded: never hang the GPU
                     __hip_atomic_store(fail, 1u, RLX_AGENT);
                     ok = false;
                     break;
@@ -909,27 +934,24 @@ k_cg_device(long nslices, long nrows,
     const int lane = threadIdx.x & (WAVE - 1);
     const long wslice0 = tid >> 6;
     const long nw = nth >> 6;
-    gu32* bar_cnt = (gu32*)(barrier_state + 0);
-    gu32* bar_gen = (gu32*)(barrier_state + 1);
-    gu32* bar_fail = (gu32*)(barrier_state + 2);
     bool alive = true;
     auto gsync = [&]() {
-        if (alive && !grid_barrier(bar_cnt, bar_gen, bar_fail)) alive = false;
+        if (alive && !grid_barrier(barrier_state)) alive = false;
         return alive;
     };
 
-    // grid-wide sum helper: every block contributes partials[bid]; block 0
-    // reduces into scal[slot]; two grid barriers bracket it.
+    // grid-wide sum helper: every block contributes partials[bid] (sc1);
+    // block 0 reduces into scal[slot]; two grid barriers bracket it.
     auto grid_sum = [&](double v, int slot) {
         v = block_reduce(v);
-        if (threadIdx.x == 0) partials[blockIdx.x] = v;
+        if (threadIdx.x == 0) st_sc1_f64(partials + blockIdx.x, v);
         if (!gsync()) return;
         if (blockIdx.x == 0) {
             double s = 0.0;
             for (int i = threadIdx.x; i < (int)gridDim.x; i += BLOCK)
                 s += partials[i];
             s = block_reduce(s);
-            if (threadIdx.x == 0) scal[slot] = s;
+            if (threadIdx.x == 0) st_sc1_f64(scal + slot, s);
         }
         gsync();
     };
@@ -953,7 +975,7 @@ k_cg_device(long nslices, long nrows,
             for (; j < len; ++j) sum += ld_nt(v + j * WAVE) * xin[c[j * WAVE]];
             const long row = s * WAVE + lane;
             if (row < nrows) {
-                yout[row] = sum;
+                st_sc1_f64(yout + row, sum);
                 if (fuse) dacc += xin[row] * sum;
             }
         }
@@ -969,14 +991,14 @@ k_cg_device(long nslices, long nrows,
     acc = 0.0;
     for (long i = tid; i < nrows; i += nth) {
         const double ri = b[i] - t[i];
-        r[i] = ri;
-        p[i] = ri;
+        st_sc1_f64(r + i, ri);
+        st_sc1_f64(p + i, ri);
         acc += ri * ri;
     }
     grid_sum(acc, S_RR);
     const double bnrm2sqr = alive ? scal[S_BNRM2] : 1.0;
     double rr = alive ? scal[S_RR] : 0.0;
-    if (tid == 0) scal[S_RR_PREV] = rr;  // report (r0,r0)
+    if (tid == 0) st_sc1_f64(scal + S_RR_PREV, rr);  // report (r0,r0)
     const double rt = res_rtol * sqrt(bnrm2sqr) > res_atol
         ? res_rtol * sqrt(bnrm2sqr) : res_atol;
     const double rtol2 = rt * rt;
@@ -991,8 +1013,8 @@ k_cg_device(long nslices, long nrows,
         acc = 0.0;
         for (long i = tid; i < nrows; i += nth) {
             const double rn = r[i] - alpha * t[i];
-            r[i] = rn;
-            x[i] += alpha * p[i];
+            st_sc1_f64(r + i, rn);
+            st_sc1_f64(x + i, x[i] + alpha * p[i]);
             acc += rn * rn;
         }
         grid_sum(acc, S_RR);
@@ -1000,13 +1022,13 @@ k_cg_device(long nslices, long nrows,
         const double rr_new = scal[S_RR];
         const double beta = rr_new / rr;
         for (long i = tid; i < nrows; i += nth)
-            p[i] = beta * p[i] + r[i];
+            st_sc1_f64(p + i, beta * p[i] + r[i]);
         rr = rr_new;
         ++k;
         if (rtol2 > 0.0 && rr <= rtol2) converged = 1;
     }
     if (tid == 0) {
-        scal[S_RR] = rr;
+        st_sc1_f64(scal + S_RR, rr);
         out2[0] = k;
         out2[1] = alive ? converged : -1;
     }
