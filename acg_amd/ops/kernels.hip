@@ -901,7 +901,7 @@ __device__ __forceinline__ bool grid_barrier(unsigned* state) {
             while (__hip_atomic_load(mygen, RLX_AGENT) == g) {
                 if (spins < 32) __builtin_amdgcn_s_sleep(2);
                 else __builtin_amdgcn_s_sleep(64);
-                if (++spins > 100000000u) {  // bounded: never hang the GPU
+                if (++spins > 4000000u) {  // ~7 s worst case: fail fast, never hang
                     __hip_atomic_store(fail, 1u, RLX_AGENT);
                     ok = false;
                     break;
@@ -1282,6 +1282,11 @@ int cg_device(long nslices, long nrows, uintptr_t sellptr, uintptr_t cols,
     hipError_t oe = hipOccupancyMaxActiveBlocksPerMultiprocessor(
         &blocks_per_cu, kern, BLOCK, 0);
     if (oe != hipSuccess || blocks_per_cu < 1) blocks_per_cu = 1;
+    // the occupancy API can over-report by one block/CU and the cooperative
+    // launch validates against the same (wrong) number -- a non-resident
+    // block would deadlock the grid barrier.  Guide guidance: <= 4 blocks
+    // of 256 threads per CU is reliably resident.
+    if (blocks_per_cu > 4) blocks_per_cu = 4;
     long grid = (long)props.multiProcessorCount * blocks_per_cu;
     long need = (nslices * WAVE + BLOCK - 1) / BLOCK;
     if (grid > need) grid = need;
