@@ -177,6 +177,7 @@ def main(argv=None) -> int:
     collective_raise(comm, err)
 
     # ---- scatter (reference acgsymcsrmatrix_scatter, acg-hip.c:1752)
+    t0 = time.perf_counter()
     if comm:
         S = comm.scatter_object(systems)
         b_local, x0_local = comm.scatter_object(b_pieces)
@@ -184,6 +185,9 @@ def main(argv=None) -> int:
     else:
         S = systems[0]
         b_local, x0_local = b_pieces[0]
+    log(f"scattered subdomains ({time.perf_counter() - t0:.2f}s)")
+    if args.verbose and args.verbose > 1:
+        S.dump(file=sys.stderr)
 
     if args.output_comm_matrix and comm:
         counts = comm.gather_object(
@@ -219,8 +223,14 @@ def main(argv=None) -> int:
         elif gpu_solver:
             from .solvers.hip import CGSolverHIP
 
+            t0 = time.perf_counter()
             solver = CGSolverHIP(S, comm=comm, device=device,
                                  profile=args.profile)
+            fmt = ("Block-SELL" if solver.bsell is not None else
+                   "sigma-SELL" if solver.sell_perm is not None else
+                   "SELL" if solver.sell is not None else "CSR-vector")
+            log(f"GPU solver init: operator format {fmt} "
+                f"({time.perf_counter() - t0:.2f}s)")
             b = b.to(device)
             x = x.to(device)
             if args.warmup:

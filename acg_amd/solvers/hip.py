@@ -217,13 +217,12 @@ class CGSolverHIP:
     # -- classic CG -------------------------------------------------------
 
     def solve(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
-              res_atol: float = 0.0, res_rtol: float = 1e-9,
-              check_every: int = 1) -> SolveResult:
+              res_atol: float = 0.0, res_rtol: float = 1e-9) -> SolveResult:
         """Classic CG (reference acgsolverhip_solvempi, cghip.c:402-1159).
 
         ``x`` must be an nlocal vector (ghost tail included); ``b`` nowned.
-        ``check_every``: host convergence-test cadence (1 = reference
-        behavior).
+        The host convergence test runs every iteration on the lag-1
+        pipeline (see solve_pipelined).
         """
         res = SolveResult(solver="cg-hip", maxits=maxits, res_atol=res_atol,
                           res_rtol=res_rtol,
@@ -267,7 +266,7 @@ class CGSolverHIP:
                    for _ in range(LAG + 1)]
         evdone = [torch.cuda.Event() for _ in range(LAG + 1)]
 
-        def body(first_capture=False):
+        def body():
             # zero the fused (p,t) accumulator, then halo+split SpMV with
             # the (p,t) reduction fused into both SpMV passes
             S.cg_prep_pt(scal)
@@ -396,7 +395,7 @@ class CGSolverHIP:
 
     def solve_pipelined(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
                         res_atol: float = 0.0, res_rtol: float = 1e-9,
-                        check_every: int = 1, use_graph: bool = True,
+                        use_graph: bool = True,
                         megafuse: bool | None = None) -> SolveResult:
         """Pipelined (Ghysels-Vanroose) CG: ONE 2-double allreduce per
         iteration, overlapped with the halo + SpMV of q = A w
