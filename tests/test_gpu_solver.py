@@ -177,6 +177,27 @@ def test_pipelined_megafused_matches_fallback(problem):
     torch.testing.assert_close(x1[:S.nowned], x2[:S.nowned], rtol=1e-6, atol=1e-8)
 
 
+def test_no_memory_growth_across_solves(problem):
+    """Repeated solves (each capturing fresh hipGraphs) must not leak."""
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A, S = problem
+    rng = np.random.default_rng(23)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+    gpu = CGSolverHIP(S, device="cuda:0")
+    x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
+    for _ in range(3):  # warm the pools
+        gpu.solve_pipelined(b, x.clone(), maxits=10, res_rtol=0.0)
+    torch.cuda.synchronize()
+    base = torch.cuda.memory_allocated()
+    for _ in range(10):
+        gpu.solve_pipelined(b, x.clone(), maxits=10, res_rtol=0.0)
+        gpu.solve(b, x.clone(), maxits=10, res_rtol=0.0)
+    torch.cuda.synchronize()
+    grown = torch.cuda.memory_allocated() - base
+    assert grown < 32 << 20, f"leaked {grown} bytes across 20 solves"
+
+
 def test_profile_mode(problem):
     """hipEvent per-op profiling produces sane per-op stats on GPU."""
     from acg_amd.solvers.hip import CGSolverHIP
