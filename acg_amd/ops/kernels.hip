@@ -359,12 +359,27 @@ k_zero_scalars(double* scal, int i0, int count) {
     for (int i = threadIdx.x; i < count; i += BLOCK) scal[i0 + i] = 0.0;
 }
 
-// before the classic halo/SpMV: zero the (p,t) accumulator
+// before the classic halo/SpMV: zero the (p,t) accumulator (only needed
+// on the CSR fallback path; the SELL/BSELL matA finalize overwrites)
 __global__ void k_cg_prep_pt(double* scal) { if (threadIdx.x == 0) scal[S_PT] = 0.0; }
 
 // after allreduce(p,t): save rr for the device-side alpha
 __global__ void k_cg_prep_rr(double* scal) {
     if (threadIdx.x == 0) scal[S_RR_PREV] = scal[S_RR];
+}
+
+// classic-iteration epilogue: rotate rr -> rr_prev, then publish the new
+// (r,r) from the update kernel's partials (one 1-block launch instead of
+// prep_rr + reduce_partials)
+__global__ void __launch_bounds__(BLOCK)
+k_cg_finalize(const double* __restrict__ partials, int nblocks, double* scal) {
+    double v = 0.0;
+    for (int i = threadIdx.x; i < nblocks; i += BLOCK) v += partials[i];
+    v = block_reduce(v);
+    if (threadIdx.x == 0) {
+        scal[S_RR_PREV] = scal[S_RR];
+        scal[S_RR] = v;
+    }
 }
 
 // dot: partials[b] = block sum of x[i]*y[i]
@@ -437,7 +452,8 @@ k_daypx_ratio(double* __restrict__ y, const double* __restrict__ x, long n,
         y[i] = b * y[i] + x[i];
 }
 
-// classic-CG fused update: alpha = rr_prev/pt (device);
+// classic-CG fused update: alpha = rr/pt (device; S_RR still holds the
+// current rr -- k_cg_finalize rotates it afterwards);
 //   r -= alpha*t;  x += alpha*p;  partials[b] = block sum of new r.r
 // One pass over r,t,x,p instead of three kernels + a dot
 // (reference: daxpy_minus_alpha + daxpy_alpha + Ddot, cghip.c:969-1026).
@@ -446,7 +462,7 @@ k_cg_fused_update(double* __restrict__ r, double* __restrict__ x,
                   const double* __restrict__ p, const double* __restrict__ t,
                   long n, const double* __restrict__ scal,
                   double* __restrict__ partials) {
-    const double alpha = scal[S_RR_PREV] / scal[S_PT];
+    const double alpha = scal[S_RR] / scal[S_PT];
     double acc = 0.0;
     const long stride = (long)gridDim.x * BLOCK;
     // t and x are single-use streams this iteration: non-temporal keeps
@@ -1210,7 +1226,9 @@ void cg_fused_update(uintptr_t r, uintptr_t x, uintptr_t p, uintptr_t t, long n,
                        (double*)r, (double*)x, (const double*)p, (const double*)t,
                        n, (const double*)scal, (double*)partials);
     check_hip("cg_fused_update");
-    reduce_partials(partials, (int)blocks, scal, S_RR, false, stream);
+    hipLaunchKernelGGL(k_cg_finalize, dim3(1), dim3(BLOCK), 0, S(stream),
+                       (const double*)partials, (int)blocks, (double*)scal);
+    check_hip("cg_finalize");
 }
 
 long sell_pipe(long nslices, long nrows_pass, long rowbase, long border_base,

@@ -234,9 +234,12 @@ def daypx_ratio(y, x, scal, num, den, n=None) -> None:
 
 
 def cg_fused_update(r, x, p, t, scal, partials, n) -> None:
-    alpha = float(scal[S_RR_PREV]) / float(scal[S_PT])
+    """alpha = rr/pt; update r,x; then rotate rr->rr_prev and publish the
+    new (r,r) (matches k_cg_fused_update + k_cg_finalize)."""
+    alpha = float(scal[S_RR]) / float(scal[S_PT])
     r[:n] -= alpha * t[:n]
     x[:n] += alpha * p[:n]
+    scal[S_RR_PREV] = scal[S_RR].clone()
     scal[S_RR] = torch.dot(r[:n], r[:n])
 
 

@@ -172,17 +172,21 @@ class CGSolverHIP:
                     scal=self.scal if fuse_dotslot >= 0 else None,
                     dotslot=fuse_dotslot)
         with self.prof.span("spmvA"):
+            # the matA pass OVERWRITES the dot slot (dot_accum=False), so
+            # no zeroing prep kernel is needed; matO accumulates on top
             if self.bsell is not None:
                 bptr, bcol, bvals, dof = self.bsell
                 ops.spmv_bsell(bptr, bcol, bvals, self.n // dof, dof,
-                               xfull, y, **fuse)
+                               xfull, y, dot_accum=False, **fuse)
             elif self.sell is not None:
                 sellptr, scols, svals = self.sell
                 ops.spmv_sell(sellptr, scols, svals, self.n, xfull, y,
-                              accum=False, perm=self.sell_perm, **fuse)
+                              accum=False, perm=self.sell_perm,
+                              dot_accum=False, **fuse)
             else:
                 ops.spmv(self.A_rowptr, self.A_colidx, self.A_vals, xfull, y,
-                         lanes=self.lanesA, accum=False, **fuse)
+                         lanes=self.lanesA, accum=False, dot_accum=False,
+                         **fuse)
         if have_halo:
             with torch.cuda.stream(self.comm_stream):
                 self.halo.end()
@@ -267,14 +271,13 @@ class CGSolverHIP:
         evdone = [torch.cuda.Event() for _ in range(LAG + 1)]
 
         def body():
-            # zero the fused (p,t) accumulator, then halo+split SpMV with
-            # the (p,t) reduction fused into both SpMV passes
-            S.cg_prep_pt(scal)
+            # halo+split SpMV with the (p,t) reduction fused into both
+            # passes (matA overwrites the slot, matO accumulates)
             self._spmv_overlapped(p, t, fuse_dotslot=S.S_PT)
             with self.prof.span("allreduce"):
                 self._allreduce_slot(S.S_PT)
-            # rr_prev = rr; then fused r/x update + finalized (r,r)
-            S.cg_prep_rr(scal)
+            # fused r/x update (alpha = rr/pt on device) + combined
+            # finalize (rr -> rr_prev rotation + new (r,r))
             with self.prof.span("update_classic"):
                 S.cg_fused_update(r, x, p, t, scal, self.partials, n)
             with self.prof.span("allreduce"):
