@@ -305,6 +305,11 @@ class CGSolverHIP:
         while k < maxits:
             if k >= LAG and check(k - LAG):
                 break
+            if k > 0:
+                # this body's finalize overwrites S_RR, which the previous
+                # iteration's lagged D2H may still be reading: order after it
+                cur0 = torch.cuda.current_stream(self.device)
+                cur0.wait_event(evdone[(k - 1) % (LAG + 1)])
             if graph is not None:
                 graph.replay()
             else:
@@ -559,6 +564,10 @@ class CGSolverHIP:
                     evdone[j].record(self.copy_stream)
             else:
                 issue_gamma_copy(k)
+            # the body's finalize overwrites gamma/delta: order it after the
+            # in-flight 8-byte copy (costs ~us, preserves the SpMV overlap)
+            torch.cuda.current_stream(self.device).wait_event(
+                evdone[k % (LAG + 1)])
             if mega:
                 wa, wb = (w, w2) if k % 2 == 0 else (w2, w)
                 g = graphs[k % 2]
