@@ -17,6 +17,20 @@ def _build(nx=32, ny=32, nparts=1, spec=STENCIL_5PT_2D, gz=1, method="block"):
     return A, systems
 
 
+def test_baseline_config1_poisson128():
+    """BASELINE.json config 1: 5-pt 2D Poisson 128x128, classic CG, CPU,
+    1 rank (the plumbing check, verbatim)."""
+    A, systems = _build(128, 128, 1)
+    S = systems[0]
+    rng = np.random.default_rng(0)
+    b = torch.from_numpy(rng.standard_normal(A.n))
+    x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64)
+    solver = CGSolverCPU(S)
+    res = solver.solve(b, x, maxits=2000, res_rtol=1e-9)
+    assert res.converged, res.summary()
+    assert res.rnrm2 <= 1e-9 * res.bnrm2 * 1.01
+
+
 def test_poisson_cg_serial_vs_scipy():
     A, systems = _build(32, 32, 1)
     S = systems[0]
