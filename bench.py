@@ -30,11 +30,13 @@ def main() -> int:
     ap.add_argument("--steps", type=int, default=200)
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--solver", choices=["auto", "pipelined", "classic"],
-                    default="pipelined",
-                    help="pipelined (default): megafused on narrow rows, one "
-                         "overlapped allreduce multi-GPU; classic is within "
-                         "noise of it on 1 GPU (box-to-box variance ~5%% "
-                         "dominates)")
+                    default="auto",
+                    help="auto (measured, tools/solver_bench.py interleaved): "
+                         "1 GPU wide rows -> classic (589 vs 605 us/it on "
+                         "Queen: 9n vs 13n update traffic); 1 GPU narrow "
+                         "rows -> megafused pipelined (2011 vs 2286 us/it on "
+                         "Poisson 384^3); multi-GPU -> pipelined (single "
+                         "overlapped allreduce)")
     ap.add_argument("--config", choices=["queen", "flan", "poisson7"], default="queen",
                     help="queen: 27-pt dof-3 Queen_4147-shaped (BASELINE configs 3-4); "
                          "flan: Flan_1565-shaped, 1.56M rows (BASELINE config 2); "
@@ -96,7 +98,10 @@ def main() -> int:
 
     solver_kind = args.solver
     if solver_kind == "auto":
-        solver_kind = "classic" if ngpus == 1 else "pipelined"
+        if ngpus > 1 or solver.megafuse_auto:
+            solver_kind = "pipelined"
+        else:
+            solver_kind = "classic"
     solve = solver.solve_pipelined if solver_kind == "pipelined" else solver.solve
 
     # warmup (untimed; also JITs RCCL channels and fills caches)
