@@ -248,6 +248,67 @@ py::tuple coo_to_sym_csr(i64 n,
     return py::make_tuple(rowptr, colidx, v);
 }
 
+// ---------------------------------------------------------------------------
+// Block-structure analysis for Block-SELL conversion: a node = dof
+// consecutive rows; its block columns are the union of the rows' col/dof
+// values.  Rows are sorted by column, so this is a dof-way merge --
+// O(nnz), OpenMP-parallel over nodes (the numpy fallback is an
+// O(nnz log nnz) single-threaded unique).
+// Returns (blocks_per_node i64[nnodes], bcols i64[nblocks] node-major
+// ascending, entry_block i64[nnz] = global block index of each entry).
+py::tuple bsell_blocks(py::array_t<i64, py::array::c_style | py::array::forcecast> rowptr,
+                       py::array_t<i64, py::array::c_style | py::array::forcecast> colidx,
+                       int dof) {
+    const i64* rp = rowptr.data();
+    const i64* ci = colidx.data();
+    const i64 nrows = (i64)rowptr.shape(0) - 1;
+    if (dof < 2 || dof > 8 || nrows % dof)
+        throw std::runtime_error("bsell_blocks: bad dof");
+    const i64 nnodes = nrows / dof;
+    const i64 nnz = rp[nrows];
+    py::array_t<i64> counts_a(nnodes);
+    i64* counts = counts_a.mutable_data();
+
+    auto merge_node = [&](i64 nd, i64* out_bcols, i64* entry_block, i64 blockbase) -> i64 {
+        i64 pos[8], end[8];
+        for (int r = 0; r < dof; ++r) {
+            pos[r] = rp[nd * dof + r];
+            end[r] = rp[nd * dof + r + 1];
+        }
+        i64 nb = 0;
+        for (;;) {
+            i64 bc = INT64_MAX;
+            for (int r = 0; r < dof; ++r)
+                if (pos[r] < end[r]) bc = std::min(bc, ci[pos[r]] / dof);
+            if (bc == INT64_MAX) break;
+            if (out_bcols) out_bcols[nb] = bc;
+            for (int r = 0; r < dof; ++r)
+                while (pos[r] < end[r] && ci[pos[r]] / dof == bc) {
+                    if (entry_block) entry_block[pos[r]] = blockbase + nb;
+                    ++pos[r];
+                }
+            ++nb;
+        }
+        return nb;
+    };
+
+    #pragma omp parallel for schedule(static)
+    for (i64 nd = 0; nd < nnodes; ++nd)
+        counts[nd] = merge_node(nd, nullptr, nullptr, 0);
+    std::vector<i64> starts(nnodes + 1);
+    starts[0] = 0;
+    for (i64 nd = 0; nd < nnodes; ++nd) starts[nd + 1] = starts[nd] + counts[nd];
+    const i64 nblocks = starts[nnodes];
+    py::array_t<i64> bcols_a(nblocks);
+    py::array_t<i64> entry_a(nnz);
+    i64* bcols = bcols_a.mutable_data();
+    i64* entry = entry_a.mutable_data();
+    #pragma omp parallel for schedule(static)
+    for (i64 nd = 0; nd < nnodes; ++nd)
+        merge_node(nd, bcols + starts[nd], entry, starts[nd]);
+    return py::make_tuple(counts_a, bcols_a, entry_a);
+}
+
 // in-place scans (reference acgprefixsum_inplace_*, prefixsum.h:72-116)
 py::array_t<i64> prefix_sum(py::array_t<i64, py::array::c_style | py::array::forcecast> a,
                             bool inclusive) {
@@ -279,5 +340,6 @@ PYBIND11_MODULE(_acg_host, m) {
           py::arg("eps") = 0.0, py::arg("col32") = true);
     m.def("coo_to_sym_csr", &coo_to_sym_csr);
     m.def("prefix_sum", &prefix_sum, py::arg("a"), py::arg("inclusive") = true);
+    m.def("bsell_blocks", &bsell_blocks);
     m.def("num_threads", &num_threads);
 }

@@ -127,16 +127,23 @@ def bsell_from_csr(rowptr, colidx, vals, dof: int, C: int = 64):
     nnodes = nrows // dof
     nnz = len(colidx)
     rows = np.repeat(np.arange(nrows, dtype=np.int64), np.diff(rowptr))
-    node = rows // dof
-    bcol_of_entry = colidx // dof
-    # unique (node, blockcol) pairs -> per-node block lists
-    key = node * (np.int64(1) << 32) | bcol_of_entry.astype(np.int64)
-    uk, inv = np.unique(key, return_inverse=True)
-    nblocks = len(uk)
-    bnode = (uk >> 32).astype(np.int64)
-    bcols = (uk & 0xFFFFFFFF).astype(np.int64)
+    try:
+        from ..host import _acg_host as H
+
+        blocks_per_node, bcols, inv = (np.asarray(a) for a in
+                                       H.bsell_blocks(rowptr, colidx, dof))
+        nblocks = len(bcols)
+        bnode = np.repeat(np.arange(nnodes, dtype=np.int64), blocks_per_node)
+    except ImportError:  # numpy fallback: O(nnz log nnz) unique
+        node = rows // dof
+        bcol_of_entry = colidx // dof
+        key = node * (np.int64(1) << 32) | bcol_of_entry.astype(np.int64)
+        uk, inv = np.unique(key, return_inverse=True)
+        nblocks = len(uk)
+        bnode = (uk >> 32).astype(np.int64)
+        bcols = (uk & 0xFFFFFFFF).astype(np.int64)
+        blocks_per_node = np.bincount(bnode, minlength=nnodes)
     density = nnz / (nblocks * dof * dof)
-    blocks_per_node = np.bincount(bnode, minlength=nnodes)
     nslices = (nnodes + C - 1) // C
     padlen = np.zeros(nslices * C, dtype=np.int64)
     padlen[:nnodes] = blocks_per_node
