@@ -9,7 +9,6 @@ performance path)."""
 
 from __future__ import annotations
 
-import math
 import time
 
 import numpy as np

@@ -12,7 +12,6 @@ prints per-solver medians.
 import argparse
 import statistics
 import sys
-import time
 from pathlib import Path
 
 import numpy as np
