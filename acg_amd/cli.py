@@ -59,8 +59,9 @@ def make_parser() -> argparse.ArgumentParser:
     p.add_argument("--residual-rtol", type=float, default=1e-9)
     p.add_argument("--epsilon", type=float, default=0.0,
                    help="diagonal shift added to A")
-    p.add_argument("--warmup", type=int, default=0,
-                   help="untimed warmup iterations before the timed solve")
+    p.add_argument("--warmup", type=int, default=10,
+                   help="untimed warmup iterations before the timed solve "
+                        "(reference default 10, acg-hip.c:480-486)")
     p.add_argument("--comm", choices=("none", "rccl", "gloo"), default=None,
                    help="default: rccl when WORLD_SIZE>1 and GPUs exist")
     p.add_argument("--manufactured-solution", action="store_true",
