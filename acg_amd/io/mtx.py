@@ -122,6 +122,8 @@ def _parse_body_text(text: bytes | str, ncols_expected: int) -> np.ndarray:
     """Parse whitespace-separated numeric rows into a 2-D float64 array."""
     if isinstance(text, bytes):
         text = text.decode()
+    if not text.strip():
+        return np.zeros((0, max(ncols_expected, 1)), dtype=np.float64)
     try:
         import pandas as pd
 

@@ -156,3 +156,34 @@ def test_cli_scipy_solver(tmp_path, capsys, monkeypatch):
     out = capsys.readouterr()
     assert rc == 0, out.err
     assert "scipy-cg" in out.err
+
+
+def test_mtx_edge_formats(tmp_path):
+    """Empty matrices, vector-coordinate, pattern, complex, comments."""
+    import numpy as np
+
+    from acg_amd.io.mtx import MtxFile, read_mtx, write_mtx
+
+    m = MtxFile(object="matrix", format="coordinate", field_="real",
+                symmetry="symmetric", nrows=3, ncols=3, nnz=0,
+                rowidx=np.zeros(0, np.int64), colidx=np.zeros(0, np.int64),
+                a=np.zeros(0))
+    p = tmp_path / "empty.mtx"
+    write_mtx(p, m)
+    assert read_mtx(p).nnz == 0
+    (tmp_path / "v.mtx").write_text(
+        "%%MatrixMarket vector coordinate real general\n5 2\n1 3.5\n4 -1.25\n")
+    mv = read_mtx(tmp_path / "v.mtx")
+    np.testing.assert_array_equal(mv.rowidx, [0, 3])
+    (tmp_path / "p.mtx").write_text(
+        "%%MatrixMarket matrix coordinate pattern symmetric\n3 3 2\n1 1\n3 2\n")
+    mp_ = read_mtx(tmp_path / "p.mtx")
+    np.testing.assert_array_equal(mp_.a, [1.0, 1.0])
+    (tmp_path / "c.mtx").write_text(
+        "%%MatrixMarket matrix coordinate complex general\n2 2 1\n1 2 1.5 -0.5\n")
+    mc = read_mtx(tmp_path / "c.mtx")
+    assert mc.a.dtype == np.complex128 and mc.a[0] == 1.5 - 0.5j
+    (tmp_path / "h.mtx").write_text(
+        "%%MatrixMarket matrix coordinate real general\n% c1\n%c2\n2 2 1\n1 1 2.0\n")
+    mh = read_mtx(tmp_path / "h.mtx")
+    assert len(mh.comments) == 2 and mh.a[0] == 2.0
