@@ -302,14 +302,10 @@ class CGSolverHIP:
             return False
 
         k = 0
+        cur = torch.cuda.current_stream(self.device)
         while k < maxits:
             if k >= LAG and check(k - LAG):
                 break
-            if k > 0:
-                # this body's finalize overwrites S_RR, which the previous
-                # iteration's lagged D2H may still be reading: order after it
-                cur0 = torch.cuda.current_stream(self.device)
-                cur0.wait_event(evdone[(k - 1) % (LAG + 1)])
             if graph is not None:
                 graph.replay()
             else:
@@ -318,14 +314,14 @@ class CGSolverHIP:
                     graph = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(graph):
                         body()
-            # issue the lagged rr D2H (after this iteration's update)
-            cur = torch.cuda.current_stream(self.device)
-            self._ev_rr.record(cur)
-            self.copy_stream.wait_event(self._ev_rr)
+            # lagged rr D2H on the SAME stream: cross-stream event chains
+            # measured ~20 us each in fire->launch latency (45 us/iter of
+            # the iteration period); the in-order 8-byte copy costs ~3 us
+            # on the critical path and makes the overwrite race impossible
+            # by stream ordering.
             j = k % (LAG + 1)
-            with torch.cuda.stream(self.copy_stream):
-                hostbuf[j].copy_(scal[S.S_RR:S.S_RR + 1], non_blocking=True)
-                evdone[j].record(self.copy_stream)
+            hostbuf[j].copy_(scal[S.S_RR:S.S_RR + 1], non_blocking=True)
+            evdone[j].record(cur)
             k += 1
             res.niterations = k
         if not converged:
@@ -501,13 +497,13 @@ class CGSolverHIP:
         evdone = [torch.cuda.Event() for _ in range(LAG + 1)]
 
         def issue_gamma_copy(k):
+            # same-stream D2H (cross-stream event chains cost ~20 us each
+            # in fire->launch latency; the in-order 8-byte copy ~3 us) --
+            # also makes the gamma-overwrite race impossible by ordering
             cur = torch.cuda.current_stream(self.device)
-            self._ev_rr.record(cur)
-            self.copy_stream.wait_event(self._ev_rr)
             j = k % (LAG + 1)
-            with torch.cuda.stream(self.copy_stream):
-                hostbuf[j].copy_(scal[S.S_GAMMA:S.S_GAMMA + 1], non_blocking=True)
-                evdone[j].record(self.copy_stream)
+            hostbuf[j].copy_(scal[S.S_GAMMA:S.S_GAMMA + 1], non_blocking=True)
+            evdone[j].record(cur)
 
         def read_gamma(j):
             evdone[j % (LAG + 1)].synchronize()
@@ -562,12 +558,11 @@ class CGSolverHIP:
                     hostbuf[j].copy_(scal[S.S_GAMMA:S.S_GAMMA + 1],
                                      non_blocking=True)
                     evdone[j].record(self.copy_stream)
+                # overlap_ar: the body's finalize overwrites gamma/delta;
+                # order it after the (side-stream) copy
+                torch.cuda.current_stream(self.device).wait_event(evdone[j])
             else:
-                issue_gamma_copy(k)
-            # the body's finalize overwrites gamma/delta: order it after the
-            # in-flight 8-byte copy (costs ~us, preserves the SpMV overlap)
-            torch.cuda.current_stream(self.device).wait_event(
-                evdone[k % (LAG + 1)])
+                issue_gamma_copy(k)  # same-stream: ordered by construction
             if mega:
                 wa, wb = (w, w2) if k % 2 == 0 else (w2, w)
                 g = graphs[k % 2]
