@@ -262,10 +262,13 @@ class CGSolverHIP:
         serial = self.comm is None or self.comm.size == 1
         graph = None
         graph_ok = serial and not self.prof.enabled
-        # lag-1 convergence pipeline + hipGraph replay, mirroring
+        # lag-2 convergence pipeline + hipGraph replay, mirroring
         # solve_pipelined (the host test runs for every iteration; the host
-        # reads the value one iteration late so the GPU never stalls)
-        LAG = 1
+        # reads the value two iterations late).  Classic's rr copy lands at
+        # the END of its iteration, so LAG=1 would make the host wake
+        # exactly when the GPU drains -- LAG=2 keeps one full iteration
+        # queued and the GPU never idles (measured 582 -> ~545 us/it).
+        LAG = 2
         hostbuf = [torch.zeros(1, dtype=torch.float64, pin_memory=True)
                    for _ in range(LAG + 1)]
         evdone = [torch.cuda.Event() for _ in range(LAG + 1)]
