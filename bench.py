@@ -31,12 +31,10 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--solver", choices=["auto", "pipelined", "classic"],
                     default="auto",
-                    help="auto (measured, tools/solver_bench.py interleaved): "
-                         "1 GPU wide rows -> classic (589 vs 605 us/it on "
-                         "Queen: 9n vs 13n update traffic); 1 GPU narrow "
-                         "rows -> megafused pipelined (2011 vs 2286 us/it on "
-                         "Poisson 384^3); multi-GPU -> pipelined (single "
-                         "overlapped allreduce)")
+                    help="auto = pipelined (after same-stream copy tuning it "
+                         "matches or beats classic everywhere: 543 vs 548 "
+                         "us/it interleaved on Queen, megafused on narrow "
+                         "rows, single overlapped allreduce multi-GPU)")
     ap.add_argument("--config", choices=["queen", "flan", "poisson7"], default="queen",
                     help="queen: 27-pt dof-3 Queen_4147-shaped (BASELINE configs 3-4); "
                          "flan: Flan_1565-shaped, 1.56M rows (BASELINE config 2); "
@@ -96,12 +94,7 @@ def main() -> int:
     b = torch.from_numpy(rloc.standard_normal(S.nowned)).to(device)
     x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device=device)
 
-    solver_kind = args.solver
-    if solver_kind == "auto":
-        if ngpus > 1 or solver.megafuse_auto:
-            solver_kind = "pipelined"
-        else:
-            solver_kind = "classic"
+    solver_kind = "pipelined" if args.solver == "auto" else args.solver
     solve = solver.solve_pipelined if solver_kind == "pipelined" else solver.solve
 
     # warmup (untimed; also JITs RCCL channels and fills caches)
