@@ -1,0 +1,8 @@
+"""`python -m acg_amd` == the aCG-compatible CLI driver."""
+
+import sys
+
+from .cli import main
+
+if __name__ == "__main__":
+    sys.exit(main())
