@@ -555,15 +555,14 @@ class CGSolverHIP:
                 break
             if overlap_ar:
                 # D2H of gamma chains off the allreduce, not the main stream
+                # (the SpMV below must NOT wait for it -- that is the
+                # overlap; only the fused update is ordered after it)
                 j = k % (LAG + 1)
                 self.copy_stream.wait_event(ev_ar)
                 with torch.cuda.stream(self.copy_stream):
                     hostbuf[j].copy_(scal[S.S_GAMMA:S.S_GAMMA + 1],
                                      non_blocking=True)
                     evdone[j].record(self.copy_stream)
-                # overlap_ar: the body's finalize overwrites gamma/delta;
-                # order it after the (side-stream) copy
-                torch.cuda.current_stream(self.device).wait_event(evdone[j])
             else:
                 issue_gamma_copy(k)  # same-stream: ordered by construction
             if mega:
@@ -583,8 +582,12 @@ class CGSolverHIP:
             else:
                 self._spmv_overlapped(w, q)
                 if overlap_ar:
-                    # the fused update reads gamma/delta: wait the allreduce
-                    torch.cuda.current_stream(self.device).wait_event(ev_ar)
+                    # the fused update reads gamma/delta (allreduce) and its
+                    # finalize overwrites them (in-flight D2H copy): order
+                    # after both
+                    cur2 = torch.cuda.current_stream(self.device)
+                    cur2.wait_event(ev_ar)
+                    cur2.wait_event(evdone[k % (LAG + 1)])
                 with self.prof.span("update"):
                     S.pipelined_fused(z, t, p, x, r, w, q, scal, self.partials,
                                       n, first)
