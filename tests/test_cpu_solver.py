@@ -215,3 +215,44 @@ def test_slab_halo_pairing():
             lo = int(R.halo.rdispls[si])
             hi = lo + int(R.halo.recvcounts[si])
             np.testing.assert_array_equal(sent_globals, R.ghost_global[lo:hi])
+
+
+def test_slab_generator_equals_generic_extractor():
+    """The analytic slab generator and the generic extractor are two
+    independent constructions of the same partition: layouts, halos and
+    operators must agree EXACTLY on slab-aligned partitions."""
+    import scipy.sparse as sp
+
+    from acg_amd.gen import stencil_local_slab
+    from acg_amd.part import extract_subdomains
+
+    spec = queen_like_spec(3)
+    gx, gy, gz, R, dof = 5, 5, 8, 2, 3
+    A = stencil_global(gx, gy, gz, spec)
+    plane = gx * gy
+    part = np.zeros(A.n, dtype=np.int32)
+    for r in range(R):
+        z0, z1 = gz * r // R, gz * (r + 1) // R
+        part[z0 * plane * dof:z1 * plane * dof] = r
+    gen = [stencil_local_slab(gx, gy, gz, spec, r, R) for r in range(R)]
+    ext = extract_subdomains(A, part, R)
+    for g, e in zip(gen, ext):
+        assert (g.nowned, g.ninterior, g.nborder, g.nghost) == \
+               (e.nowned, e.ninterior, e.nborder, e.nghost)
+        np.testing.assert_array_equal(g.owned_global, e.owned_global)
+        np.testing.assert_array_equal(g.ghost_global, e.ghost_global)
+        np.testing.assert_array_equal(g.halo.senders, e.halo.senders)
+        np.testing.assert_array_equal(g.halo.sendidx, e.halo.sendidx)
+
+        def toop(S):
+            a = sp.csr_matrix((S.A_vals, S.A_colidx.astype(np.int64),
+                               S.A_rowptr),
+                              shape=(S.nowned, S.nowned + S.nghost))
+            if S.nnzO:
+                o = sp.csr_matrix((S.O_vals, S.O_colidx.astype(np.int64),
+                                   S.O_rowptr),
+                                  shape=(S.nborder, S.nowned + S.nghost))
+                return sp.vstack([a[:S.ninterior], a[S.ninterior:] + o]).tocsr()
+            return a
+
+        assert abs(toop(g) - toop(e)).max() < 1e-14
