@@ -139,6 +139,26 @@ def test_device_generation_matches_host(specname, g, nranks_sim):
                                       np.asarray(H.halo.sendidx, dtype=np.int64))
 
 
+def test_csr_vector_fallback_path(problem):
+    """Forced CSR-vector operator (use_sell=False): both solvers still
+    converge and agree with the SELL/BSELL path."""
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A, S = problem
+    rng = np.random.default_rng(31)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+    fast = CGSolverHIP(S, device="cuda:0")
+    csr = CGSolverHIP(S, device="cuda:0", use_sell=False)
+    assert csr.sell is None and csr.A_rowptr is not None
+    x1 = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
+    r1 = fast.solve(b, x1, maxits=400, res_rtol=1e-10)
+    x2 = torch.zeros_like(x1)
+    r2 = csr.solve(b, x2, maxits=400, res_rtol=1e-10)
+    assert r1.converged and r2.converged
+    assert abs(r1.niterations - r2.niterations) <= 2
+    torch.testing.assert_close(x1[:S.nowned], x2[:S.nowned], rtol=1e-6, atol=1e-8)
+
+
 def test_pipelined_graph_matches_eager(problem):
     """hipGraph-replayed pipelined CG == eager pipelined CG."""
     from acg_amd.solvers.hip import CGSolverHIP
