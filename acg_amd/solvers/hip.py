@@ -344,7 +344,11 @@ class CGSolverHIP:
         if self.prof.enabled:
             from .profiling import annotate_op_stats
 
-            annotate_op_stats(res, self.local, self.prof.collect())
+            idxb = None
+            if self.bsell is not None:
+                idxb = 4.0 / (self.bsell[3] ** 2)
+            annotate_op_stats(res, self.local, self.prof.collect(),
+                              idx_bytes_per_nnz=idxb)
             self.prof.reset()
         return res
 
@@ -618,6 +622,10 @@ class CGSolverHIP:
         if self.prof.enabled:
             from .profiling import annotate_op_stats
 
-            annotate_op_stats(res, self.local, self.prof.collect())
+            idxb = None
+            if self.bsell is not None:
+                idxb = 4.0 / (self.bsell[3] ** 2)
+            annotate_op_stats(res, self.local, self.prof.collect(),
+                              idx_bytes_per_nnz=idxb)
             self.prof.reset()
         return res

@@ -73,14 +73,21 @@ class _Span:
         return False
 
 
-def annotate_op_stats(res: SolveResult, local, ops_times: dict) -> None:
+def annotate_op_stats(res: SolveResult, local, ops_times: dict,
+                      idx_bytes_per_nnz: float | None = None) -> None:
     """Attach analytic flop/byte counts to measured per-op seconds
-    (reference: always-on counters, cghip.h:109-118)."""
+    (reference: always-on counters, cghip.h:109-118).
+
+    ``idx_bytes_per_nnz``: index bytes per nonzero of the operator format
+    actually used (4 for int32 CSR/SELL, 4/dof^2 for Block-SELL)."""
     n = local.nowned
     it = max(res.niterations, 1)
     nnzA, nnzO = local.nnzA, local.nnzO
-    _ac = getattr(local, "A_colidx", None)
-    colb = _ac.dtype.itemsize if _ac is not None and hasattr(_ac, "dtype") else 4
+    if idx_bytes_per_nnz is not None:
+        colb = idx_bytes_per_nnz
+    else:
+        _ac = getattr(local, "A_colidx", None)
+        colb = _ac.dtype.itemsize if _ac is not None and hasattr(_ac, "dtype") else 4
     model = {
         "spmvA": (2.0 * nnzA, nnzA * (8 + colb) + 16.0 * n),
         "spmvO": (2.0 * nnzO, nnzO * (8 + colb) + 16.0 * local.nborder),
