@@ -2,23 +2,32 @@
 //
 // Hand-written HIP replacing the reference's hipSPARSE/hipBLAS calls and its
 // CUDA-era kernels (reference: acg/cg-kernels-hip.hip, acg/halo-kernels-hip.hip,
-// acg/cghip.c:463-585).  Everything here is designed for CDNA4:
-//   - 64-wide wavefronts (one wave = one SELL slice; shuffle reductions over 64),
-//   - two sparse operator formats:
-//       * CSR vector kernel (LANES lanes/row) for irregular rows and matO,
-//       * SELL-C-64 (sliced ELLPACK, slice=wave) for regular rows: vals/cols
-//         stored column-major per 64-row slice so every wave load is one
-//         contiguous 512 B line set, one row per lane, no cross-lane reduce.
-//         Measured (Queen-shaped, ~81 nnz/row): CSR-vector ~3.7 TB/s; SELL
-//         removes the gather-side shuffle and the per-row pointer walk.
-//   - reductions produce ONE per-block partial (LDS + wave shuffle), summed by
-//     a tiny finalize kernel: no fp64 atomic contention (a 3000-block
-//     atomicAdd onto one cacheline measured 74 us for a 25 MB dot on this
-//     chip) and bitwise-deterministic results, unlike the reference's
-//     unsafeAtomicAdd dots (cg-kernels-hip.hip:1229-1286).
-//   - device-resident scalars: alpha/beta are computed on device from the
-//     scalar slab; the only per-iteration D2H is the 8-byte convergence norm
-//     (reference cghip.c:996-1001).
+// acg/cghip.c:463-585).  Contents, in file order:
+//   1. deterministic block reductions (partials + finalize; no fp64 atomics:
+//      a 3000-block atomicAdd onto one cacheline measured 74 us for a 25 MB
+//      dot on this chip, vs ~8 us for this scheme -- the reference's
+//      unsafeAtomicAdd dots, cg-kernels-hip.hip:1229-1286, are both slower
+//      and non-deterministic),
+//   2. sparse operators, chosen per matrix:
+//      * CSR vector kernel (4-64 lanes/row) for irregular rows and matO,
+//      * SELL-C-64 / sigma-SELL (sliced ELLPACK, slice = one 64-lane wave):
+//        one row per lane, vals/cols column-major per slice so every wave
+//        load is a contiguous 512 B line set; NT loads + unroll-8 variants,
+//      * Block-SELL for dense dof x dof block structure (FEM): one int32
+//        index per block, pair-major values (16 B/lane dwordx4 loads),
+//   3. fused CG kernels with device-resident scalars (alpha/beta computed
+//      in-kernel from the 8-slot slab; the only per-iteration D2H is the
+//      8-byte convergence norm): classic fused update + combined finalize,
+//      pipelined 6-vector update with BOTH next-iteration dots fused in,
+//      and the megafused single-kernel iteration (SpMV + update + dots,
+//      double-buffered w, split matA/matO passes for halo overlap),
+//   4. on-GPU stencil-operator generation (SELL/BSELL built directly in
+//      HBM at write speed; enables the 288 GB-per-GPU Poisson sizing),
+//   5. the monolithic device-side CG (whole solve in ONE cooperative
+//      launch) with a hand-rolled agent-scope grid barrier (sc1
+//      write-through payloads, per-XCD generation lines, parity counters,
+//      bounded spins -- cooperative grid.sync measured ~150 us/sync),
+//   6. halo pack gather (no unpack exists: ghosts are received in place).
 //
 // The scalar slab layout (fp64 slots) is shared with solvers/cg_hip.py:
 #define S_RR 0         // (r,r) current
