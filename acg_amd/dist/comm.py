@@ -66,6 +66,14 @@ class Comm:
         """
         if self.kind == "none" or self.size == 1:
             return t
+        if self.kind == "gloo" and t.is_cuda:
+            # CPU-staged path so the GPU solver's multi-rank code can be
+            # exercised end-to-end on a single GPU (gloo cannot carry CUDA
+            # tensors).  Test harness only -- production multi-GPU is RCCL.
+            h = t.detach().cpu()
+            self._dist.all_reduce(h, op=self._dist.ReduceOp.SUM)
+            t.copy_(h)
+            return t
         self._dist.all_reduce(t, op=self._dist.ReduceOp.SUM)
         return t
 

@@ -43,6 +43,8 @@ class HaloExchange:
         self.sendidx = idx.to(self.device).long()
         self.sendbuf = torch.empty(halo.sendsize, dtype=dtype, device=self.device)
         self._reqs: list = []
+        self._staged_recv: list = []
+        self._x = None
         # per-iteration traffic counters (reference cghip.h:109-118 stats)
         self.nexchanges = 0
         self.bytes_sent = 0
@@ -63,15 +65,27 @@ class HaloExchange:
             gpu_ops.pack_gather(self.sendbuf, x, self.sendidx)
         else:
             torch.index_select(x, 0, self.sendidx, out=self.sendbuf)
+        # gloo cannot carry CUDA tensors: stage through host buffers so the
+        # GPU solver's multi-rank path can be tested end-to-end on one GPU.
+        # Test harness only -- production multi-GPU runs RCCL (device direct).
+        staged = x.is_cuda and getattr(self.comm, "kind", "rccl") == "gloo"
+        sendsrc = self.sendbuf.cpu() if staged else self.sendbuf
+        self._staged_recv = []
         ops = []
         for i in range(h.nsenders):
             lo = self.nowned + int(h.rdispls[i])
             hi = lo + int(h.recvcounts[i])
-            ops.append(dist.P2POp(dist.irecv, x[lo:hi], int(h.senders[i])))
+            if staged:
+                buf = torch.empty(hi - lo, dtype=self.dtype)
+                self._staged_recv.append((buf, lo, hi))
+                ops.append(dist.P2POp(dist.irecv, buf, int(h.senders[i])))
+            else:
+                ops.append(dist.P2POp(dist.irecv, x[lo:hi], int(h.senders[i])))
         for i in range(h.nrecipients):
             lo = int(h.sdispls[i])
             hi = lo + int(h.sendcounts[i])
-            ops.append(dist.P2POp(dist.isend, self.sendbuf[lo:hi], int(h.recipients[i])))
+            ops.append(dist.P2POp(dist.isend, sendsrc[lo:hi], int(h.recipients[i])))
+        self._x = x if staged else None
         self._reqs = dist.batch_isend_irecv(ops) if ops else []
         self.nexchanges += 1
         esize = x.element_size()
@@ -86,6 +100,11 @@ class HaloExchange:
         for r in self._reqs:
             r.wait()
         self._reqs = []
+        if self._staged_recv:
+            for buf, lo, hi in self._staged_recv:
+                self._x[lo:hi].copy_(buf)
+            self._staged_recv = []
+            self._x = None
 
     def exchange(self, x: torch.Tensor) -> None:
         self.begin(x)
