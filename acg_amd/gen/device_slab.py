@@ -43,7 +43,12 @@ class DeviceSlabSystem:
 
 
 def device_stencil_slab(gx: int, gy: int, gz: int, spec: dict,
-                        rank: int, nranks: int, device) -> DeviceSlabSystem:
+                        rank: int, nranks: int, device,
+                        operator: bool = True) -> DeviceSlabSystem:
+    """``operator=False`` skips assembling the SELL/BSELL arrays (row
+    lengths are still counted for nnz stats): matrix-free solves (dof=1)
+    need no memory-resident operator at all, freeing the ~12 B/nnz for
+    larger grids per GPU."""
     from ..ops import gpu_ops
 
     K = gpu_ops.K
@@ -99,6 +104,8 @@ def device_stencil_slab(gx: int, gy: int, gz: int, spec: dict,
         # avoid an extra multi-GB copy for dof=1 (the 8.6e9-row Poisson)
         rowlen = rowlen_nodes if dof == 1 else rowlen_nodes.repeat_interleave(dof)
         nnz = int(rowlen.sum())
+        if not operator:
+            return None, None, None, nnz
         nrows = nodes * dof
         nslices = (nrows + 63) // 64
         padded = torch.zeros(nslices * 64, dtype=torch.int64, device=device)
@@ -119,10 +126,18 @@ def device_stencil_slab(gx: int, gy: int, gz: int, spec: dict,
     A_sellptr, A_cols, A_vals, nnzA = build(0, nown_nodes, 0)
     O_sellptr, O_cols, O_vals, nnzO = build(1, nborder_nodes, ninterior_nodes)
 
+    # matrix-free operator tables (dof=1 constant-coefficient stencils):
+    # everything k_stencil_spmv / k_stencil_pipe need to apply the operator
+    # without any assembled matrix (ops.gpu_ops.stencil_spmv).
+    mf_tables = None
+    if dof == 1:
+        mf_tables = (zs_own, pb, offs, ksten, float(D[0, 0]),
+                     gx, gy, gz, nown_nodes)
+
     # Block-SELL for matA when the operator has dense dof x dof blocks:
     # one int32 index per block instead of per entry (4 -> 4/dof^2 B/nnz).
     A_bsell = None
-    if 2 <= dof <= 4 and nown_nodes > 0:
+    if operator and 2 <= dof <= 4 and nown_nodes > 0:
         blocklen = torch.empty(nown_nodes, dtype=torch.int64, device=device)
         K.stencil_blocklen(nown_nodes, gx, gy, gz, nown_nodes,
                            zs_own.data_ptr(), pb.data_ptr(), offs.data_ptr(),
@@ -187,8 +202,9 @@ def device_stencil_slab(gx: int, gy: int, gz: int, spec: dict,
         rank=rank, nparts=nranks, n_global=gx * gy * gz * dof,
         nowned=nowned, ninterior=ninterior_nodes * dof,
         nborder=nborder_nodes * dof, nghost=nghost,
-        A_sell=(A_sellptr, A_cols, A_vals),
+        A_sell=(A_sellptr, A_cols, A_vals) if operator else None,
         A_bsell=A_bsell,
-        O_sell=(O_sellptr, O_cols, O_vals),
+        O_sell=(O_sellptr, O_cols, O_vals) if operator else None,
+        mf_tables=mf_tables,
         _nnzA=nnzA, _nnzO=nnzO, halo=halo, device=device,
     )

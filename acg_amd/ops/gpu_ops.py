@@ -126,6 +126,43 @@ def spmv_bsell(bptr: torch.Tensor, bcol: torch.Tensor, bvals: torch.Tensor,
                  scal.data_ptr() if fuse else 0, dotslot, dot_accum, _stream())
 
 
+def stencil_spmv(mf, nrows_nodes: int, row0_node: int, x: torch.Tensor,
+                 y: torch.Tensor, *, mato: bool = False,
+                 partials: torch.Tensor | None = None,
+                 scal: torch.Tensor | None = None, dotslot: int = -1,
+                 dot_accum: bool = True) -> None:
+    """Matrix-free stencil SpMV (dof=1 constant-coefficient operators).
+    ``mf`` = (zs_of_plane, pb, offs, ksten, diag, gx, gy, gz, nown_nodes)."""
+    zs, pb, offs, ksten, diag, gx, gy, gz, nown_nodes = mf
+    if nrows_nodes <= 0:
+        return
+    fuse = scal is not None and dotslot >= 0
+    K.stencil_spmv(nrows_nodes, row0_node, gx, gy, gz, nown_nodes,
+                   zs.data_ptr(), pb.data_ptr(), offs.data_ptr(), ksten, diag,
+                   x.data_ptr(), y.data_ptr(), mato,
+                   partials.data_ptr() if fuse else 0,
+                   scal.data_ptr() if fuse else 0, dotslot, dot_accum,
+                   _stream())
+
+
+def stencil_pipe(mf, nrows_nodes: int, row0_node: int, border_base: int,
+                 w_old, qpart, z, t, p, x, r, w_new, scal, first: bool,
+                 partials, partials_off: int, mato: bool) -> int:
+    """Megafused matrix-free pipelined iteration pass (stencil SpMV +
+    6-vector update + dots); same protocol as sell_pipe."""
+    zs, pb, offs, ksten, diag, gx, gy, gz, nown_nodes = mf
+    if nrows_nodes <= 0:
+        return 0
+    return K.stencil_pipe(nrows_nodes, row0_node, border_base, gx, gy, gz,
+                          nown_nodes, zs.data_ptr(), pb.data_ptr(),
+                          offs.data_ptr(), ksten, diag, w_old.data_ptr(),
+                          qpart.data_ptr() if qpart is not None else 0,
+                          z.data_ptr(), t.data_ptr(), p.data_ptr(),
+                          x.data_ptr(), r.data_ptr(), w_new.data_ptr(),
+                          scal.data_ptr(), 1 if first else 0,
+                          partials.data_ptr(), partials_off, mato, _stream())
+
+
 def zero_scalars(scal: torch.Tensor, i0: int = 0, count: int | None = None) -> None:
     K.zero_scalars(scal.data_ptr(), i0, count if count is not None else scal.numel() - i0,
                    _stream())

@@ -860,6 +860,176 @@ void stencil_fill(long nrows_nodes, long row0_node, int gx, int gy, int gz,
 }
 
 // ---------------------------------------------------------------------------
+// MATRIX-FREE stencil operator (dof=1, constant coefficients -- 5/7/27-pt
+// Poisson-type).  The assembled operator re-reads 12 B/nnz of vals+cols per
+// SpMV that the stencil makes redundant: y[i] = diag*x[i] + sum w*x[nb] is
+// fully determined by the grid coordinates.  Applying it matrix-free reads
+// ONLY x (largely L2-cached across the 7/27 neighbour touches) and writes y,
+// dropping per-iteration HBM traffic by the whole vals+cols stream -- and
+// removes the operator from HBM entirely (a 2048^3 7-pt slab is ~90 GB of
+// SELL; matrix-free it is 0).  Beyond the reference (aCG always assembles);
+// opt-in because real matrices (Queen_4147) have position-dependent values
+// and MUST be measured with the memory-resident operator.
+//
+// Same matA/matO split as the assembled path: the matA pass covers owned
+// couplings (diag included), the matO pass adds ghost-plane couplings to
+// border rows after the halo lands.
+template <bool MATO, bool FUSE_DOT>
+__global__ void __launch_bounds__(BLOCK)
+k_stencil_spmv(long nrows_nodes, long row0_node, int gx, int gy, int gz,
+               long nown_nodes,
+               const int* __restrict__ zs_of_plane,
+               const long* __restrict__ pb,
+               const double* __restrict__ offs, int ksten, double diag,
+               const double* __restrict__ x, double* __restrict__ y,
+               double* __restrict__ partials) {
+    const long stride = (long)gridDim.x * BLOCK;
+    const long plane_nodes = (long)gx * gy;
+    double dacc = 0.0;
+    for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nrows_nodes;
+         i += stride) {
+        const long node = row0_node + i;
+        const int pl = (int)(node / plane_nodes);
+        const long rem = node - (long)pl * plane_nodes;
+        const int xi = (int)(rem % gx), yi = (int)(rem / gx);
+        const int zi = zs_of_plane[pl];
+        double sum = MATO ? 0.0 : diag * x[node];
+        for (int o = 0; o < ksten; ++o) {
+            const int dx = (int)offs[o * 4 + 0], dy = (int)offs[o * 4 + 1],
+                      dz = (int)offs[o * 4 + 2];
+            const long cn = stencil_col_node(xi, yi, zi, dx, dy, dz,
+                                             gx, gy, gz, pb);
+            if (cn < 0) continue;
+            if ((cn >= nown_nodes) == MATO) sum += offs[o * 4 + 3] * x[cn];
+        }
+        if (MATO) y[node] += sum; else y[node] = sum;
+        if (FUSE_DOT) dacc += x[node] * sum;
+    }
+    if (FUSE_DOT) {
+        dacc = block_reduce(dacc);
+        if (threadIdx.x == 0) partials[blockIdx.x] = dacc;
+    }
+}
+
+// Megafused matrix-free pipelined iteration: identical epilogue and qpart /
+// partials protocol to k_sell_pipe (see there), with the SpMV computed from
+// the stencil instead of SELL loads.  Per-iteration traffic collapses to
+// the 11n-double vector stream alone.
+template <bool MATO>
+__global__ void __launch_bounds__(BLOCK)
+k_stencil_pipe(long nrows_nodes, long row0_node, long border_base,
+               int gx, int gy, int gz, long nown_nodes,
+               const int* __restrict__ zs_of_plane,
+               const long* __restrict__ pb,
+               const double* __restrict__ offs, int ksten, double diag,
+               const double* __restrict__ w_old, double* __restrict__ qpart,
+               double* __restrict__ z, double* __restrict__ t,
+               double* __restrict__ p, double* __restrict__ xv,
+               double* __restrict__ r, double* __restrict__ w_new,
+               const double* __restrict__ scal, int first,
+               double* __restrict__ partials, long partials_off) {
+    const long stride = (long)gridDim.x * BLOCK;
+    const long plane_nodes = (long)gx * gy;
+    double beta, alpha;
+    pipelined_coeffs(scal, first, &beta, &alpha);
+    double g = 0.0, d = 0.0;
+    for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nrows_nodes;
+         i += stride) {
+        const long node = row0_node + i;
+        const int pl = (int)(node / plane_nodes);
+        const long rem = node - (long)pl * plane_nodes;
+        const int xi = (int)(rem % gx), yi = (int)(rem / gx);
+        const int zi = zs_of_plane[pl];
+        double sum = MATO ? 0.0 : diag * w_old[node];
+        for (int o = 0; o < ksten; ++o) {
+            const int dx = (int)offs[o * 4 + 0], dy = (int)offs[o * 4 + 1],
+                      dz = (int)offs[o * 4 + 2];
+            const long cn = stencil_col_node(xi, yi, zi, dx, dy, dz,
+                                             gx, gy, gz, pb);
+            if (cn < 0) continue;
+            if ((cn >= nown_nodes) == MATO) sum += offs[o * 4 + 3] * w_old[cn];
+        }
+        const long row = node;  // dof = 1
+        if (!MATO && row >= border_base) {
+            qpart[row - border_base] = sum;  // defer to the matO pass
+        } else {
+            const double q = MATO ? sum + qpart[row - border_base] : sum;
+            const double zi_ = q + beta * ld_nt(z + row);
+            const double ti = w_old[row] + beta * ld_nt(t + row);
+            const double pi = r[row] + beta * ld_nt(p + row);
+            __builtin_nontemporal_store(zi_, z + row);
+            __builtin_nontemporal_store(ti, t + row);
+            __builtin_nontemporal_store(pi, p + row);
+            __builtin_nontemporal_store(ld_nt(xv + row) + alpha * pi, xv + row);
+            const double rn = r[row] - alpha * ti;
+            const double wn = w_old[row] - alpha * zi_;
+            r[row] = rn; w_new[row] = wn;
+            g += rn * rn;
+            d += wn * rn;
+        }
+    }
+    g = block_reduce(g);
+    __syncthreads();
+    d = block_reduce(d);
+    if (threadIdx.x == 0) {
+        partials[partials_off + blockIdx.x] = g;
+        partials[MAXG + partials_off + blockIdx.x] = d;
+    }
+}
+
+void stencil_spmv(long nrows_nodes, long row0_node, int gx, int gy, int gz,
+                  long nown_nodes, uintptr_t zs_of_plane, uintptr_t pb,
+                  uintptr_t offs, int ksten, double diag, uintptr_t x,
+                  uintptr_t y, bool mato, uintptr_t partials, uintptr_t scal,
+                  int dotslot, bool dot_accum, uintptr_t stream) {
+    if (nrows_nodes == 0) return;
+    const long blocks = elem_grid(nrows_nodes);
+    const bool fuse = partials != 0 && dotslot >= 0;
+    dim3 g((unsigned)blocks), b(BLOCK);
+    #define LMS(MATO, FD) \
+        hipLaunchKernelGGL((k_stencil_spmv<MATO, FD>), g, b, 0, \
+            (hipStream_t)stream, nrows_nodes, row0_node, gx, gy, gz, \
+            nown_nodes, (const int*)zs_of_plane, (const long*)pb, \
+            (const double*)offs, ksten, diag, (const double*)x, (double*)y, \
+            (double*)partials)
+    if (mato) { if (fuse) { LMS(true, true); } else { LMS(true, false); } }
+    else      { if (fuse) { LMS(false, true); } else { LMS(false, false); } }
+    #undef LMS
+    check_hip("stencil_spmv");
+    if (fuse) {
+        hipLaunchKernelGGL(k_reduce_partials, dim3(1), dim3(BLOCK), 0,
+                           (hipStream_t)stream, (const double*)partials,
+                           (int)blocks, (double*)scal, dotslot, dot_accum ? 1 : 0);
+        check_hip("stencil_spmv_reduce");
+    }
+}
+
+long stencil_pipe(long nrows_nodes, long row0_node, long border_base,
+                  int gx, int gy, int gz, long nown_nodes,
+                  uintptr_t zs_of_plane, uintptr_t pb, uintptr_t offs,
+                  int ksten, double diag, uintptr_t w_old, uintptr_t qpart,
+                  uintptr_t z, uintptr_t t, uintptr_t p, uintptr_t xv,
+                  uintptr_t r, uintptr_t w_new, uintptr_t scal, int first,
+                  uintptr_t partials, long partials_off, bool mato,
+                  uintptr_t stream) {
+    if (nrows_nodes == 0) return 0;
+    long blocks = elem_grid(nrows_nodes);
+    if (blocks + partials_off > MAXG) blocks = MAXG - partials_off;
+    dim3 g((unsigned)blocks), b(BLOCK);
+    #define LMP(MATO) \
+        hipLaunchKernelGGL((k_stencil_pipe<MATO>), g, b, 0, (hipStream_t)stream, \
+            nrows_nodes, row0_node, border_base, gx, gy, gz, nown_nodes, \
+            (const int*)zs_of_plane, (const long*)pb, (const double*)offs, \
+            ksten, diag, (const double*)w_old, (double*)qpart, (double*)z, \
+            (double*)t, (double*)p, (double*)xv, (double*)r, (double*)w_new, \
+            (const double*)scal, first, (double*)partials, partials_off)
+    if (mato) { LMP(true); } else { LMP(false); }
+    #undef LMP
+    check_hip("stencil_pipe");
+    return blocks;
+}
+
+// ---------------------------------------------------------------------------
 // Monolithic device-side CG: the ENTIRE solver loop in one cooperative
 // launch -- zero per-iteration launch/sync overhead (reference
 // acgsolverhip_cg_kernel, cg-kernels-hip.hip:1386-1747; single-GPU, like
@@ -1375,6 +1545,8 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.def("spmv_bsell", &spmv_bsell);
     m.def("stencil_blocklen", &stencil_blocklen);
     m.def("stencil_bfill", &stencil_bfill);
+    m.def("stencil_spmv", &stencil_spmv);
+    m.def("stencil_pipe", &stencil_pipe);
     m.attr("S_RR") = S_RR;
     m.attr("S_PT") = S_PT;
     m.attr("S_RR_PREV") = S_RR_PREV;

@@ -40,7 +40,7 @@ class CGSolverHIP:
 
     def __init__(self, local: LocalSystem, comm=None, device=None,
                  lanes: int | None = None, use_sell: bool = True,
-                 profile: bool = False):
+                 profile: bool = False, matfree: bool = False):
         self.local = local
         self.comm = comm
         if device is None:
@@ -62,11 +62,24 @@ class CGSolverHIP:
         self.O_rowptr = self.O_colidx = self.O_vals = None
         self.lanesA = self.lanesO = lanes or 16
         self.bsell = None
+        # matrix-free analytic operator (dof=1 stencils, opt-in): the SpMV
+        # reads NO matrix data -- see ops/kernels.hip k_stencil_spmv.
+        self.matfree = None
+        if matfree:
+            mf = getattr(L, "mf_tables", None)
+            if mf is None:
+                raise ValueError(
+                    "matfree=True needs a device-generated dof=1 stencil "
+                    "system (gen.device_slab with spec dof==1)")
+            self.matfree = mf
         if hasattr(L, "A_sell"):
             # device-generated system (gen.device_slab): SELL already in HBM
             self.sell = L.A_sell
             self.sellO = L.O_sell
             self.bsell = getattr(L, "A_bsell", None)
+            if self.sell is None and self.matfree is None:
+                raise ValueError("system generated with operator=False "
+                                 "requires matfree=True")
         else:
             self.A_rowptr = up(L.A_rowptr)
             self.A_colidx = up(L.A_colidx)
@@ -125,12 +138,16 @@ class CGSolverHIP:
         # for narrow rows (7-pt Poisson) the q elimination is ~8% of the
         # iteration's traffic.  Auto-enable below ~16 nnz/row.
         self.can_megafuse = (
-            self.sell is not None and self.sell_perm is None
-            and self.sell[1].dtype == torch.int32
-            and (L.nnzO == 0 or (self.sellO is not None
-                                 and self.sellO[1].dtype == torch.int32)))
+            self.matfree is not None
+            or (self.sell is not None and self.sell_perm is None
+                and self.sell[1].dtype == torch.int32
+                and (L.nnzO == 0 or (self.sellO is not None
+                                     and self.sellO[1].dtype == torch.int32))))
+        # matrix-free: megafusion always pays (the SpMV reads no matrix, so
+        # the vector-traffic saving has no x-gather-locality cost to trade)
         self.megafuse_auto = (self.can_megafuse
-                              and L.nnzA / max(L.nowned, 1) <= 16.0)
+                              and (self.matfree is not None
+                                   or L.nnzA / max(L.nowned, 1) <= 16.0))
         self.halo = HaloExchange(L.halo, L.nowned, self.device, comm)
         self.scal = ops.alloc_scalars(self.device)
         self.partials = ops.alloc_partials(self.device)
@@ -174,7 +191,10 @@ class CGSolverHIP:
         with self.prof.span("spmvA"):
             # the matA pass OVERWRITES the dot slot (dot_accum=False), so
             # no zeroing prep kernel is needed; matO accumulates on top
-            if self.bsell is not None:
+            if self.matfree is not None:
+                ops.stencil_spmv(self.matfree, self.n, 0, xfull, y,
+                                 mato=False, dot_accum=False, **fuse)
+            elif self.bsell is not None:
                 bptr, bcol, bvals, dof = self.bsell
                 ops.spmv_bsell(bptr, bcol, bvals, self.n // dof, dof,
                                xfull, y, dot_accum=False, **fuse)
@@ -196,7 +216,10 @@ class CGSolverHIP:
             cur.wait_event(self._ev_recv)
         if L.nborder > 0 and self.local.nnzO > 0:
             with self.prof.span("spmvO"):
-                if self.sellO is not None:
+                if self.matfree is not None:
+                    ops.stencil_spmv(self.matfree, L.nborder, L.ninterior,
+                                     xfull, y, mato=True, **fuse)
+                elif self.sellO is not None:
                     optr, ocols, ovals = self.sellO
                     ops.spmv_sell(optr, ocols, ovals, L.nborder, xfull, y,
                                   rowbase=L.ninterior, accum=True, **fuse)
@@ -345,7 +368,9 @@ class CGSolverHIP:
             from .profiling import annotate_op_stats
 
             idxb = None
-            if self.bsell is not None:
+            if self.matfree is not None:
+                idxb = -8.0  # matrix-free: no vals (8 B) and no cols read
+            elif self.bsell is not None:
                 idxb = 4.0 / (self.bsell[3] ** 2)
             annotate_op_stats(res, self.local, self.prof.collect(),
                               idx_bytes_per_nnz=idxb)
@@ -472,11 +497,16 @@ class CGSolverHIP:
                 self.comm_stream.wait_event(self._ev_p)
                 with torch.cuda.stream(self.comm_stream):
                     self.halo.begin(wa)
-            sp, sc, sv = self.sell
             border_base = L.ninterior if L.nnzO > 0 else n
-            nbA = S.sell_pipe(sp, sc, sv, n, 0, border_base, wa, qpart,
-                              z, t, p, x, r, wb, scal, first,
-                              self.partials, 0, mato=False)
+            if self.matfree is not None:
+                nbA = S.stencil_pipe(self.matfree, n, 0, border_base, wa,
+                                     qpart, z, t, p, x, r, wb, scal, first,
+                                     self.partials, 0, mato=False)
+            else:
+                sp, sc, sv = self.sell
+                nbA = S.sell_pipe(sp, sc, sv, n, 0, border_base, wa, qpart,
+                                  z, t, p, x, r, wb, scal, first,
+                                  self.partials, 0, mato=False)
             nb = nbA
             if have_halo:
                 with torch.cuda.stream(self.comm_stream):
@@ -484,10 +514,17 @@ class CGSolverHIP:
                     self._ev_recv.record(self.comm_stream)
                 cur.wait_event(self._ev_recv)
             if L.nnzO > 0:
-                op_, oc, ov = self.sellO
-                nbO = S.sell_pipe(op_, oc, ov, L.nborder, L.ninterior,
-                                  L.ninterior, wa, qpart, z, t, p, x, r, wb,
-                                  scal, first, self.partials, nbA, mato=True)
+                if self.matfree is not None:
+                    nbO = S.stencil_pipe(self.matfree, L.nborder, L.ninterior,
+                                         L.ninterior, wa, qpart, z, t, p, x,
+                                         r, wb, scal, first, self.partials,
+                                         nbA, mato=True)
+                else:
+                    op_, oc, ov = self.sellO
+                    nbO = S.sell_pipe(op_, oc, ov, L.nborder, L.ninterior,
+                                      L.ninterior, wa, qpart, z, t, p, x, r,
+                                      wb, scal, first, self.partials, nbA,
+                                      mato=True)
                 nb += nbO
             S.pipelined_finalize(self.partials, nb, scal, first)
         # lag-1 convergence pipeline: gamma_k is copied to the host as soon
@@ -623,7 +660,9 @@ class CGSolverHIP:
             from .profiling import annotate_op_stats
 
             idxb = None
-            if self.bsell is not None:
+            if self.matfree is not None:
+                idxb = -8.0  # matrix-free: no vals (8 B) and no cols read
+            elif self.bsell is not None:
                 idxb = 4.0 / (self.bsell[3] ** 2)
             annotate_op_stats(res, self.local, self.prof.collect(),
                               idx_bytes_per_nnz=idxb)

@@ -47,6 +47,12 @@ def main() -> int:
                     help="matrix generation: on-GPU SELL (default) or host numpy")
     ap.add_argument("--lanes", type=int, default=None,
                     help="override SpMV lanes-per-row (4/8/16/32/64)")
+    ap.add_argument("--matfree", action="store_true",
+                    help="matrix-free analytic operator (dof=1 stencil "
+                         "configs only; beyond-reference opt-in -- the "
+                         "default keeps the assembled, memory-resident "
+                         "operator so numbers reflect the sparse-matrix "
+                         "workload BASELINE names)")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -82,13 +88,17 @@ def main() -> int:
         spec = dict(STENCIL_7PT_3D)
         spec["dof"] = dof
         model = f"poisson3d-7pt-G{G}"
+    if args.matfree and (dof != 1 or args.gen != "device"):
+        raise SystemExit("--matfree needs a dof=1 stencil config with --gen device")
     if args.gen == "device":
         from acg_amd.gen.device_slab import device_stencil_slab
 
-        S = device_stencil_slab(G, G, G, spec, rank, ngpus, device)
+        S = device_stencil_slab(G, G, G, spec, rank, ngpus, device,
+                                operator=not args.matfree)
     else:
         S = stencil_local_slab(G, G, G, spec, rank, ngpus)
-    solver = CGSolverHIP(S, comm=comm, device=device, lanes=args.lanes)
+    solver = CGSolverHIP(S, comm=comm, device=device, lanes=args.lanes,
+                         matfree=args.matfree)
 
     rloc = np.random.default_rng(10_000 + rank)
     b = torch.from_numpy(rloc.standard_normal(S.nowned)).to(device)
@@ -157,7 +167,7 @@ def main() -> int:
                 "model": model,
                 "rows": nrows_global,
                 "nnz": nnz_global,
-                "solver": f"cg-{solver_kind}",
+                "solver": f"cg-{solver_kind}" + ("-matfree" if args.matfree else ""),
                 "time_to_solution_s": elapsed,
                 "gflops": args.steps * (2.0 * nnz_global + 10.0 * nrows_global) / elapsed / 1e9,
                 "parallelism": f"slab{ngpus}-rccl",

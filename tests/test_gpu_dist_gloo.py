@@ -114,6 +114,26 @@ def _check_vs_direct(results):
     assert covered.all()
 
 
+def _body_cg_gpu_ws4(comm):
+    """4 ranks on one GPU, generic rgb partition: multi-neighbour halos
+    through the full extract_subdomains path with the HIP solver."""
+    from acg_amd.gen import STENCIL_27PT_3D
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A = stencil_global(6, 6, 6, STENCIL_27PT_3D)
+    part = partition_rows(A, comm.size, method="rgb", seed=1)
+    S = extract_subdomains(A, part, comm.size)[comm.rank]
+    rng = np.random.default_rng(7)
+    b_global = rng.standard_normal(A.n)
+    b = torch.from_numpy(b_global[S.owned_global]).cuda()
+    solver = CGSolverHIP(S, comm=comm, device="cuda:0")
+    x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda:0")
+    res = solver.solve_pipelined(b, x, maxits=800, res_rtol=1e-10)
+    assert res.converged, res.summary()
+    return (S.owned_global, x[:S.nowned].cpu().numpy(), res.niterations)
+
+
 def test_gpu_halo_exchange_2proc():
     _run_dist("_body_halo_gpu", world=2, port=29610)
 
@@ -124,3 +144,17 @@ def test_gpu_cg_classic_2proc():
 
 def test_gpu_cg_pipelined_2proc():
     _check_vs_direct(_run_dist("_body_cg_gpu_pipelined", world=2, port=29612))
+
+
+def test_gpu_cg_ws4_rgb():
+    from acg_amd.gen import STENCIL_27PT_3D
+
+    results = _run_dist("_body_cg_gpu_ws4", world=4, port=29613)
+    A = stencil_global(6, 6, 6, STENCIL_27PT_3D)
+    rng = np.random.default_rng(7)
+    b_global = rng.standard_normal(A.n)
+    import scipy.sparse.linalg as spla
+
+    x_ref = spla.spsolve(A.to_scipy_full().tocsc(), b_global)
+    for _, (owned_global, xloc, _nit) in results.items():
+        np.testing.assert_allclose(xloc, x_ref[owned_global], rtol=1e-6, atol=1e-8)
