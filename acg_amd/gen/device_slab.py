@@ -131,8 +131,20 @@ def device_stencil_slab(gx: int, gy: int, gz: int, spec: dict,
     # without any assembled matrix (ops.gpu_ops.stencil_spmv).
     mf_tables = None
     if dof == 1:
-        mf_tables = (zs_own, pb, offs, ksten, float(D[0, 0]),
-                     gx, gy, gz, nown_nodes)
+        # 7-pt fast path: the z-column-walk kernels need the per-axis
+        # weights and the slab bounds (w_old is then read exactly once,
+        # with the z +- 1 neighbours carried in registers).
+        w7 = None
+        if ksten == 6:
+            wmap = {(int(dx), int(dy), int(dz)): float(w)
+                    for (dx, dy, dz, w) in spec["offsets"]}
+            keys = [(-1, 0, 0), (1, 0, 0), (0, -1, 0),
+                    (0, 1, 0), (0, 0, -1), (0, 0, 1)]
+            if set(wmap) == set(keys):
+                w7 = tuple(wmap[k] for k in keys)
+        mf_tables = dict(zs=zs_own, pb=pb, offs=offs, ksten=ksten,
+                         diag=float(D[0, 0]), gx=gx, gy=gy, gz=gz,
+                         nown_nodes=nown_nodes, z0=z0, z1=z1, w7=w7)
 
     # Block-SELL for matA when the operator has dense dof x dof blocks:
     # one int32 index per block instead of per entry (4 -> 4/dof^2 B/nnz).

@@ -132,17 +132,24 @@ def stencil_spmv(mf, nrows_nodes: int, row0_node: int, x: torch.Tensor,
                  scal: torch.Tensor | None = None, dotslot: int = -1,
                  dot_accum: bool = True) -> None:
     """Matrix-free stencil SpMV (dof=1 constant-coefficient operators).
-    ``mf`` = (zs_of_plane, pb, offs, ksten, diag, gx, gy, gz, nown_nodes)."""
-    zs, pb, offs, ksten, diag, gx, gy, gz, nown_nodes = mf
+    ``mf`` = gen.device_slab mf_tables dict.  7-pt matA passes take the
+    z-column-walk kernel (w_old read exactly once)."""
     if nrows_nodes <= 0:
         return
     fuse = scal is not None and dotslot >= 0
-    K.stencil_spmv(nrows_nodes, row0_node, gx, gy, gz, nown_nodes,
-                   zs.data_ptr(), pb.data_ptr(), offs.data_ptr(), ksten, diag,
-                   x.data_ptr(), y.data_ptr(), mato,
-                   partials.data_ptr() if fuse else 0,
-                   scal.data_ptr() if fuse else 0, dotslot, dot_accum,
-                   _stream())
+    pp = partials.data_ptr() if fuse else 0
+    sp = scal.data_ptr() if fuse else 0
+    if not mato and mf["w7"] is not None:
+        K.stencil_spmv7(mf["gx"] * mf["gy"], mf["gx"], mf["z0"], mf["z1"],
+                        mf["gz"], mf["nown_nodes"], mf["pb"].data_ptr(),
+                        mf["diag"], *mf["w7"], x.data_ptr(), y.data_ptr(),
+                        pp, sp, dotslot, dot_accum, _stream())
+        return
+    K.stencil_spmv(nrows_nodes, row0_node, mf["gx"], mf["gy"], mf["gz"],
+                   mf["nown_nodes"], mf["zs"].data_ptr(),
+                   mf["pb"].data_ptr(), mf["offs"].data_ptr(), mf["ksten"],
+                   mf["diag"], x.data_ptr(), y.data_ptr(), mato,
+                   pp, sp, dotslot, dot_accum, _stream())
 
 
 def stencil_pipe(mf, nrows_nodes: int, row0_node: int, border_base: int,
@@ -150,13 +157,23 @@ def stencil_pipe(mf, nrows_nodes: int, row0_node: int, border_base: int,
                  partials, partials_off: int, mato: bool) -> int:
     """Megafused matrix-free pipelined iteration pass (stencil SpMV +
     6-vector update + dots); same protocol as sell_pipe."""
-    zs, pb, offs, ksten, diag, gx, gy, gz, nown_nodes = mf
     if nrows_nodes <= 0:
         return 0
-    return K.stencil_pipe(nrows_nodes, row0_node, border_base, gx, gy, gz,
-                          nown_nodes, zs.data_ptr(), pb.data_ptr(),
-                          offs.data_ptr(), ksten, diag, w_old.data_ptr(),
-                          qpart.data_ptr() if qpart is not None else 0,
+    qp = qpart.data_ptr() if qpart is not None else 0
+    if not mato and mf["w7"] is not None:
+        return K.stencil_pipe7(mf["gx"] * mf["gy"], mf["gx"], mf["z0"],
+                               mf["z1"], mf["gz"], border_base,
+                               mf["nown_nodes"], mf["pb"].data_ptr(),
+                               mf["diag"], *mf["w7"], w_old.data_ptr(), qp,
+                               z.data_ptr(), t.data_ptr(), p.data_ptr(),
+                               x.data_ptr(), r.data_ptr(), w_new.data_ptr(),
+                               scal.data_ptr(), 1 if first else 0,
+                               partials.data_ptr(), partials_off, _stream())
+    return K.stencil_pipe(nrows_nodes, row0_node, border_base, mf["gx"],
+                          mf["gy"], mf["gz"], mf["nown_nodes"],
+                          mf["zs"].data_ptr(), mf["pb"].data_ptr(),
+                          mf["offs"].data_ptr(), mf["ksten"], mf["diag"],
+                          w_old.data_ptr(), qp,
                           z.data_ptr(), t.data_ptr(), p.data_ptr(),
                           x.data_ptr(), r.data_ptr(), w_new.data_ptr(),
                           scal.data_ptr(), 1 if first else 0,

@@ -977,6 +977,179 @@ k_stencil_pipe(long nrows_nodes, long row0_node, long border_base,
     }
 }
 
+// 7-pt z-column-walk specialisation: one thread owns one (x,y) column and
+// walks the owned planes z0..z1-1, carrying the z-1/z/z+1 values of w_old
+// in registers.  The generic kernel re-reads the z +- 1 planes through an
+// L2 that cannot hold them (a 512^2 plane is 2 MB x thousands of in-flight
+// blocks), costing ~2 extra w_old streams; the walk reads w_old EXACTLY
+// once.  x +- 1 / y +- gx neighbours come from L1 (adjacent lanes load
+// them).  matA pass only -- ghost-plane couplings stay with the generic
+// matO kernel (2 planes at most).
+template <bool FUSE_DOT>
+__global__ void __launch_bounds__(BLOCK)
+k_stencil_spmv7(long plane_nodes, int gx, int z0, int z1, int gz,
+                long nown_nodes, const long* __restrict__ pb,
+                double diag, double wxm, double wxp, double wym, double wyp,
+                double wzm, double wzp,
+                const double* __restrict__ x, double* __restrict__ y,
+                double* __restrict__ partials) {
+    const long stride = (long)gridDim.x * BLOCK;
+    double dacc = 0.0;
+    for (long xy = (long)blockIdx.x * BLOCK + threadIdx.x; xy < plane_nodes;
+         xy += stride) {
+        const int xi = (int)(xy % gx);
+        const long yi = xy / gx;
+        const bool hxm = xi > 0, hxp = xi < gx - 1;
+        const bool hym = yi > 0, hyp = yi < plane_nodes / gx - 1;
+        long base_c = pb[z0 + 1];
+        double xc = x[base_c + xy];
+        const long base_m = (z0 > 0) ? pb[z0] : -1;
+        bool hm = base_m >= 0 && base_m < nown_nodes;
+        double xm = hm ? x[base_m + xy] : 0.0;
+        for (int zz = z0; zz < z1; ++zz) {
+            const long node = base_c + xy;
+            const long base_p = (zz + 1 < gz) ? pb[zz + 2] : -1;
+            const bool hp = base_p >= 0 && base_p < nown_nodes;
+            const double xp = hp ? x[base_p + xy] : 0.0;
+            double sum = diag * xc;
+            if (hxm) sum += wxm * x[node - 1];
+            if (hxp) sum += wxp * x[node + 1];
+            if (hym) sum += wym * x[node - gx];
+            if (hyp) sum += wyp * x[node + gx];
+            if (hm) sum += wzm * xm;
+            if (hp) sum += wzp * xp;
+            y[node] = sum;
+            if (FUSE_DOT) dacc += xc * sum;
+            xm = xc; hm = true;
+            xc = xp; base_c = base_p;
+        }
+    }
+    if (FUSE_DOT) {
+        dacc = block_reduce(dacc);
+        if (threadIdx.x == 0) partials[blockIdx.x] = dacc;
+    }
+}
+
+// megafused 7-pt column walk: k_stencil_pipe's epilogue on the walk above.
+__global__ void __launch_bounds__(BLOCK)
+k_stencil_pipe7(long plane_nodes, int gx, int z0, int z1, int gz,
+                long border_base, long nown_nodes,
+                const long* __restrict__ pb,
+                double diag, double wxm, double wxp, double wym, double wyp,
+                double wzm, double wzp,
+                const double* __restrict__ w_old, double* __restrict__ qpart,
+                double* __restrict__ z, double* __restrict__ t,
+                double* __restrict__ p, double* __restrict__ xv,
+                double* __restrict__ r, double* __restrict__ w_new,
+                const double* __restrict__ scal, int first,
+                double* __restrict__ partials, long partials_off) {
+    const long stride = (long)gridDim.x * BLOCK;
+    double beta, alpha;
+    pipelined_coeffs(scal, first, &beta, &alpha);
+    double g = 0.0, d = 0.0;
+    for (long xy = (long)blockIdx.x * BLOCK + threadIdx.x; xy < plane_nodes;
+         xy += stride) {
+        const int xi = (int)(xy % gx);
+        const long yi = xy / gx;
+        const bool hxm = xi > 0, hxp = xi < gx - 1;
+        const bool hym = yi > 0, hyp = yi < plane_nodes / gx - 1;
+        long base_c = pb[z0 + 1];
+        double xc = w_old[base_c + xy];
+        const long base_m = (z0 > 0) ? pb[z0] : -1;
+        bool hm = base_m >= 0 && base_m < nown_nodes;
+        double xm = hm ? w_old[base_m + xy] : 0.0;
+        for (int zz = z0; zz < z1; ++zz) {
+            const long node = base_c + xy;
+            const long base_p = (zz + 1 < gz) ? pb[zz + 2] : -1;
+            const bool hp = base_p >= 0 && base_p < nown_nodes;
+            const double xp = hp ? w_old[base_p + xy] : 0.0;
+            double sum = diag * xc;
+            if (hxm) sum += wxm * w_old[node - 1];
+            if (hxp) sum += wxp * w_old[node + 1];
+            if (hym) sum += wym * w_old[node - gx];
+            if (hyp) sum += wyp * w_old[node + gx];
+            if (hm) sum += wzm * xm;
+            if (hp) sum += wzp * xp;
+            if (node >= border_base) {
+                qpart[node - border_base] = sum;  // matO finishes border rows
+            } else {
+                const double zi_ = sum + beta * ld_nt(z + node);
+                const double ti = xc + beta * ld_nt(t + node);
+                const double pi = r[node] + beta * ld_nt(p + node);
+                __builtin_nontemporal_store(zi_, z + node);
+                __builtin_nontemporal_store(ti, t + node);
+                __builtin_nontemporal_store(pi, p + node);
+                __builtin_nontemporal_store(ld_nt(xv + node) + alpha * pi,
+                                            xv + node);
+                const double rn = r[node] - alpha * ti;
+                const double wn = xc - alpha * zi_;
+                r[node] = rn; w_new[node] = wn;
+                g += rn * rn;
+                d += wn * rn;
+            }
+            xm = xc; hm = true;
+            xc = xp; base_c = base_p;
+        }
+    }
+    g = block_reduce(g);
+    __syncthreads();
+    d = block_reduce(d);
+    if (threadIdx.x == 0) {
+        partials[partials_off + blockIdx.x] = g;
+        partials[MAXG + partials_off + blockIdx.x] = d;
+    }
+}
+
+void stencil_spmv7(long plane_nodes, int gx, int z0, int z1, int gz,
+                   long nown_nodes, uintptr_t pb, double diag,
+                   double wxm, double wxp, double wym, double wyp,
+                   double wzm, double wzp, uintptr_t x, uintptr_t y,
+                   uintptr_t partials, uintptr_t scal, int dotslot,
+                   bool dot_accum, uintptr_t stream) {
+    if (plane_nodes == 0 || z1 <= z0) return;
+    long blocks = (plane_nodes + BLOCK - 1) / BLOCK;
+    if (blocks > MAXG) blocks = MAXG;
+    const bool fuse = partials != 0 && dotslot >= 0;
+    dim3 g((unsigned)blocks), b(BLOCK);
+    #define LS7(FD) \
+        hipLaunchKernelGGL((k_stencil_spmv7<FD>), g, b, 0, (hipStream_t)stream, \
+            plane_nodes, gx, z0, z1, gz, nown_nodes, (const long*)pb, diag, \
+            wxm, wxp, wym, wyp, wzm, wzp, (const double*)x, (double*)y, \
+            (double*)partials)
+    if (fuse) { LS7(true); } else { LS7(false); }
+    #undef LS7
+    check_hip("stencil_spmv7");
+    if (fuse) {
+        hipLaunchKernelGGL(k_reduce_partials, dim3(1), dim3(BLOCK), 0,
+                           (hipStream_t)stream, (const double*)partials,
+                           (int)blocks, (double*)scal, dotslot, dot_accum ? 1 : 0);
+        check_hip("stencil_spmv7_reduce");
+    }
+}
+
+long stencil_pipe7(long plane_nodes, int gx, int z0, int z1, int gz,
+                   long border_base, long nown_nodes, uintptr_t pb,
+                   double diag, double wxm, double wxp, double wym,
+                   double wyp, double wzm, double wzp, uintptr_t w_old,
+                   uintptr_t qpart, uintptr_t z, uintptr_t t, uintptr_t p,
+                   uintptr_t xv, uintptr_t r, uintptr_t w_new, uintptr_t scal,
+                   int first, uintptr_t partials, long partials_off,
+                   uintptr_t stream) {
+    if (plane_nodes == 0 || z1 <= z0) return 0;
+    long blocks = (plane_nodes + BLOCK - 1) / BLOCK;
+    if (blocks + partials_off > MAXG) blocks = MAXG - partials_off;
+    dim3 g((unsigned)blocks), b(BLOCK);
+    hipLaunchKernelGGL(k_stencil_pipe7, g, b, 0, (hipStream_t)stream,
+                       plane_nodes, gx, z0, z1, gz, border_base, nown_nodes,
+                       (const long*)pb, diag, wxm, wxp, wym, wyp, wzm, wzp,
+                       (const double*)w_old, (double*)qpart, (double*)z,
+                       (double*)t, (double*)p, (double*)xv, (double*)r,
+                       (double*)w_new, (const double*)scal, first,
+                       (double*)partials, partials_off);
+    check_hip("stencil_pipe7");
+    return blocks;
+}
+
 void stencil_spmv(long nrows_nodes, long row0_node, int gx, int gy, int gz,
                   long nown_nodes, uintptr_t zs_of_plane, uintptr_t pb,
                   uintptr_t offs, int ksten, double diag, uintptr_t x,
@@ -1547,6 +1720,8 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.def("stencil_bfill", &stencil_bfill);
     m.def("stencil_spmv", &stencil_spmv);
     m.def("stencil_pipe", &stencil_pipe);
+    m.def("stencil_spmv7", &stencil_spmv7);
+    m.def("stencil_pipe7", &stencil_pipe7);
     m.attr("S_RR") = S_RR;
     m.attr("S_PT") = S_PT;
     m.attr("S_RR_PREV") = S_RR_PREV;
