@@ -49,6 +49,12 @@ def main():
                                                     maxits=args.steps,
                                                     res_rtol=0.0),
     }
+    if getattr(S, "mf_tables", None) is not None:
+        mfsolver = CGSolverHIP(S, device="cuda:0", matfree=True)
+        arms["matfree-pipelined"] = lambda: mfsolver.solve_pipelined(
+            b, x.clone(), maxits=args.steps, res_rtol=0.0)
+        arms["matfree-classic"] = lambda: mfsolver.solve(
+            b, x.clone(), maxits=args.steps, res_rtol=0.0)
     if solver.can_megafuse:
         arms["pipelined-nomega"] = lambda: solver.solve_pipelined(
             b, x.clone(), maxits=args.steps, res_rtol=0.0, megafuse=False)
