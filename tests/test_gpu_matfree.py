@@ -85,6 +85,31 @@ def test_matfree_classic_matches_assembled():
     np.testing.assert_allclose(x_mf, x_as, rtol=1e-8, atol=1e-10)
 
 
+def test_matfree_27pt_pipelined_matches_assembled():
+    """27-pt has no column-walk fast path: this exercises the GENERIC
+    megafused k_stencil_pipe end-to-end (w7 is None)."""
+    from acg_amd.gen.device_slab import device_stencil_slab
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    S = device_stencil_slab(14, 14, 14, _spec("27pt"), 0, 1, "cuda:0")
+    assert S.mf_tables["w7"] is None
+    rng = np.random.default_rng(9)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+
+    def run(matfree):
+        solver = CGSolverHIP(S, device="cuda:0", matfree=matfree)
+        x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64,
+                        device="cuda")
+        res = solver.solve_pipelined(b, x, maxits=400, res_rtol=1e-10)
+        assert res.converged, res.summary()
+        return x[:S.nowned].cpu().numpy(), res
+
+    x_mf, r_mf = run(True)
+    x_as, r_as = run(False)
+    assert abs(r_mf.niterations - r_as.niterations) <= 2
+    np.testing.assert_allclose(x_mf, x_as, rtol=1e-8, atol=1e-10)
+
+
 def test_matfree_megafused_matches_nomega():
     """Default matfree pipelined runs the megafused k_stencil_pipe (+ graph
     replay); it must agree with the non-megafused matfree path."""
