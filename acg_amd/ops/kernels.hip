@@ -1137,7 +1137,8 @@ long stencil_pipe7(long plane_nodes, int gx, int z0, int z1, int gz,
                    uintptr_t stream) {
     if (plane_nodes == 0 || z1 <= z0) return 0;
     long blocks = (plane_nodes + BLOCK - 1) / BLOCK;
-    if (blocks + partials_off > MAXG) blocks = MAXG - partials_off;
+    const long cap = (partials_off == 0) ? (MAXG - 1024) : (MAXG - partials_off);
+    if (blocks > cap) blocks = cap;  // see sell_pipe
     dim3 g((unsigned)blocks), b(BLOCK);
     hipLaunchKernelGGL(k_stencil_pipe7, g, b, 0, (hipStream_t)stream,
                        plane_nodes, gx, z0, z1, gz, border_base, nown_nodes,
@@ -1187,7 +1188,8 @@ long stencil_pipe(long nrows_nodes, long row0_node, long border_base,
                   uintptr_t stream) {
     if (nrows_nodes == 0) return 0;
     long blocks = elem_grid(nrows_nodes);
-    if (blocks + partials_off > MAXG) blocks = MAXG - partials_off;
+    const long cap = (partials_off == 0) ? (MAXG - 1024) : (MAXG - partials_off);
+    if (blocks > cap) blocks = cap;  // see sell_pipe
     dim3 g((unsigned)blocks), b(BLOCK);
     #define LMP(MATO) \
         hipLaunchKernelGGL((k_stencil_pipe<MATO>), g, b, 0, (hipStream_t)stream, \
@@ -1591,7 +1593,11 @@ long sell_pipe(long nslices, long nrows_pass, long rowbase, long border_base,
                long partials_off, bool mato, uintptr_t stream) {
     if (nrows_pass == 0) return 0;
     long blocks = (nslices * WAVE + BLOCK - 1) / BLOCK;
-    if (blocks + partials_off > MAXG) blocks = MAXG - partials_off;
+    // the matA pass (partials_off == 0) must leave partial slots for the
+    // matO pass: at >=3.9M rows an uncapped matA claims all MAXG blocks
+    // and matO would launch with grid 0 (invalid configuration)
+    const long cap = (partials_off == 0) ? (MAXG - 1024) : (MAXG - partials_off);
+    if (blocks > cap) blocks = cap;
     dim3 g((unsigned)blocks), b(BLOCK);
     #define LP(MATO) \
         hipLaunchKernelGGL((k_sell_pipe<int, true, 8, MATO>), g, b, 0, S(stream), \

@@ -155,6 +155,37 @@ def test_column_walk_matches_generic(nranks):
                                    rtol=1e-14, atol=1e-13)
 
 
+@pytest.mark.parametrize("matfree", [False, True])
+def test_megafused_large_matA_leaves_matO_blocks(matfree):
+    """Regression: at >=3.9M owned rows the matA pass used to claim all
+    MAXG partials blocks, so a matO pass (any multi-rank system) launched
+    with grid 0 -> 'invalid configuration argument'.  Rank 0 of 2 at
+    256^3 crosses the threshold (8.4M rows) with a live matO pass."""
+    from acg_amd.gen import STENCIL_7PT_3D
+    from acg_amd.gen.device_slab import device_stencil_slab
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    S = device_stencil_slab(256, 256, 256, dict(STENCIL_7PT_3D), 0, 2,
+                            "cuda:0", operator=not matfree)
+    assert S.nnzO > 0
+    rng = np.random.default_rng(3)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+
+    def run(megafuse):
+        solver = CGSolverHIP(S, device="cuda:0", matfree=matfree)
+        x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64,
+                        device="cuda")
+        solver.solve_pipelined(b, x, maxits=25, res_rtol=0.0,
+                               megafuse=megafuse)
+        return x[:S.nowned].cpu().numpy()
+
+    if matfree:
+        np.testing.assert_allclose(run(True), run(False), rtol=1e-8,
+                                   atol=1e-10)
+    else:
+        run(True)  # operator=True build is the expensive part; mega only
+
+
 def test_matfree_requires_stencil_system():
     from acg_amd.gen import queen_like_spec
     from acg_amd.gen.device_slab import device_stencil_slab
