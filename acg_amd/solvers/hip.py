@@ -163,6 +163,21 @@ class CGSolverHIP:
         self._ev_rr = torch.cuda.Event()
         # always-on counters (reference cghip.h:109-118)
         self.niterations_total = 0
+        # persistent per-method workspace vectors + captured iteration
+        # graphs: repeated solves on one solver instance replay the graphs
+        # captured on the FIRST solve (capture costs ~5-10 ms; re-capturing
+        # inside every solve costs 2-10% at typical step counts).  Solution
+        # state lives in the internal "xi" buffer so graph-referenced
+        # addresses never change; the caller's x is copied in/out.
+        self._ws: dict = {}
+        self._graphs: dict = {}
+
+    def _workspace(self, key: str, names) -> dict:
+        ws = self._ws.setdefault(key, {})
+        for nm, nghost in names:
+            if nm not in ws:
+                ws[nm] = self._vec(nghost=nghost)
+        return ws
 
     # -- pieces -----------------------------------------------------------
 
@@ -257,16 +272,17 @@ class CGSolverHIP:
         n = self.n
         S = ops
         scal = self.scal
-        r = self._vec()
-        t = self._vec()
-        p = self._vec(nghost=True)
+        ws = self._workspace("classic", [("r", False), ("t", False),
+                                         ("p", True), ("xi", True)])
+        r, t, p, xi = ws["r"], ws["t"], ws["p"], ws["xi"]
+        xi.copy_(x)
         torch.cuda.synchronize(self.device)
         t0 = time.perf_counter()
         # bnrm2
         S.dot(b, b, self.partials, scal, S.S_BNRM2, n=n)
         self._allreduce_slot(S.S_BNRM2)
         # r0 = b - A x0;  p = r0
-        self._spmv_overlapped(x, t)
+        self._spmv_overlapped(xi, t)
         torch.sub(b[:n], t, out=r)
         p[:n] = r
         S.dot(r, r, self.partials, scal, S.S_RR, n=n)
@@ -283,8 +299,8 @@ class CGSolverHIP:
             return res
         converged = False
         serial = self.comm is None or self.comm.size == 1
-        graph = None
         graph_ok = serial and not self.prof.enabled
+        graph = self._graphs.get("classic") if graph_ok else None
         # lag-2 convergence pipeline + hipGraph replay, mirroring
         # solve_pipelined (the host test runs for every iteration; the host
         # reads the value two iterations late).  Classic's rr copy lands at
@@ -305,7 +321,7 @@ class CGSolverHIP:
             # fused r/x update (alpha = rr/pt on device) + combined
             # finalize (rr -> rr_prev rotation + new (r,r))
             with self.prof.span("update_classic"):
-                S.cg_fused_update(r, x, p, t, scal, self.partials, n)
+                S.cg_fused_update(r, xi, p, t, scal, self.partials, n)
             with self.prof.span("allreduce"):
                 self._allreduce_slot(S.S_RR)
             # p = (rr/rr_prev) p + r
@@ -340,6 +356,7 @@ class CGSolverHIP:
                     graph = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(graph):
                         body()
+                    self._graphs["classic"] = graph
             # lagged rr D2H on the SAME stream: cross-stream event chains
             # measured ~20 us each in fire->launch latency (45 us/iter of
             # the iteration period); the in-order 8-byte copy costs ~3 us
@@ -356,6 +373,7 @@ class CGSolverHIP:
                     break
         torch.cuda.synchronize(self.device)
         res.tsolve = time.perf_counter() - t0
+        x.copy_(xi)
         rr = self._host_scalar(S.S_RR)
         res.rnrm2 = math.sqrt(max(rr, 0.0))
         res.converged = converged or (rtol2 > 0 and rr <= rtol2)
@@ -455,24 +473,32 @@ class CGSolverHIP:
         L = self.local
         scal = self.scal
         mega = self.megafuse_auto if megafuse is None else (megafuse and self.can_megafuse)
-        r = self._vec(nghost=True)
-        w = self._vec(nghost=True)
-        z = self._vec()
-        t = self._vec()
-        p = self._vec()
-        tmp = self._vec()
+        wskey = f"pipelined{'_mega' if mega else ''}"
+        names = [("r", True), ("w", True), ("z", False), ("t", False),
+                 ("p", False), ("tmp", False), ("xi", True)]
+        names += [("w2", True)] if mega else [("q", False)]
+        ws = self._workspace(wskey, names)
+        r, w, z, t, p, tmp, xi = (ws["r"], ws["w"], ws["z"], ws["t"],
+                                  ws["p"], ws["tmp"], ws["xi"])
         # megafused path double-buffers w (SpMV gathers w_old while the
         # fused epilogue writes w_new) and stages border q in qpart;
         # the separate q vector exists only on the fallback path.
-        w2 = self._vec(nghost=True) if mega else None
-        qpart = (torch.zeros(max(L.nborder, 1), dtype=torch.float64,
-                             device=self.device) if mega else None)
-        q = None if mega else self._vec()
+        w2 = ws.get("w2")
+        q = ws.get("q")
+        if mega and "qpart" not in ws:
+            ws["qpart"] = torch.zeros(max(L.nborder, 1), dtype=torch.float64,
+                                      device=self.device)
+        qpart = ws.get("qpart")
+        xi.copy_(x)
+        # first=True multiplies z/t/p by beta=0: stale non-finite values
+        # from an aborted earlier solve must not poison 0*x
+        for v in (z, t, p):
+            v.zero_()
         torch.cuda.synchronize(self.device)
         t0 = time.perf_counter()
         S.dot(b, b, self.partials, scal, S.S_BNRM2, n=n)
         self._allreduce_slot(S.S_BNRM2)
-        self._spmv_overlapped(x, tmp)
+        self._spmv_overlapped(xi, tmp)
         torch.sub(b[:n], tmp, out=r[:n])
         self._spmv_overlapped(r, w)  # w = A r
         res.bnrm2 = math.sqrt(max(self._host_scalar(S.S_BNRM2), 0.0))
@@ -483,9 +509,11 @@ class CGSolverHIP:
         converged = False
         gamma_host = None
         serial = self.comm is None or self.comm.size == 1
-        graph = None
-        graphs = [None, None]  # megafused: even/odd w ping-pong
         graph_ok = use_graph and serial and not self.prof.enabled
+        graph = self._graphs.get(wskey) if graph_ok and not mega else None
+        # megafused: even/odd w ping-pong graphs
+        graphs = (self._graphs.get(wskey, [None, None])
+                  if graph_ok and mega else [None, None])
 
         def mega_body(wa, wb, first):
             """One megafused iteration: halo(wa) || matA pass (SpMV + update
@@ -500,12 +528,12 @@ class CGSolverHIP:
             border_base = L.ninterior if L.nnzO > 0 else n
             if self.matfree is not None:
                 nbA = S.stencil_pipe(self.matfree, n, 0, border_base, wa,
-                                     qpart, z, t, p, x, r, wb, scal, first,
+                                     qpart, z, t, p, xi, r, wb, scal, first,
                                      self.partials, 0, mato=False)
             else:
                 sp, sc, sv = self.sell
                 nbA = S.sell_pipe(sp, sc, sv, n, 0, border_base, wa, qpart,
-                                  z, t, p, x, r, wb, scal, first,
+                                  z, t, p, xi, r, wb, scal, first,
                                   self.partials, 0, mato=False)
             nb = nbA
             if have_halo:
@@ -516,13 +544,13 @@ class CGSolverHIP:
             if L.nnzO > 0:
                 if self.matfree is not None:
                     nbO = S.stencil_pipe(self.matfree, L.nborder, L.ninterior,
-                                         L.ninterior, wa, qpart, z, t, p, x,
+                                         L.ninterior, wa, qpart, z, t, p, xi,
                                          r, wb, scal, first, self.partials,
                                          nbA, mato=True)
                 else:
                     op_, oc, ov = self.sellO
                     nbO = S.sell_pipe(op_, oc, ov, L.nborder, L.ninterior,
-                                      L.ninterior, wa, qpart, z, t, p, x, r,
+                                      L.ninterior, wa, qpart, z, t, p, xi, r,
                                       wb, scal, first, self.partials, nbA,
                                       mato=True)
                 nb += nbO
@@ -609,16 +637,17 @@ class CGSolverHIP:
             if mega:
                 wa, wb = (w, w2) if k % 2 == 0 else (w2, w)
                 g = graphs[k % 2]
-                if g is not None:
+                if g is not None and k > 0:  # graphs are first=False bodies
                     g.replay()
                 else:
                     mega_body(wa, wb, first)
-                    if graph_ok and k in (2, 3):
+                    if graph_ok and graphs[k % 2] is None and k in (2, 3):
                         gg = torch.cuda.CUDAGraph()
                         with torch.cuda.graph(gg):
                             mega_body(wa, wb, False)
                         graphs[k % 2] = gg
-            elif graph is not None:
+                        self._graphs[wskey] = graphs
+            elif graph is not None and k > 0:
                 graph.replay()
             else:
                 self._spmv_overlapped(w, q)
@@ -630,15 +659,16 @@ class CGSolverHIP:
                     cur2.wait_event(ev_ar)
                     cur2.wait_event(evdone[k % (LAG + 1)])
                 with self.prof.span("update"):
-                    S.pipelined_fused(z, t, p, x, r, w, q, scal, self.partials,
+                    S.pipelined_fused(z, t, p, xi, r, w, q, scal, self.partials,
                                       n, first)
-                if graph_ok and k == 2:
+                if graph_ok and graph is None and k == 2:
                     # steady state (first=False): capture SpMV + fused update
                     graph = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(graph):
                         self._spmv_overlapped(w, q)
-                        S.pipelined_fused(z, t, p, x, r, w, q, scal,
+                        S.pipelined_fused(z, t, p, xi, r, w, q, scal,
                                           self.partials, n, False)
+                    self._graphs[wskey] = graph
             k += 1
             res.niterations = k
         if not converged:
@@ -648,6 +678,7 @@ class CGSolverHIP:
                     break
         torch.cuda.synchronize(self.device)
         res.tsolve = time.perf_counter() - t0
+        x.copy_(xi)
         if not converged and gamma_host is not None:
             res.rnrm2 = math.sqrt(max(gamma_host, 0.0))
         res.converged = converged
