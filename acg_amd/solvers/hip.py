@@ -272,7 +272,7 @@ class CGSolverHIP:
 
     def solve(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
               res_atol: float = 0.0, res_rtol: float = 1e-9,
-              use_graph: bool = True) -> SolveResult:
+              use_graph: bool = False) -> SolveResult:
         """Classic CG (reference acgsolverhip_solvempi, cghip.c:402-1159).
 
         ``x`` must be an nlocal vector (ghost tail included); ``b`` nowned.
@@ -462,21 +462,22 @@ class CGSolverHIP:
 
     def solve_pipelined(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
                         res_atol: float = 0.0, res_rtol: float = 1e-9,
-                        use_graph: bool = True,
+                        use_graph: bool = False,
                         megafuse: bool | None = None) -> SolveResult:
         """Pipelined (Ghysels-Vanroose) CG: ONE 2-double allreduce per
         iteration, overlapped with the halo + SpMV of q = A w
         (reference acgsolverhip_solve_pipelined, cghip.c:1187-1933).
 
-        ``use_graph``: on single-GPU runs, capture the steady-state
-        iteration (SpMV + fused update) into a hipGraph and replay it --
-        one graph launch instead of ~5 kernel launches per iteration
-        (launch-bound inner loops belong in graphs on MI355X).  The host
-        convergence test and its 8-byte D2H stay outside the graph.  The
-        replay enqueues the update before the host reads the previous
-        gamma, so on the converging iteration one extra (valid) update has
-        already been applied to x; the reported niterations/rnrm2 match
-        the eager path.
+        ``use_graph``: capture the steady-state iteration into a hipGraph
+        and replay it (single-GPU only; captured once per solver, cached).
+        Default OFF: with the same-stream lag pipeline the eager body is
+        1-3 cheap launches and hipGraphLaunch measures ~8 us/it SLOWER
+        across classic/megafused/matrix-free regimes (profiles/RESULTS.md)
+        -- graphs only pay when the body has many launches or launch
+        latency is high.  Replay semantics when enabled: the update is
+        enqueued before the host reads the previous gamma, so on the
+        converging iteration one extra (valid) update has already been
+        applied to x; reported niterations/rnrm2 match the eager path.
         """
         res = SolveResult(solver="cg-hip-pipelined", maxits=maxits,
                           res_atol=res_atol, res_rtol=res_rtol,

@@ -31,10 +31,11 @@ def main() -> int:
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--solver", choices=["auto", "pipelined", "classic"],
                     default="auto",
-                    help="auto = pipelined (after same-stream copy tuning it "
-                         "matches or beats classic everywhere: 543 vs 548 "
-                         "us/it interleaved on Queen, megafused on narrow "
-                         "rows, single overlapped allreduce multi-GPU)")
+                    help="auto = measured policy (profiles/RESULTS.md): "
+                         "classic on 1 GPU for wide rows (9n vs 13n update "
+                         "traffic; 504 vs 531 us/it interleaved on Queen), "
+                         "megafused pipelined for narrow rows, pipelined "
+                         "(single overlapped allreduce) for multi-GPU")
     ap.add_argument("--config", choices=["queen", "flan", "poisson7"], default="queen",
                     help="queen: 27-pt dof-3 Queen_4147-shaped (BASELINE configs 3-4); "
                          "flan: Flan_1565-shaped, 1.56M rows (BASELINE config 2); "
@@ -104,7 +105,11 @@ def main() -> int:
     b = torch.from_numpy(rloc.standard_normal(S.nowned)).to(device)
     x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device=device)
 
-    solver_kind = "pipelined" if args.solver == "auto" else args.solver
+    if args.solver == "auto":
+        solver_kind = ("classic" if ngpus == 1 and not solver.megafuse_auto
+                       else "pipelined")
+    else:
+        solver_kind = args.solver
     solve = solver.solve_pipelined if solver_kind == "pipelined" else solver.solve
 
     # warmup (untimed; also JITs RCCL channels and fills caches)
