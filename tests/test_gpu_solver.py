@@ -331,3 +331,37 @@ def test_slab_generated_gpu_solve():
     x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
     res = solver.solve(b, x, maxits=300, res_rtol=1e-9)
     assert res.converged, res.summary()
+
+
+def test_repeated_solves_deterministic_and_isolated(problem):
+    """Workspace vectors and captured graphs persist across solve calls:
+    repeated identical solves must be bitwise identical (deterministic
+    reductions + fixed kernel geometry), interleaving classic/pipelined/
+    graph variants must not leak state, and the caller's x must receive
+    the solution (it lives in an internal staggered buffer)."""
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A, S = problem
+    rng = np.random.default_rng(23)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+    solver = CGSolverHIP(S, device="cuda:0")
+
+    def run(fn, **kw):
+        x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64,
+                        device="cuda")
+        r = fn(b, x, maxits=60, res_rtol=0.0, **kw)
+        assert r.niterations == 60
+        return x[:S.nowned].cpu().numpy()
+
+    p1 = run(solver.solve_pipelined)
+    c1 = run(solver.solve)
+    pg = run(solver.solve_pipelined, use_graph=True)
+    p2 = run(solver.solve_pipelined)
+    c2 = run(solver.solve)
+    pg2 = run(solver.solve_pipelined, use_graph=True)
+    np.testing.assert_array_equal(p1, p2)     # bitwise repeatable
+    np.testing.assert_array_equal(pg, pg2)    # cached-graph replays too
+    np.testing.assert_array_equal(c1, c2)
+    np.testing.assert_allclose(p1, pg, rtol=1e-12, atol=1e-12)
+    np.testing.assert_allclose(p1, c1, rtol=1e-5, atol=1e-7)
+    assert np.all(np.isfinite(p1)) and np.abs(p1).max() > 0
