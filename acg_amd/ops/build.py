@@ -1,5 +1,8 @@
 """In-tree build of the gfx950 HIP kernel extension.
 
+Reference analog: the CMake HIP build (CMakeLists.txt `--offload-arch`,
+ACG_WITH_HIP) reduced to two direct compiler invocations.
+
 The built ``.so`` lives next to the sources (inside the package) so the
 repo snapshot carries it to GPU boxes; a JIT cache under ~/.cache would
 not travel.  Compilation is a direct ``hipcc --offload-arch=gfx950`` of

@@ -1,5 +1,9 @@
 """Tensor-level wrappers around the gfx950 HIP kernels (_acg_kernels.so).
 
+Reference analog: the BLAS/SpMV call layer of acg/cghip.c (hipsparse
+SpMV setup :463-585, hipblas dot/axpy wrappers) -- here every hot op is
+one of our own kernels (ops/kernels.hip cites the per-kernel analogs).
+
 Raw pointers + the caller's current HIP stream are passed to the
 extension; all type/shape checks happen here.  If the extension is
 missing on a machine with a GPU, import fails loudly (no eager fallback).
