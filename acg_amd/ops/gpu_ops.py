@@ -184,6 +184,23 @@ def stencil_pipe(mf, nrows_nodes: int, row0_node: int, border_base: int,
                           partials.data_ptr(), partials_off, mato, _stream())
 
 
+def spmv_bsell_daypx(bptr, bcol, bvals, nnodes: int, dof: int,
+                     pold: torch.Tensor, rvec: torch.Tensor,
+                     pnew: torch.Tensor, y: torch.Tensor,
+                     scal: torch.Tensor, partials: torch.Tensor,
+                     dotslot: int) -> None:
+    """Classic-CG fold: p_new = beta*p_old + r materialised inside the
+    BSELL SpMV (gather computes it on the fly), (p_new, t) dot fused.
+    beta = scal[RR]/scal[RR_PREV]; serial matA-only (no matO pass)."""
+    nslices = bptr.numel() - 1
+    if nslices <= 0:
+        return
+    K.spmv_bsell_daypx(nslices, nnodes, dof, bptr.data_ptr(), bcol.data_ptr(),
+                       bvals.data_ptr(), pold.data_ptr(), rvec.data_ptr(),
+                       pnew.data_ptr(), y.data_ptr(), scal.data_ptr(),
+                       partials.data_ptr(), dotslot, _stream())
+
+
 def zero_scalars(scal: torch.Tensor, i0: int = 0, count: int | None = None) -> None:
     K.zero_scalars(scal.data_ptr(), i0, count if count is not None else scal.numel() - i0,
                    _stream())

@@ -286,6 +286,71 @@ k_spmv_bsell(long nslices, long nnodes,
     }
 }
 
+// Classic-CG daypx folded into the BSELL SpMV (serial matA-only path):
+// instead of a separate p = beta p + r kernel (3n streams) the gather
+// computes beta*p_old[c] + r[c] on the fly and the row side materialises
+// p_new = beta*p_old + r into a SEPARATE buffer (ping-pong: folding into
+// one buffer would be a write-after-read race against other blocks'
+// gathers -- the separate daypx kernel's global barrier is exactly what
+// a single buffer needs).  beta = rr/rr_prev from the device scalar
+// slab (the previous update's finalize); the host seeds rr_prev = inf
+// before iteration 0 so beta = 0 reproduces p0 = r0.  The fused (p,t)
+// dot uses the in-register p_new row values.
+template <int DOF>
+__global__ void __launch_bounds__(BLOCK)
+k_spmv_bsell_daypx(long nslices, long nnodes,
+                   const long* __restrict__ bptr,
+                   const int* __restrict__ bcol,
+                   const double* __restrict__ bvals,
+                   const double* __restrict__ pold,
+                   const double* __restrict__ rvec,
+                   double* __restrict__ pnew,
+                   double* __restrict__ y,
+                   const double* __restrict__ scal,
+                   double* __restrict__ partials) {
+    const int lane = threadIdx.x & (WAVE - 1);
+    const long wslice = ((long)blockIdx.x * BLOCK + threadIdx.x) >> 6;
+    const long nw = ((long)gridDim.x * BLOCK) >> 6;
+    const double beta = scal[S_RR] / scal[S_RR_PREV];
+    double dacc = 0.0;
+    for (long s = wslice; s < nslices; s += nw) {
+        const long b0 = bptr[s];
+        const long blen = (bptr[s + 1] - b0) >> 6;
+        const int* __restrict__ c = bcol + b0 + lane;
+        const double* __restrict__ v = bvals + b0 * (DOF * DOF);
+        double acc[DOF];
+        #pragma unroll
+        for (int r = 0; r < DOF; ++r) acc[r] = 0.0;
+        for (long j = 0; j < blen; ++j) {
+            const int cb = ld_nt(c + j * WAVE);
+            double xv[DOF];
+            #pragma unroll
+            for (int cc = 0; cc < DOF; ++cc)
+                xv[cc] = beta * pold[(long)cb * DOF + cc]
+                         + rvec[(long)cb * DOF + cc];
+            const double* __restrict__ vj = v + j * (DOF * DOF) * WAVE;
+            #pragma unroll
+            for (int k = 0; k < DOF * DOF; ++k) {
+                const double a = ld_nt(vj + bval_off<DOF * DOF>(k, lane));
+                acc[k / DOF] += a * xv[k % DOF];
+            }
+        }
+        const long node = s * WAVE + lane;
+        if (node < nnodes) {
+            #pragma unroll
+            for (int r = 0; r < DOF; ++r) {
+                const double pn = beta * pold[node * DOF + r]
+                                  + rvec[node * DOF + r];
+                pnew[node * DOF + r] = pn;
+                y[node * DOF + r] = acc[r];
+                dacc += pn * acc[r];
+            }
+        }
+    }
+    dacc = block_reduce(dacc);
+    if (threadIdx.x == 0) partials[blockIdx.x] = dacc;
+}
+
 // device-side BSELL generation for the block-stencil slab (block-level
 // analog of k_stencil_rowlen / k_stencil_fill; matA only -- owned x owned)
 __global__ void __launch_bounds__(BLOCK)
@@ -796,6 +861,34 @@ void spmv_bsell(long nslices, long nnodes, int dof, uintptr_t bptr,
                            (int)blocks, (double*)scal, dotslot, dot_accum ? 1 : 0);
         check_hip("spmv_bsell_reduce");
     }
+}
+
+void spmv_bsell_daypx(long nslices, long nnodes, int dof, uintptr_t bptr,
+                      uintptr_t bcol, uintptr_t bvals, uintptr_t pold,
+                      uintptr_t rvec, uintptr_t pnew, uintptr_t y,
+                      uintptr_t scal, uintptr_t partials, int dotslot,
+                      uintptr_t stream) {
+    if (nnodes == 0) return;
+    long blocks = (nslices * WAVE + BLOCK - 1) / BLOCK;
+    if (blocks > MAXG) blocks = MAXG;
+    dim3 g((unsigned)blocks), b(BLOCK);
+    #define LBD(D) \
+        hipLaunchKernelGGL((k_spmv_bsell_daypx<D>), g, b, 0, (hipStream_t)stream, \
+            nslices, nnodes, (const long*)bptr, (const int*)bcol, \
+            (const double*)bvals, (const double*)pold, (const double*)rvec, \
+            (double*)pnew, (double*)y, (const double*)scal, (double*)partials)
+    switch (dof) {
+        case 2: LBD(2); break;
+        case 3: LBD(3); break;
+        case 4: LBD(4); break;
+        default: throw std::runtime_error("spmv_bsell_daypx: dof must be 2/3/4");
+    }
+    #undef LBD
+    check_hip("spmv_bsell_daypx");
+    hipLaunchKernelGGL(k_reduce_partials, dim3(1), dim3(BLOCK), 0,
+                       (hipStream_t)stream, (const double*)partials,
+                       (int)blocks, (double*)scal, dotslot, 0);
+    check_hip("spmv_bsell_daypx_reduce");
 }
 
 void stencil_blocklen(long nnodes, int gx, int gy, int gz, long nown_nodes,
@@ -1722,6 +1815,7 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.def("stencil_rowlen", &stencil_rowlen);
     m.def("stencil_fill", &stencil_fill);
     m.def("spmv_bsell", &spmv_bsell);
+    m.def("spmv_bsell_daypx", &spmv_bsell_daypx);
     m.def("stencil_blocklen", &stencil_blocklen);
     m.def("stencil_bfill", &stencil_bfill);
     m.def("stencil_spmv", &stencil_spmv);

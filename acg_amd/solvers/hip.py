@@ -272,7 +272,8 @@ class CGSolverHIP:
 
     def solve(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
               res_atol: float = 0.0, res_rtol: float = 1e-9,
-              use_graph: bool = False) -> SolveResult:
+              use_graph: bool = False,
+              fold_daypx: bool | None = None) -> SolveResult:
         """Classic CG (reference acgsolverhip_solvempi, cghip.c:402-1159).
 
         ``x`` must be an nlocal vector (ghost tail included); ``b`` nowned.
@@ -312,7 +313,17 @@ class CGSolverHIP:
             return res
         converged = False
         serial = self.comm is None or self.comm.size == 1
-        graph_ok = use_graph and serial and not self.prof.enabled
+        # daypx folded into the BSELL SpMV (serial matA-only): the gather
+        # computes beta*p_old + r on the fly and the row side materialises
+        # p_new into a ping-pong buffer -- the separate 3n-stream daypx
+        # kernel disappears (~3% of the Queen iteration).  beta needs
+        # rr_prev = inf seeded below so iteration 0 reproduces p0 = r0.
+        fold = (serial and self.bsell is not None and self.local.nnzO == 0
+                and fold_daypx is not False)
+        if fold:
+            p2 = self._workspace("classic", [("p2", True)])["p2"]
+            scal[S.S_RR_PREV] = math.inf
+        graph_ok = use_graph and serial and not self.prof.enabled and not fold
         graph = self._graphs.get("classic") if graph_ok else None
         # lag-2 convergence pipeline + hipGraph replay, mirroring
         # solve_pipelined (the host test runs for every iteration; the host
@@ -325,7 +336,17 @@ class CGSolverHIP:
                    for _ in range(LAG + 1)]
         evdone = [torch.cuda.Event() for _ in range(LAG + 1)]
 
-        def body():
+        def body(k: int = 0):
+            if fold:
+                bptr, bcol, bvals, dof = self.bsell
+                pold, pnew = (p, p2) if k % 2 == 0 else (p2, p)
+                with self.prof.span("spmvA"):
+                    S.spmv_bsell_daypx(bptr, bcol, bvals, n // dof, dof,
+                                       pold, r, pnew, t, scal,
+                                       self.partials, S.S_PT)
+                with self.prof.span("update_classic"):
+                    S.cg_fused_update(r, xi, pnew, t, scal, self.partials, n)
+                return
             # halo+split SpMV with the (p,t) reduction fused into both
             # passes (matA overwrites the slot, matO accumulates)
             self._spmv_overlapped(p, t, fuse_dotslot=S.S_PT)
@@ -364,7 +385,7 @@ class CGSolverHIP:
             if graph is not None:
                 graph.replay()
             else:
-                body()
+                body(k)
                 if graph_ok and k == 1:
                     graph = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(graph):

@@ -365,3 +365,30 @@ def test_repeated_solves_deterministic_and_isolated(problem):
     np.testing.assert_allclose(p1, pg, rtol=1e-12, atol=1e-12)
     np.testing.assert_allclose(p1, c1, rtol=1e-5, atol=1e-7)
     assert np.all(np.isfinite(p1)) and np.abs(p1).max() > 0
+
+
+def test_classic_daypx_fold_matches_unfolded(problem):
+    """Serial BSELL classic CG folds daypx into the SpMV (p ping-pong);
+    must match the unfolded path in iterations and solution."""
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A, S = problem
+    solver = CGSolverHIP(S, device="cuda:0")
+    if solver.bsell is None or S.nnzO > 0:
+        pytest.skip("fold needs serial BSELL matA-only")
+    rng = np.random.default_rng(31)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+
+    def run(**kw):
+        x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64,
+                        device="cuda")
+        res = solver.solve(b, x, maxits=500, res_rtol=1e-9, **kw)
+        assert res.converged, res.summary()
+        return x[:S.nowned].cpu().numpy(), res
+
+    xf, rf = run()                    # fold auto-on
+    xu, ru = run(fold_daypx=False)
+    assert abs(rf.niterations - ru.niterations) <= 2
+    np.testing.assert_allclose(xf, xu, rtol=1e-8, atol=1e-10)
+    xf2, _ = run()
+    np.testing.assert_array_equal(xf, xf2)  # fold path deterministic
