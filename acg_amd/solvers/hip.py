@@ -315,11 +315,13 @@ class CGSolverHIP:
         serial = self.comm is None or self.comm.size == 1
         # daypx folded into the BSELL SpMV (serial matA-only): the gather
         # computes beta*p_old + r on the fly and the row side materialises
-        # p_new into a ping-pong buffer -- the separate 3n-stream daypx
-        # kernel disappears (~3% of the Queen iteration).  beta needs
-        # rr_prev = inf seeded below so iteration 0 reproduces p0 = r0.
+        # p_new into a ping-pong buffer, eliminating the 3n-stream daypx
+        # kernel.  MEASURED NEGATIVE and therefore opt-in: doubling the
+        # gather footprint (p_old AND r) costs +60 us/it on Queen vs the
+        # ~15 us the daypx saves (553.6 vs 493.8 us/it interleaved) --
+        # the SpMV gather is the roofline resource, not kernel count.
         fold = (serial and self.bsell is not None and self.local.nnzO == 0
-                and fold_daypx is not False)
+                and fold_daypx is True)
         if fold:
             p2 = self._workspace("classic", [("p2", True)])["p2"]
             scal[S.S_RR_PREV] = math.inf

@@ -386,9 +386,9 @@ def test_classic_daypx_fold_matches_unfolded(problem):
         assert res.converged, res.summary()
         return x[:S.nowned].cpu().numpy(), res
 
-    xf, rf = run()                    # fold auto-on
-    xu, ru = run(fold_daypx=False)
+    xf, rf = run(fold_daypx=True)     # measured-negative, opt-in
+    xu, ru = run()
     assert abs(rf.niterations - ru.niterations) <= 2
     np.testing.assert_allclose(xf, xu, rtol=1e-8, atol=1e-10)
-    xf2, _ = run()
+    xf2, _ = run(fold_daypx=True)
     np.testing.assert_array_equal(xf, xf2)  # fold path deterministic
