@@ -29,7 +29,7 @@
 //      bounded spins -- cooperative grid.sync measured ~150 us/sync),
 //   6. halo pack gather (no unpack exists: ghosts are received in place).
 //
-// The scalar slab layout (fp64 slots) is shared with solvers/cg_hip.py:
+// The scalar slab layout (fp64 slots) is shared with solvers/hip.py:
 #define S_RR 0         // (r,r) current
 #define S_PT 1         // (p,t)
 #define S_RR_PREV 2    // (r,r) previous
