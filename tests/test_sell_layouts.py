@@ -75,3 +75,17 @@ def test_sell_sigma_layout(nrows, dens, seed):
     # and the padding must actually shrink vs sigma=1 for skewed rows
     plain = sell_from_csr(rowptr, colidx, vals)[0]
     assert int(sellptr[-1]) <= int(plain[-1])
+
+
+def test_pick_lanes_policy():
+    """Measured lanes-per-row policy (profiles/RESULTS.md: 16 lanes beats
+    64 by ~25% on ~80 nnz/row): ~6 nnz per lane target."""
+    from acg_amd.ops.gpu_ops import pick_lanes
+
+    assert pick_lanes(5) == 4
+    assert pick_lanes(24) == 4
+    assert pick_lanes(25) == 8
+    assert pick_lanes(80) == 16      # Queen-shaped
+    assert pick_lanes(7) == 4        # 7-pt Poisson
+    assert pick_lanes(150) == 32
+    assert pick_lanes(500) == 64
