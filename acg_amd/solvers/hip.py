@@ -485,22 +485,25 @@ class CGSolverHIP:
 
     def solve_pipelined(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
                         res_atol: float = 0.0, res_rtol: float = 1e-9,
-                        use_graph: bool = False,
+                        use_graph: bool | None = None,
                         megafuse: bool | None = None) -> SolveResult:
         """Pipelined (Ghysels-Vanroose) CG: ONE 2-double allreduce per
         iteration, overlapped with the halo + SpMV of q = A w
         (reference acgsolverhip_solve_pipelined, cghip.c:1187-1933).
 
         ``use_graph``: capture the steady-state iteration into a hipGraph
-        and replay it (single-GPU only; captured once per solver, cached).
-        Default OFF: with the same-stream lag pipeline the eager body is
-        1-3 cheap launches and hipGraphLaunch measures ~8 us/it SLOWER
-        across classic/megafused/matrix-free regimes (profiles/RESULTS.md)
-        -- graphs only pay when the body has many launches or launch
-        latency is high.  Replay semantics when enabled: the update is
-        enqueued before the host reads the previous gamma, so on the
-        converging iteration one extra (valid) update has already been
-        applied to x; reported niterations/rnrm2 match the eager path.
+        and replay it (captured once per solver, cached).  Default None =
+        measured policy: OFF on one GPU (the eager 1-3 launch body is ~8
+        us/it cheaper than hipGraphLaunch, profiles/RESULTS.md), ON for
+        multi-GPU where the eager iteration's ~100 us of host-side c10d /
+        launch work rivals the per-rank GPU time -- there the WHOLE
+        iteration (side-stream allreduce fork, RCCL halo, split SpMV,
+        fused update) is captured, with automatic permanent fallback to
+        eager if the backend cannot capture (e.g. gloo).  Replay
+        semantics: the update is enqueued before the host reads the
+        previous gamma, so on the converging iteration 1-2 extra (valid)
+        updates have already been applied to x; reported
+        niterations/rnrm2 match the eager path.
         """
         res = SolveResult(solver="cg-hip-pipelined", maxits=maxits,
                           res_atol=res_atol, res_rtol=res_rtol,
@@ -546,6 +549,11 @@ class CGSolverHIP:
         converged = False
         gamma_host = None
         serial = self.comm is None or self.comm.size == 1
+        if use_graph is None:
+            # measured policy: eager wins on one GPU (~8 us/it cheaper than
+            # hipGraphLaunch); captured iterations win the host-bound
+            # distributed loop (capture failure falls back to eager)
+            use_graph = not serial
         graph_ok = use_graph and serial and not self.prof.enabled
         graph = self._graphs.get(wskey) if graph_ok and not mega else None
         # megafused: even/odd w ping-pong graphs
@@ -635,9 +643,101 @@ class CGSolverHIP:
 
         ev_gd = torch.cuda.Event()
         ev_ar = torch.cuda.Event()
+        # Multi-GPU graph capture: the eager distributed iteration costs
+        # ~100 us of host work (c10d op lists for the halo, allreduce
+        # setup, kernel launches) which rivals the GPU time per iteration
+        # at 8-GPU Queen scale.  The ENTIRE steady-state iteration --
+        # side-stream allreduce fork, halo send/recv, split SpMV, fused
+        # update + finalize -- is captured once and replayed (RCCL
+        # collectives and event forks are capturable; the 8-byte gamma
+        # D2H stays outside, same-stream after the replay).  Capture is
+        # rank-local and records without executing, so no rank-sync is
+        # involved; ANY capture failure (e.g. the gloo test backend
+        # cannot capture) permanently falls back to the eager path for
+        # this solver, and mixed replay/eager ranks remain correct
+        # because replay issues the identical collective sequence.
+        dist_key = wskey + ":dist"  # dist bodies INCLUDE the allreduce --
+        dist_failed_key = wskey + ":capture_failed"  # never mix with serial
+        dist_graph_ok = (use_graph and not serial and not self.prof.enabled
+                         and not self._graphs.get(dist_failed_key, False))
+        dgraph = self._graphs.get(dist_key) if dist_graph_ok and not mega else None
+        dgraphs = (self._graphs.get(dist_key, [None, None])
+                   if dist_graph_ok and mega else [None, None])
+
+        # in-graph snapshot of the ALLREDUCED gamma: the replayed body both
+        # allreduces gamma_{k-1} and overwrites it (finalize), so the host
+        # copy after replay must read this intermediate, not scal itself --
+        # otherwise each rank would test its LOCAL gamma and ranks could
+        # break at different iterations (collective deadlock).
+        if "gsnap" not in ws:
+            ws["gsnap"] = torch.zeros(1, dtype=torch.float64,
+                                      device=self.device)
+        gsnap = ws["gsnap"]
+
+        def body_dist(wa=None, wb=None):
+            """One steady-state (first=False) distributed iteration in the
+            shape the capture needs (allreduce included)."""
+            if mega:
+                self._allreduce_slot(S.S_GAMMA, 2)
+                gsnap.copy_(scal[S.S_GAMMA:S.S_GAMMA + 1])
+                mega_body(wa, wb, False)
+            else:
+                cur = torch.cuda.current_stream(self.device)
+                ev_gd.record(cur)
+                self.allred_stream.wait_event(ev_gd)
+                with torch.cuda.stream(self.allred_stream):
+                    self._allreduce_slot(S.S_GAMMA, 2)
+                    ev_ar.record(self.allred_stream)
+                self._spmv_overlapped(w, q)
+                cur.wait_event(ev_ar)
+                gsnap.copy_(scal[S.S_GAMMA:S.S_GAMMA + 1])
+                S.pipelined_fused(z, t, p, xi, r, w, q, scal, self.partials,
+                                  n, False)
+
+        def try_capture_dist(kk):
+            nonlocal dgraph
+            if self._graphs.get(dist_failed_key):
+                return
+            try:
+                gg = torch.cuda.CUDAGraph()
+                if mega:
+                    wa, wb = (w, w2) if kk % 2 == 0 else (w2, w)
+                    with torch.cuda.graph(gg):
+                        body_dist(wa, wb)
+                    dgraphs[kk % 2] = gg
+                    self._graphs[dist_key] = dgraphs
+                else:
+                    with torch.cuda.graph(gg):
+                        body_dist()
+                    dgraph = gg
+                    self._graphs[dist_key] = gg
+            except Exception as e:
+                self._graphs[dist_failed_key] = True
+                torch.cuda.synchronize(self.device)
+                if (self.comm is None or self.comm.rank == 0):
+                    print(f"[acg_amd] distributed graph capture unavailable "
+                          f"({type(e).__name__}); eager iterations",
+                          file=__import__("sys").stderr)
+
         k = 0
         while k < maxits:
             first = (k == 0)
+            if not serial and dist_graph_ok:
+                g = dgraphs[k % 2] if mega else dgraph
+                if g is not None and k > 0:
+                    # whole iteration (incl. allreduce) is in the graph
+                    if k >= LAG and check(k - LAG):
+                        break
+                    g.replay()
+                    # same slot/value as the eager path: the ALLREDUCED
+                    # previous gamma, snapshotted inside the replay
+                    cur = torch.cuda.current_stream(self.device)
+                    j = k % (LAG + 1)
+                    hostbuf[j].copy_(gsnap, non_blocking=True)
+                    evdone[j].record(cur)
+                    k += 1
+                    res.niterations = k
+                    continue
             # ONE 2-double allreduce per iteration (gamma,delta adjacent).
             # Multi-GPU non-megafused: the allreduce rides a dedicated side
             # stream so SpMV(q = A w) -- which does not read the scalars --
@@ -706,6 +806,10 @@ class CGSolverHIP:
                         S.pipelined_fused(z, t, p, xi, r, w, q, scal,
                                           self.partials, n, False)
                     self._graphs[wskey] = graph
+            if dist_graph_ok and not serial and (
+                    (mega and k in (2, 3) and dgraphs[k % 2] is None)
+                    or (not mega and k == 2 and dgraph is None)):
+                try_capture_dist(k)
             k += 1
             res.niterations = k
         if not converged:
