@@ -31,4 +31,24 @@ try:
     print("CAPTURE-OK", float(t[0]), float(buf[0]))
 except Exception as e:
     print("CAPTURE-FAIL", type(e).__name__, str(e)[:200])
+
+# P2P (halo shape): grouped isend/irecv to self under capture
+try:
+    src = torch.arange(4, dtype=torch.float64, device="cuda")
+    dst = torch.zeros(4, dtype=torch.float64, device="cuda")
+    ops = [dist.P2POp(dist.irecv, dst, 0), dist.P2POp(dist.isend, src, 0)]
+    for r in dist.batch_isend_irecv(ops):
+        r.wait()
+    torch.cuda.synchronize()  # warm
+    g2 = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g2):
+        ops = [dist.P2POp(dist.irecv, dst, 0), dist.P2POp(dist.isend, src, 0)]
+        for r in dist.batch_isend_irecv(ops):
+            r.wait()
+    src += 1.0
+    g2.replay()
+    torch.cuda.synchronize()
+    print("P2P-CAPTURE-OK", dst.tolist())
+except Exception as e:
+    print("P2P-CAPTURE-FAIL", type(e).__name__, str(e)[:200])
 dist.destroy_process_group()
