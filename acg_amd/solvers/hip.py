@@ -24,6 +24,7 @@ re-design:
 from __future__ import annotations
 
 import math
+import os
 import time
 
 import numpy as np
@@ -658,7 +659,10 @@ class CGSolverHIP:
         # because replay issues the identical collective sequence.
         dist_key = wskey + ":dist"  # dist bodies INCLUDE the allreduce --
         dist_failed_key = wskey + ":capture_failed"  # never mix with serial
+        # ACG_DIST_GRAPH=0 is the operational kill switch for the captured
+        # distributed iteration (falls back to the eager path everywhere)
         dist_graph_ok = (use_graph and not serial and not self.prof.enabled
+                         and os.environ.get("ACG_DIST_GRAPH", "1") != "0"
                          and not self._graphs.get(dist_failed_key, False))
         dgraph = self._graphs.get(dist_key) if dist_graph_ok and not mega else None
         dgraphs = (self._graphs.get(dist_key, [None, None])
