@@ -68,8 +68,11 @@ def main() -> int:
     device = torch.device("cuda", local_rank % max(torch.cuda.device_count(), 1))
     torch.cuda.set_device(device)
     # init RCCL whenever torchrun launched us (even world=1: exercises the
-    # same bootstrap the driver's multi-GPU runs use)
-    comm = Comm("rccl", device=device) if "RANK" in os.environ else None
+    # same bootstrap the driver's multi-GPU runs use).  ACG_BENCH_COMM=gloo
+    # is test-only: it lets the full multi-rank bench flow run 2-process on
+    # ONE GPU via the CPU-staged gloo path (tests/test_gpu_dist_gloo.py).
+    kind = os.environ.get("ACG_BENCH_COMM", "rccl")
+    comm = Comm(kind, device=device) if "RANK" in os.environ else None
 
     if args.config == "queen":
         dof = args.dof or 3
@@ -127,8 +130,9 @@ def main() -> int:
     t1 = time.perf_counter()
     elapsed = t1 - t0
     # MAX over ranks
+    red_dev = device if kind == "rccl" else "cpu"
     if comm is not None:
-        et = torch.tensor([elapsed], dtype=torch.float64, device=device)
+        et = torch.tensor([elapsed], dtype=torch.float64, device=red_dev)
         import torch.distributed as dist
 
         dist.all_reduce(et, op=dist.ReduceOp.MAX)
@@ -140,7 +144,8 @@ def main() -> int:
     if comm is not None:
         import torch.distributed as dist
 
-        nt = torch.tensor([float(nnz_local)], dtype=torch.float64, device=device)
+        nt = torch.tensor([float(nnz_local)], dtype=torch.float64,
+                          device=red_dev)
         dist.all_reduce(nt, op=dist.ReduceOp.SUM)
         nnz_global = float(nt.item())
     else:

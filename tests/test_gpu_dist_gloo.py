@@ -196,3 +196,30 @@ def test_gpu_cg_ws4_rgb():
     x_ref = spla.spsolve(A.to_scipy_full().tocsc(), b_global)
     for _, (owned_global, xloc, _nit) in results.items():
         np.testing.assert_allclose(xloc, x_ref[owned_global], rtol=1e-6, atol=1e-8)
+
+
+def test_bench_2proc_end_to_end(tmp_path):
+    """The driver-facing bench contract at world_size=2 on one GPU
+    (ACG_BENCH_COMM=gloo test override): full flow -- slab generation per
+    rank, multi-rank solve, MAX-over-ranks timing, nnz SUM, one JSON line
+    from rank 0."""
+    import json
+    import subprocess
+    import sys as _sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    env = dict(__import__("os").environ)
+    env["ACG_BENCH_COMM"] = "gloo"
+    cmd = [_sys.executable, "-m", "torch.distributed.run", "--nnodes", "1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", "29640", "bench.py", "--gpus", "2",
+           "--steps", "5", "--warmup", "2", "--grid", "14"]
+    r = subprocess.run(cmd, capture_output=True, text=True, cwd=repo,
+                       env=env, timeout=420)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    line = next(l for l in r.stdout.splitlines() if l.startswith("{"))
+    d = json.loads(line)
+    assert d["n_gpus"] == 2 and d["steps"] == 5
+    assert d["value"] > 0 and d["config"]["solver"] == "cg-pipelined"
+    assert d["config"]["rows"] == 3 * 14 ** 3
