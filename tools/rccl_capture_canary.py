@@ -32,7 +32,11 @@ try:
 except Exception as e:
     print("CAPTURE-FAIL", type(e).__name__, str(e)[:200])
 
-# P2P (halo shape): grouped isend/irecv to self under capture
+# P2P (halo shape): grouped isend/irecv to self under capture.
+# KNOWN: at world_size=1 NCCL hard-aborts the process on a SELF-send
+# (invalid usage) before the capture question arises -- this section only
+# yields data on a >=2-GPU node.  The solver's capture fallback makes a
+# P2P-capture 'no' harmless either way.
 try:
     src = torch.arange(4, dtype=torch.float64, device="cuda")
     dst = torch.zeros(4, dtype=torch.float64, device="cuda")
