@@ -328,7 +328,7 @@ class CGSolverHIP:
             scal[S.S_RR_PREV] = math.inf
         graph_ok = use_graph and serial and not self.prof.enabled and not fold
         graph = self._graphs.get("classic") if graph_ok else None
-        # lag-2 convergence pipeline + hipGraph replay, mirroring
+        # lag-2 convergence pipeline (+ optional hipGraph replay), mirroring
         # solve_pipelined (the host test runs for every iteration; the host
         # reads the value two iterations late).  Classic's rr copy lands at
         # the END of its iteration, so LAG=1 would make the host wake
