@@ -32,10 +32,10 @@ def main() -> int:
     ap.add_argument("--solver", choices=["auto", "pipelined", "classic"],
                     default="auto",
                     help="auto = measured policy (profiles/RESULTS.md): "
-                         "classic on 1 GPU for wide rows (9n vs 13n update "
-                         "traffic; 504 vs 531 us/it interleaved on Queen), "
-                         "megafused pipelined for narrow rows, pipelined "
-                         "(single overlapped allreduce) for multi-GPU")
+                         "megafused pipelined for narrow rows; pipelined "
+                         "(overlapped allreduce) for multi-GPU; on 1 GPU "
+                         "wide rows a 30-it warmup probe picks classic vs "
+                         "pipelined (their ordering flips per instance)")
     ap.add_argument("--config", choices=["queen", "flan", "poisson7"], default="queen",
                     help="queen: 27-pt dof-3 Queen_4147-shaped (BASELINE configs 3-4); "
                          "flan: Flan_1565-shaped, 1.56M rows (BASELINE config 2); "
