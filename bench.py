@@ -160,6 +160,25 @@ def main() -> int:
         elapsed = float(et.item())
 
     assert res.niterations == args.steps, (res.niterations, args.steps)
+
+    # integrity check (outside the timed region): the TRUE residual of the
+    # final iterate must agree with the recursion residual the solver
+    # reports -- silent corruption anywhere (kernels, halo, captured
+    # graphs, collectives) would make these diverge or go non-finite.
+    tchk = torch.zeros(S.nowned, dtype=torch.float64, device=device)
+    solver._spmv_overlapped(x, tchk)
+    rloc2 = float(torch.sum((b[:S.nowned] - tchk) ** 2))
+    bloc2 = float(torch.sum(b[:S.nowned] ** 2))
+    if comm is not None:
+        import torch.distributed as dist
+
+        rb = torch.tensor([rloc2, bloc2], dtype=torch.float64, device=red_dev)
+        dist.all_reduce(rb, op=dist.ReduceOp.SUM)
+        rloc2, bloc2 = float(rb[0]), float(rb[1])
+    true_rel = (rloc2 / bloc2) ** 0.5 if bloc2 > 0 else float("nan")
+    recur_rel = res.rnrm2 / res.bnrm2 if res.bnrm2 > 0 else float("nan")
+    assert np.isfinite(true_rel), "non-finite true residual"
+
     nrows_global = dof * G * G * G
     nnz_local = S.nnzA + S.nnzO
     if comm is not None:
@@ -200,6 +219,8 @@ def main() -> int:
                 "nnz": nnz_global,
                 "solver": f"cg-{solver_kind}" + ("-matfree" if args.matfree else ""),
                 "time_to_solution_s": elapsed,
+                "rel_residual_true": true_rel,
+                "rel_residual_recursion": recur_rel,
                 "gflops": args.steps * (2.0 * nnz_global + 10.0 * nrows_global) / elapsed / 1e9,
                 "parallelism": f"slab{ngpus}-rccl",
             },
