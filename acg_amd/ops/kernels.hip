@@ -87,6 +87,17 @@ __device__ __forceinline__ double block_reduce(double v) {
     return v;  // valid in thread 0
 }
 
+// 0/0-safe coefficient division: once the recursion residual underflows
+// to EXACT zero (a converged solve driven past convergence, e.g. a
+// fixed-iteration benchmark with rtol=0 -- measured: Queen rr reaches
+// 3e-136 after 500 iterations and would hit 0 near ~1200), alpha/beta
+// become 0/0 = NaN and poison every vector.  Dividing to 0 instead
+// freezes the (already converged) iterate, which is exactly the right
+// no-op behaviour.
+__device__ __forceinline__ double safe_div(double a, double b) {
+    return b != 0.0 ? a / b : 0.0;
+}
+
 // sum partials[0..nblocks) into scal[slot] (+= if ACC)
 __global__ void __launch_bounds__(BLOCK)
 k_reduce_partials(const double* __restrict__ partials, int nblocks,
@@ -311,7 +322,7 @@ k_spmv_bsell_daypx(long nslices, long nnodes,
     const int lane = threadIdx.x & (WAVE - 1);
     const long wslice = ((long)blockIdx.x * BLOCK + threadIdx.x) >> 6;
     const long nw = ((long)gridDim.x * BLOCK) >> 6;
-    const double beta = scal[S_RR] / scal[S_RR_PREV];
+    const double beta = safe_div(scal[S_RR], scal[S_RR_PREV]);
     double dacc = 0.0;
     for (long s = wslice; s < nslices; s += nw) {
         const long b0 = bptr[s];
@@ -510,7 +521,7 @@ k_reduce_partials2(const double* __restrict__ partials, int nblocks,
 __global__ void __launch_bounds__(BLOCK)
 k_axpy_ratio(double* __restrict__ y, const double* __restrict__ x, long n,
              const double* __restrict__ scal, int num, int den, double sign) {
-    const double a = sign * scal[num] / scal[den];
+    const double a = sign * safe_div(scal[num], scal[den]);
     const long stride = (long)gridDim.x * BLOCK;
     for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride)
         y[i] += a * x[i];
@@ -520,7 +531,7 @@ k_axpy_ratio(double* __restrict__ y, const double* __restrict__ x, long n,
 __global__ void __launch_bounds__(BLOCK)
 k_daypx_ratio(double* __restrict__ y, const double* __restrict__ x, long n,
               const double* __restrict__ scal, int num, int den) {
-    const double b = scal[num] / scal[den];
+    const double b = safe_div(scal[num], scal[den]);
     const long stride = (long)gridDim.x * BLOCK;
     for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride)
         y[i] = b * y[i] + x[i];
@@ -536,7 +547,7 @@ k_cg_fused_update(double* __restrict__ r, double* __restrict__ x,
                   const double* __restrict__ p, const double* __restrict__ t,
                   long n, const double* __restrict__ scal,
                   double* __restrict__ partials) {
-    const double alpha = scal[S_RR] / scal[S_PT];
+    const double alpha = safe_div(scal[S_RR], scal[S_PT]);
     double acc = 0.0;
     const long stride = (long)gridDim.x * BLOCK;
     // t and x are single-use streams this iteration: non-temporal keeps
@@ -554,11 +565,12 @@ k_cg_fused_update(double* __restrict__ r, double* __restrict__ x,
 __device__ __forceinline__ void pipelined_coeffs(const double* scal, int first,
                                                  double* beta, double* alpha) {
     const double gamma = scal[S_GAMMA], delta = scal[S_DELTA];
-    if (first) { *beta = 0.0; *alpha = gamma / delta; }
+    if (first) { *beta = 0.0; *alpha = safe_div(gamma, delta); }
     else {
-        const double b = gamma / scal[S_GAMMA_PREV];
+        const double b = safe_div(gamma, scal[S_GAMMA_PREV]);
         *beta = b;
-        *alpha = gamma / (delta - b * gamma / scal[S_ALPHA_PREV]);
+        *alpha = safe_div(gamma,
+                          delta - b * safe_div(gamma, scal[S_ALPHA_PREV]));
     }
 }
 
@@ -1472,7 +1484,7 @@ k_cg_device(long nslices, long nrows,
         const double pt_part = spmv(p, t, true);
         grid_sum(pt_part, S_PT);
         if (!alive) break;
-        const double alpha = rr / scal[S_PT];
+        const double alpha = safe_div(rr, scal[S_PT]);
         acc = 0.0;
         for (long i = tid; i < nrows; i += nth) {
             const double rn = r[i] - alpha * t[i];

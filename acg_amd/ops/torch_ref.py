@@ -228,22 +228,29 @@ def dot2(r, w, partials, scal, n, accumulate=False) -> None:
         scal[S_DELTA] = d
 
 
+def _sdiv(a, b):
+    # 0/0-safe coefficient division (see kernels.hip safe_div): a converged
+    # solve driven past convergence underflows the recursion residual to
+    # exact 0; dividing to 0 freezes the iterate instead of NaN-poisoning it
+    return a / b if float(b) != 0.0 else torch.zeros_like(a) if torch.is_tensor(a) else 0.0
+
+
 def axpy_ratio(y, x, scal, num, den, sign=1.0, n=None) -> None:
     n = y.numel() if n is None else n
-    a = sign * float(scal[num]) / float(scal[den])
+    a = sign * _sdiv(float(scal[num]), float(scal[den]))
     y[:n] += a * x[:n]
 
 
 def daypx_ratio(y, x, scal, num, den, n=None) -> None:
     n = y.numel() if n is None else n
-    b = float(scal[num]) / float(scal[den])
+    b = _sdiv(float(scal[num]), float(scal[den]))
     y[:n] = b * y[:n] + x[:n]
 
 
 def cg_fused_update(r, x, p, t, scal, partials, n) -> None:
     """alpha = rr/pt; update r,x; then rotate rr->rr_prev and publish the
     new (r,r) (matches k_cg_fused_update + k_cg_finalize)."""
-    alpha = float(scal[S_RR]) / float(scal[S_PT])
+    alpha = _sdiv(float(scal[S_RR]), float(scal[S_PT]))
     r[:n] -= alpha * t[:n]
     x[:n] += alpha * p[:n]
     scal[S_RR_PREV] = scal[S_RR].clone()
@@ -254,9 +261,9 @@ def _pipelined_coeffs(scal, first: bool):
     gamma = float(scal[S_GAMMA])
     delta = float(scal[S_DELTA])
     if first:
-        return 0.0, gamma / delta
-    beta = gamma / float(scal[S_GAMMA_PREV])
-    alpha = gamma / (delta - beta * gamma / float(scal[S_ALPHA_PREV]))
+        return 0.0, _sdiv(gamma, delta)
+    beta = _sdiv(gamma, float(scal[S_GAMMA_PREV]))
+    alpha = _sdiv(gamma, delta - beta * _sdiv(gamma, float(scal[S_ALPHA_PREV])))
     return beta, alpha
 
 

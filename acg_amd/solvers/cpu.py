@@ -97,7 +97,7 @@ class CGSolverCPU:
         for k in range(maxits):
             self._spmv(p, t)
             pt = self._dot(p, t)
-            alpha = rr / pt
+            alpha = rr / pt if pt != 0.0 else 0.0  # 0/0 at underflow: freeze
             r -= alpha * t
             x[:n] += alpha * p[:n]
             rr_new = self._dot(r, r)
@@ -113,7 +113,7 @@ class CGSolverCPU:
                 res.converged = True
                 res.rnrm2 = math.sqrt(rr_new)
                 break
-            beta = rr_new / rr
+            beta = rr_new / rr if rr != 0.0 else 0.0
             p[:n] = r + beta * p[:n]
             rr = rr_new
             res.rnrm2 = math.sqrt(rr_new)
@@ -160,10 +160,12 @@ class CGSolverCPU:
                 break
             self._spmv(w, q)  # q = A w (halo on w inside)
             if k == 0:
-                beta, alpha = 0.0, gamma / delta
+                beta = 0.0
+                alpha = gamma / delta if delta != 0.0 else 0.0
             else:
-                beta = gamma / gamma_prev
-                alpha = gamma / (delta - beta * gamma / alpha_prev)
+                beta = gamma / gamma_prev if gamma_prev != 0.0 else 0.0
+                den = delta - beta * (gamma / alpha_prev if alpha_prev != 0.0 else 0.0)
+                alpha = gamma / den if den != 0.0 else 0.0
             z[:] = q + beta * z
             t[:] = w[:n] + beta * t
             p[:] = r[:n] + beta * p

@@ -96,3 +96,25 @@ def test_general_symmetry_rejected():
                 rowidx=np.array([0]), colidx=np.array([1]), a=np.array([1.0]))
     with pytest.raises(AcgError):
         SymCSRMatrix.from_mtxfile(m)
+
+
+def test_cpu_solver_survives_residual_underflow():
+    """Driven far past convergence with rtol=0 (fixed-iteration benches),
+    the recursion residual underflows to exact 0: coefficients must
+    freeze (0) instead of becoming 0/0 = NaN."""
+    import torch
+
+    from acg_amd.gen import STENCIL_5PT_2D, stencil_global
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.solvers.cpu import CGSolverCPU
+
+    A = stencil_global(12, 12, 1, STENCIL_5PT_2D)
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    rng = np.random.default_rng(0)
+    b = torch.from_numpy(rng.standard_normal(S.nowned))
+    for method in ("solve", "solve_pipelined"):
+        solver = CGSolverCPU(S)
+        x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64)
+        res = getattr(solver, method)(b, x, maxits=3000, res_rtol=0.0)
+        assert res.niterations == 3000, (method, res.niterations)
+        assert torch.isfinite(x).all(), method
