@@ -8,8 +8,8 @@ module implements its role natively:
   matrices such as the Poisson and Queen-like stencil operators the
   benchmarks use — minimises halo for banded orderings).
 - ``rgb``: recursive graph bisection via BFS level structures from a
-  pseudo-peripheral vertex, with balanced median split — a METIS-recursive
-  stand-in for general irregular graphs.
+  pseudo-peripheral vertex, with weighted splits for arbitrary part
+  counts — a METIS-recursive stand-in for general irregular graphs.
 
 A precomputed partition vector can also be read from / written to a
 Matrix Market integer array file, compatible with the reference's
@@ -58,13 +58,16 @@ def _pseudo_peripheral(G, start: int) -> int:
     return v
 
 
-def _bisect(G, nodes: np.ndarray, rng) -> tuple[np.ndarray, np.ndarray]:
-    """Split ``nodes`` into two balanced halves along a BFS level structure."""
+def _bisect(G, nodes: np.ndarray, rng,
+            nleft: int | None = None) -> tuple[np.ndarray, np.ndarray]:
+    """Split ``nodes`` along a BFS level structure: first ``nleft`` (default
+    half) in BFS order go left -- weighted splits support arbitrary part
+    counts like METIS-recursive."""
     from scipy.sparse.csgraph import breadth_first_order
 
     sub = G[nodes][:, nodes]
     n = len(nodes)
-    half = n // 2
+    half = n // 2 if nleft is None else int(nleft)
     # BFS over the subgraph (may be disconnected: loop over components)
     visited = np.zeros(n, dtype=bool)
     orderall = np.empty(n, dtype=np.int64)
@@ -103,8 +106,8 @@ def partition_rows(A, nparts: int, seed: int = 0, method: str = "auto") -> np.nd
         part = ((np.arange(n, dtype=np.int64) * nparts) // n).astype(np.int32)
         return part
     if method == "rgb":
-        if nparts & (nparts - 1):
-            raise AcgError(ErrCode.NOT_SUPPORTED, "rgb requires power-of-two nparts")
+        if nparts > n:
+            raise AcgError(ErrCode.INVALID_VALUE, "more parts than rows")
         G = _full_adjacency(A)
         rng = np.random.default_rng(seed)
         part = np.zeros(n, dtype=np.int32)
@@ -114,9 +117,13 @@ def partition_rows(A, nparts: int, seed: int = 0, method: str = "auto") -> np.nd
             if k == 1:
                 part[nodes] = base
                 continue
-            left, right = _bisect(G, nodes, rng)
-            groups.append((left, base, k // 2))
-            groups.append((right, base + k // 2, k // 2))
+            # weighted bisection (kl:kr) supports arbitrary nparts, like
+            # METIS_PartGraphRecursive
+            kl = (k + 1) // 2
+            kr = k - kl
+            left, right = _bisect(G, nodes, rng, (len(nodes) * kl) // k)
+            groups.append((left, base, kl))
+            groups.append((right, base + kl, kr))
         return part
     raise AcgError(ErrCode.NOT_SUPPORTED, f"partition method {method!r}")
 

@@ -38,3 +38,20 @@ def test_random_partition_fuzz():
                                        rtol=1e-10, atol=1e-9)
             covered[S.owned_global] = True
         assert covered.all()
+
+
+def test_partition_deterministic_and_balanced():
+    """Same seed -> identical partition; parts stay within 2x of perfect
+    balance for both methods (the reference's METIS gives ~1.03; our
+    stand-ins must at least never starve a rank)."""
+    from acg_amd.part import partition_rows
+
+    A = stencil_global(9, 8, 7, STENCIL_27PT_3D)
+    for method in ("block", "rgb"):
+        for nparts in (2, 5, 8):
+            p1 = partition_rows(A, nparts, method=method, seed=3)
+            p2 = partition_rows(A, nparts, method=method, seed=3)
+            np.testing.assert_array_equal(p1, p2)
+            counts = np.bincount(p1, minlength=nparts)
+            assert counts.min() > 0
+            assert counts.max() <= 2.0 * A.n / nparts, (method, nparts, counts)
