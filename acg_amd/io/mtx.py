@@ -262,6 +262,11 @@ def write_mtx(path_or_file, m: MtxFile, binary: bool = False, idxsize: int = 64,
         if m.object == "matrix" and m.format == "coordinate":
             _w(f"{m.nrows} {m.ncols} {m.nnz}\n")
             if binary:
+                if m.field_ not in ("real", "integer"):
+                    # matches the reader and mtx2bin (--double/--integer):
+                    # the reference binary layout stores a value array
+                    raise AcgError(ErrCode.NOT_SUPPORTED,
+                                   "binary supports real|integer fields")
                 idt = np.int32 if idxsize == 32 else np.int64
                 f.write((m.rowidx + 1).astype(idt).tobytes())
                 f.write((m.colidx + 1).astype(idt).tobytes())
