@@ -87,3 +87,40 @@ def test_array_vector_roundtrip_fuzz(tmp_path, container):
             write_mtx(path, m)
             r = read_mtx(path)
         np.testing.assert_allclose(np.asarray(r.a), a, rtol=0, atol=1e-14)
+
+
+def test_numfmt_grammar_fuzz():
+    """Random printf specifiers over the reference grammar (fmtspec.c):
+    every accepted spec must format like C/python %-formatting; malformed
+    ones must be rejected, never crash."""
+    import itertools
+
+    from acg_amd.utils.numfmt import FmtSpec, parse_numfmt
+
+    rng = np.random.default_rng(11)
+    convs = "eEfFgG"
+    vals = [0.0, 1.0, -1.5, 3.141592653589793e-8, 2.718e120, -7e-300]
+    for _ in range(200):
+        flags = "".join(rng.choice(list("-+ #0"),
+                                   size=rng.integers(0, 3), replace=False))
+        width = str(rng.integers(0, 25)) if rng.random() < 0.6 else ""
+        prec = f".{rng.integers(0, 18)}" if rng.random() < 0.7 else ""
+        conv = convs[rng.integers(len(convs))]
+        spec = f"%{flags}{width}{prec}{conv}"
+        f = parse_numfmt(spec)
+        for v in vals:
+            got = f.format(v)
+            want = spec.replace("'", "") % v
+            assert got == want, (spec, v, got, want)
+    # hex-float conversions format without crashing and round-trip
+    fa = parse_numfmt("%a")
+    for v in vals:
+        assert float.fromhex(fa.format(v)) == v
+    # malformed specs are rejected
+    for bad in ("%d", "%.5", "%", "abc", "%5.2x", "%ld", "%*g", "%.*e", "%%g"):
+        try:
+            FmtSpec(bad)
+            rejected = False
+        except ValueError:
+            rejected = True
+        assert rejected, bad
