@@ -156,3 +156,23 @@ def test_symcsr_assembly_fuzz():
         # dsymv agrees too
         x = rng.standard_normal(n)
         np.testing.assert_allclose(A.dsymv(x), full @ x, rtol=1e-12, atol=1e-12)
+
+
+def test_bsell_native_matches_numpy_fallback(monkeypatch):
+    """bsell_from_csr has a native C++ block-merge path and a numpy unique
+    fallback: identical outputs on random dof-blocked matrices."""
+    import sys
+
+    from acg_amd.gen import queen_like_spec, stencil_global
+    from acg_amd.ops import torch_ref
+    from acg_amd.part import extract_subdomains, partition_rows
+
+    A = stencil_global(5, 4, 6, queen_like_spec(3))
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    native = torch_ref.bsell_from_csr(S.A_rowptr, S.A_colidx, S.A_vals, 3)
+    monkeypatch.setitem(sys.modules, "acg_amd.host._acg_host", None)
+    fallback = torch_ref.bsell_from_csr(S.A_rowptr, S.A_colidx, S.A_vals, 3)
+    assert native is not None and fallback is not None
+    for a, b in zip(native[:3], fallback[:3]):
+        np.testing.assert_array_equal(np.asarray(a), np.asarray(b))
+    assert native[3] == fallback[3]
