@@ -417,3 +417,31 @@ def test_gpu_solver_survives_residual_underflow(problem, method):
     # classic floors at ~1e-15; pipelined's recursion drift leaves the
     # frozen iterate at ~1e-12 true residual (textbook behaviour)
     assert rel < 1e-10, rel
+
+
+@pytest.mark.parametrize("solver_name", ["acg", "acg-pipelined", "acg-device"])
+def test_cli_gpu_end_to_end(tmp_path, capsys, monkeypatch, solver_name):
+    """Full CLI pipeline on a real GPU: mtx file -> assembly -> extraction
+    -> CGSolverHIP (all three GPU solver types) -> solution to stdout."""
+    from acg_amd import cli
+    from acg_amd.gen import STENCIL_5PT_2D, stencil_global
+    from acg_amd.io.mtx import MtxFile, write_mtx
+
+    A = stencil_global(24, 24, 1, STENCIL_5PT_2D)
+    rows = np.repeat(np.arange(A.n), np.diff(A.rowptr))
+    m = MtxFile(object="matrix", format="coordinate", field_="real",
+                symmetry="symmetric", nrows=A.n, ncols=A.n,
+                nnz=A.nnz_stored, rowidx=A.colidx.astype(np.int64),
+                colidx=rows.astype(np.int64), a=A.vals)
+    path = tmp_path / "p.mtx"
+    write_mtx(path, m)
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    monkeypatch.delenv("RANK", raising=False)
+    rc = cli.main([str(path), "--solver", solver_name,
+                   "--manufactured-solution", "--max-iterations", "2000",
+                   "--residual-rtol", "1e-9", "-v"])
+    out = capsys.readouterr()
+    assert rc == 0, out.err
+    assert out.out.startswith("%%MatrixMarket matrix array real general")
+    vals = np.array([float(v) for v in out.out.strip().splitlines()[2:]])
+    assert len(vals) == A.n and np.isfinite(vals).all()
