@@ -79,6 +79,36 @@ def make_parser() -> argparse.ArgumentParser:
     return p
 
 
+def _vector_from_mtx(m, n: int, what: str) -> np.ndarray:
+    """Dense fp64 vector from an MtxFile: array format positionally, or
+    coordinate (sparse) format scattered through rowidx -- never assign
+    coordinate values positionally (a permuted file would silently produce
+    a wrong b; reference validates at acg-hip.c:1786-2021)."""
+    vals = np.asarray(m.a, dtype=np.float64)
+    if m.format == "array":
+        if m.ncols not in (0, 1) and m.nrows * m.ncols != n:
+            raise AcgError(ErrCode.INVALID_VALUE,
+                           f"{what}: array shape {m.nrows}x{m.ncols} != {n}")
+        if len(vals) != n:
+            raise AcgError(ErrCode.INVALID_VALUE,
+                           f"{what}: length {len(vals)} != matrix rows {n}")
+        return vals
+    if m.format == "coordinate":
+        if m.ncols != 1:
+            raise AcgError(ErrCode.INVALID_VALUE,
+                           f"{what}: coordinate vector must have 1 column")
+        if m.nrows != n:
+            raise AcgError(ErrCode.INVALID_VALUE,
+                           f"{what}: length {m.nrows} != matrix rows {n}")
+        out = np.zeros(n, dtype=np.float64)
+        ri = np.asarray(m.rowidx, dtype=np.int64)
+        if len(ri) and (ri.min() < 0 or ri.max() >= n):
+            raise AcgError(ErrCode.INVALID_VALUE, f"{what}: row index out of range")
+        out[ri] = vals
+        return out
+    raise AcgError(ErrCode.INVALID_FORMAT, f"{what}: unsupported format {m.format!r}")
+
+
 def main(argv=None) -> int:
     args = make_parser().parse_args(argv)
     import torch
@@ -159,9 +189,7 @@ def main(argv=None) -> int:
                 b_global = A.dsymv(xsol)
             elif args.b:
                 mb = read_mtx(args.b, gzipped=args.gzip)
-                b_global = np.asarray(mb.a, dtype=np.float64)
-                if len(b_global) != A.n:
-                    raise AcgError(ErrCode.INVALID_VALUE, "b length mismatch")
+                b_global = _vector_from_mtx(mb, A.n, "b")
                 xsol = None
             else:
                 b_global = np.ones(A.n, dtype=np.float64)
@@ -169,7 +197,7 @@ def main(argv=None) -> int:
             x0_global = None
             if args.x0:
                 mx = read_mtx(args.x0, gzipped=args.gzip)
-                x0_global = np.asarray(mx.a, dtype=np.float64)
+                x0_global = _vector_from_mtx(mx, A.n, "x0")
             b_pieces = [(b_global[S.owned_global],
                          None if x0_global is None else x0_global[S.owned_global])
                         for S in systems]
@@ -211,6 +239,7 @@ def main(argv=None) -> int:
     # ---- solve (reference acg-hip.c:2192-2247)
     err = None
     res = None
+    solver = None
     try:
         if solver_name in ("scipy", "scipy-pipelined"):
             from .solvers.oracle import solve_scipy
@@ -234,10 +263,26 @@ def main(argv=None) -> int:
                 f"({time.perf_counter() - t0:.2f}s)")
             b = b.to(device)
             x = x.to(device)
+            diff_requested = args.diff_atol > 0 or args.diff_rtol > 0
+            if diff_requested and solver_name != "acg":
+                # reference parity: the GPU pipelined/device solvers REJECT
+                # diff tolerances (cghip.c:427, 1212) rather than silently
+                # iterating to maxits; classic implements them (see
+                # CGSolverHIP.solve diff mode)
+                raise AcgError(ErrCode.NOT_SUPPORTED,
+                               f"--diff-atol/--diff-rtol are not supported "
+                               f"by {solver_name} (use acg or cpu)")
             if args.warmup:
-                meth = solver.solve_pipelined if solver_name == "acg-pipelined" \
-                    else solver.solve
-                meth(b, x.clone(), maxits=args.warmup, res_rtol=0.0)
+                if solver_name == "acg-pipelined":
+                    solver.solve_pipelined(b, x.clone(), maxits=args.warmup,
+                                           res_rtol=0.0)
+                elif solver_name == "acg-device":
+                    # warm the actual cooperative kernel (reference warms the
+                    # kernel it will time, cg-kernels-hip.hip:1925-1989)
+                    solver.solve_device(b, x.clone(), maxits=args.warmup,
+                                        res_rtol=0.0)
+                else:
+                    solver.solve(b, x.clone(), maxits=args.warmup, res_rtol=0.0)
             if solver_name == "acg-pipelined":
                 res = solver.solve_pipelined(b, x, maxits=args.max_iterations,
                                              res_atol=args.residual_atol,
@@ -249,7 +294,9 @@ def main(argv=None) -> int:
             else:
                 res = solver.solve(b, x, maxits=args.max_iterations,
                                    res_atol=args.residual_atol,
-                                   res_rtol=args.residual_rtol)
+                                   res_rtol=args.residual_rtol,
+                                   diff_atol=args.diff_atol,
+                                   diff_rtol=args.diff_rtol)
         else:
             from .solvers.cpu import CGSolverCPU
 
@@ -272,6 +319,38 @@ def main(argv=None) -> int:
     from .solvers.profiling import write_stats
 
     write_stats(res, S, comm, file=sys.stderr)
+
+    # true-residual integrity check: ||b - A x|| of the returned iterate,
+    # independent of the solver's residual recursion (bench.py computes the
+    # same; silent corruption anywhere would make these diverge)
+    if res is not None:
+        try:
+            if gpu_solver:
+                tchk = _t.zeros(S.nowned, dtype=_t.float64, device=device)
+                solver._spmv_overlapped(x, tchk)
+            else:
+                from .solvers.cpu import CGSolverCPU
+
+                chk = solver if isinstance(solver, CGSolverCPU) \
+                    else CGSolverCPU(S, comm=comm)
+                tchk = _t.zeros(S.nowned, dtype=_t.float64)
+                chk._spmv(x, tchk)
+            rl2 = float(_t.sum((b[:S.nowned] - tchk) ** 2))
+            bl2 = float(_t.sum(b[:S.nowned] ** 2))
+            if comm:
+                import torch.distributed as dist
+
+                rb = _t.tensor([rl2, bl2], dtype=_t.float64,
+                               device=tchk.device if commkind == "rccl" else "cpu")
+                dist.all_reduce(rb, op=dist.ReduceOp.SUM)
+                rl2, bl2 = float(rb[0]), float(rb[1])
+            if rank == 0 and bl2 > 0:
+                print(f"true residual: ||b-Ax||/||b|| = {(rl2 / bl2) ** 0.5:.6e}",
+                      file=sys.stderr)
+        except Exception as e:  # side-check only: never fail the solve over it
+            if rank == 0:
+                print(f"true residual: unavailable ({type(e).__name__}: {e})",
+                      file=sys.stderr)
 
     x_host = x[:S.nowned].cpu().numpy()
 
@@ -300,7 +379,12 @@ def main(argv=None) -> int:
             write_mtx(sys.stdout, vector_to_mtx(xg), numfmt=numfmt)
     if comm:
         comm.finalize()
-    return 0 if (res is None or res.converged or args.residual_rtol == 0) else 2
+    # exit 2 on non-convergence whenever ANY stopping criterion was active
+    # (residual atol/rtol or diff atol/rtol); 0 when none was requested
+    # (fixed-iteration run)
+    criterion = (args.residual_rtol > 0 or args.residual_atol > 0
+                 or args.diff_atol > 0 or args.diff_rtol > 0)
+    return 0 if (res is None or res.converged or not criterion) else 2
 
 
 if __name__ == "__main__":

@@ -85,6 +85,12 @@ class SymCSRMatrix:
         v = np.asarray(vals, dtype=np.float64)
         if i.shape != j.shape or i.shape != v.shape:
             raise AcgError(ErrCode.INVALID_VALUE, "COO array length mismatch")
+        # validate BOTH indices up front: negative numpy indices wrap
+        # silently and the native scatter would write past its buffers
+        if len(i) and (int(i.min()) < 0 or int(j.min()) < 0
+                       or int(i.max()) >= n or int(j.max()) >= n):
+            raise AcgError(ErrCode.INVALID_VALUE,
+                           f"COO index outside [0, {n})")
         try:
             from ..host import _acg_host as H
 

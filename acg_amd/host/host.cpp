@@ -88,6 +88,23 @@ py::tuple sym_expand_full(i64 n,
     const double* vu = val_u.data();
     const i64 nnz_u = rp[n];
 
+    // validate column range before the counting scatter (flag, not throw:
+    // throwing inside an OpenMP region is std::terminate)
+    std::atomic<i64> badidx(-1);
+    #pragma omp parallel for schedule(static)
+    for (i64 k = 0; k < nnz_u; ++k) {
+        if (cu[k] < 0 || cu[k] >= n) {
+            i64 expect = -1;
+            badidx.compare_exchange_strong(expect, k, std::memory_order_relaxed);
+        }
+    }
+    if (badidx.load() >= 0) {
+        i64 k = badidx.load();
+        throw std::out_of_range(
+            "sym_expand_full: column " + std::to_string(cu[k]) +
+            " at entry " + std::to_string(k) + " outside [0," +
+            std::to_string(n) + ")");
+    }
     std::vector<std::atomic<i64>> cnt(n);
     for (i64 i = 0; i < n; ++i) cnt[i].store(0, std::memory_order_relaxed);
     #pragma omp parallel for schedule(static)
@@ -180,12 +197,29 @@ py::tuple coo_to_sym_csr(i64 n,
     const i64* ci = cols.data();
     const double* vi = vals.data();
 
+    // Bounds-validate BOTH indices before any scatter touches memory, and
+    // record violations in a flag instead of throwing: a throw inside an
+    // OpenMP region is std::terminate, not a Python exception.
+    std::atomic<i64> badidx(-1);
+    #pragma omp parallel for schedule(static)
+    for (i64 k = 0; k < nnz; ++k) {
+        if (ri[k] < 0 || ri[k] >= n || ci[k] < 0 || ci[k] >= n) {
+            i64 expect = -1;
+            badidx.compare_exchange_strong(expect, k, std::memory_order_relaxed);
+        }
+    }
+    if (badidx.load() >= 0) {
+        i64 k = badidx.load();
+        throw std::out_of_range(
+            "coo_to_sym_csr: entry " + std::to_string(k) + " (" +
+            std::to_string(ri[k]) + "," + std::to_string(ci[k]) +
+            ") outside [0," + std::to_string(n) + ")");
+    }
     std::vector<std::atomic<i64>> cnt(n);
     for (i64 i = 0; i < n; ++i) cnt[i].store(0, std::memory_order_relaxed);
     #pragma omp parallel for schedule(static)
     for (i64 k = 0; k < nnz; ++k) {
         i64 r = ri[k] <= ci[k] ? ri[k] : ci[k];
-        if (r < 0 || r >= n) throw std::runtime_error("row index out of range");
         cnt[r].fetch_add(1, std::memory_order_relaxed);
     }
     std::vector<i64> rp(n + 1);

@@ -182,9 +182,10 @@ def bsell_from_csr(rowptr, colidx, vals, dof: int, C: int = 64):
 
 def spmv_sell(sellptr, cols, vals, nrows, x, y, *, rowbase: int = 0,
               accum: bool = False, partials=None, scal=None,
-              dotslot: int = -1, dot_accum: bool = True) -> None:
+              dotslot: int = -1, dot_accum: bool = True, C: int = 64) -> None:
     nslices = sellptr.numel() - 1
-    C = 64
+    assert (nrows + C - 1) // C == nslices, \
+        f"SELL structure was built with a different chunk size than C={C}"
     contrib = torch.zeros(nrows, dtype=torch.float64, device=x.device)
     for s in range(nslices):
         base = int(sellptr[s])

@@ -274,12 +274,19 @@ class CGSolverHIP:
     def solve(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
               res_atol: float = 0.0, res_rtol: float = 1e-9,
               use_graph: bool = False,
-              fold_daypx: bool | None = None) -> SolveResult:
+              fold_daypx: bool | None = None,
+              diff_atol: float = 0.0, diff_rtol: float = 0.0) -> SolveResult:
         """Classic CG (reference acgsolverhip_solvempi, cghip.c:402-1159).
 
         ``x`` must be an nlocal vector (ghost tail included); ``b`` nowned.
         The host convergence test runs every iteration on the lag-1
         pipeline (see solve_pipelined).
+
+        ``diff_atol``/``diff_rtol`` stop on the solution change
+        ``|alpha|*||p|| <= max(diff_atol, diff_rtol*||b||)`` -- a
+        beyond-reference capability (the reference GPU solvers REJECT diff
+        tolerances, cghip.c:427); engaging it switches to a blocking
+        per-iteration host test (no lag pipeline, no graph replay).
         """
         res = SolveResult(solver="cg-hip", maxits=maxits, res_atol=res_atol,
                           res_rtol=res_rtol,
@@ -307,13 +314,18 @@ class CGSolverHIP:
         res.bnrm2 = math.sqrt(max(bnrm2sqr, 0.0))
         res.r0nrm2 = math.sqrt(max(rr, 0.0))
         rtol2 = max(res_atol, res_rtol * res.bnrm2) ** 2
-        if rtol2 > 0 and rr <= rtol2:
+        dtol = max(diff_atol, diff_rtol * res.bnrm2)
+        if rtol2 > 0 and rr <= rtol2 and dtol == 0.0:
             res.converged = True
             res.rnrm2 = math.sqrt(rr)
             res.tsolve = time.perf_counter() - t0
             return res
         converged = False
         serial = self.comm is None or self.comm.size == 1
+        if dtol > 0:
+            converged = self._solve_diffmode(res, r, t, p, xi, rtol2, dtol,
+                                             maxits)
+            return self._finish_classic(res, x, xi, converged, rtol2, t0)
         # daypx folded into the BSELL SpMV (serial matA-only): the gather
         # computes beta*p_old + r on the fly and the row side materialises
         # p_new into a ping-pong buffer, eliminating the 3n-stream daypx
@@ -408,6 +420,45 @@ class CGSolverHIP:
             for j in range(max(maxits - LAG, 0), maxits):
                 if check(j):
                     break
+        return self._finish_classic(res, x, xi, converged, rtol2, t0)
+
+    def _solve_diffmode(self, res, r, t, p, xi, rtol2: float, dtol: float,
+                        maxits: int) -> bool:
+        """Classic iterations with the solution-change stopping criterion
+        ``|alpha|*||p|| <= dtol`` (CPU-solver semantics, cg.c:1059-1068).
+        Blocking host test each iteration: alpha = rr_prev/pt comes from
+        the device scalar slab AFTER the fused update rotated rr ->
+        rr_prev; ||p||^2 is one extra fused-dot + allreduce per iteration
+        (S_GAMMA used as scratch -- classic never touches it)."""
+        S = ops
+        n = self.n
+        scal = self.scal
+        converged = False
+        for k in range(maxits):
+            self._spmv_overlapped(p, t, fuse_dotslot=S.S_PT)
+            self._allreduce_slot(S.S_PT)
+            S.dot(p, p, self.partials, scal, S.S_GAMMA, n=n)
+            self._allreduce_slot(S.S_GAMMA)
+            S.cg_fused_update(r, xi, p, t, scal, self.partials, n)
+            self._allreduce_slot(S.S_RR)
+            res.niterations = k + 1
+            rr_old = self._host_scalar(S.S_RR_PREV)
+            pt = self._host_scalar(S.S_PT)
+            pp = self._host_scalar(S.S_GAMMA)
+            rr = self._host_scalar(S.S_RR)
+            if not math.isfinite(rr):
+                raise FloatingPointError(f"CG diverged: rr={rr} at it {k + 1}")
+            alpha = rr_old / pt if pt != 0.0 else 0.0
+            dx = abs(alpha) * math.sqrt(max(pp, 0.0))
+            if (rtol2 > 0 and rr <= rtol2) or dx <= dtol:
+                converged = True
+                break
+            S.daypx_ratio(p, r, scal, S.S_RR, S.S_RR_PREV, n=n)
+        return converged
+
+    def _finish_classic(self, res, x, xi, converged: bool, rtol2: float,
+                        t0: float):
+        S = ops
         torch.cuda.synchronize(self.device)
         res.tsolve = time.perf_counter() - t0
         x.copy_(xi)
@@ -415,7 +466,7 @@ class CGSolverHIP:
         res.rnrm2 = math.sqrt(max(rr, 0.0))
         res.converged = converged or (rtol2 > 0 and rr <= rtol2)
         nnz_full = self.local.nnzA + self.local.nnzO
-        res.nflops = res.niterations * cg_flops_per_iter(nnz_full, n)
+        res.nflops = res.niterations * cg_flops_per_iter(nnz_full, self.n)
         res.halo_bytes_sent = self.halo.bytes_sent
         res.halo_msgs_sent = self.halo.nmsgs_sent
         self.niterations_total += res.niterations

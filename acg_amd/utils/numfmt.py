@@ -41,11 +41,38 @@ class FmtSpec:
         # Python has no %a hex-float conversion via %-formatting; map to float.hex.
         self._hex = m.group("conversion") in "aA"
         self._py = py
+        # honour the ' (thousands-grouping) flag like the reference fmtspec:
+        # %-formatting cannot group, so build an equivalent format() spec
+        # ({:[flags][width],[.prec][conv]}) and use it when ' is present
+        self._groupfmt = None
+        if "'" in m.group("flags") and not self._hex:
+            flags = m.group("flags")
+            conv = m.group("conversion")
+            f = ""
+            if "-" in flags:
+                f += "<"
+            if "+" in flags:
+                f += "+"
+            elif " " in flags:
+                f += " "
+            if "#" in flags:
+                f += "#"
+            if "0" in flags and "-" not in flags:
+                f += "0"
+            if m.group("width"):
+                f += m.group("width")
+            f += ","
+            if m.group("precision"):
+                f += m.group("precision")
+            f += conv
+            self._groupfmt = f
 
     def format(self, value: float) -> str:
         if self._hex:
             s = float(value).hex()
             return s.upper() if self.spec[-1] == "A" else s
+        if self._groupfmt is not None:
+            return format(float(value), self._groupfmt)
         return self._py % value
 
     def __call__(self, value: float) -> str:

@@ -1,0 +1,167 @@
+"""Round-2 robustness fixes: COO bounds validation,
+numfmt thousands grouping, diff-tolerance handling, CLI exit codes,
+coordinate-format RHS."""
+
+import numpy as np
+import pytest
+import torch
+
+from acg_amd.core.symcsr import SymCSRMatrix
+from acg_amd.gen import STENCIL_5PT_2D, stencil_global
+from acg_amd.io.mtx import MtxFile, write_mtx
+from acg_amd.utils.errors import AcgError
+from acg_amd.utils.numfmt import parse_numfmt
+
+
+def _poisson_mtx(tmp_path, nx=16, ny=16, name="A.mtx"):
+    A = stencil_global(nx, ny, 1, STENCIL_5PT_2D)
+    rows = np.repeat(np.arange(A.n), np.diff(A.rowptr))
+    m = MtxFile(object="matrix", format="coordinate", field_="real",
+                symmetry="symmetric", nrows=A.n, ncols=A.n,
+                nnz=A.nnz_stored, rowidx=rows, colidx=A.colidx, a=A.vals)
+    path = tmp_path / name
+    write_mtx(path, m)
+    return A, path
+
+
+# ---- COO index validation (ADVICE: heap overflow via malformed mtx) ----
+
+def test_from_coo_col_out_of_range_raises():
+    with pytest.raises(AcgError):
+        SymCSRMatrix.from_coo(4, [0, 1], [1, 10], [1.0, 2.0])
+
+
+def test_from_coo_row_out_of_range_raises():
+    with pytest.raises(AcgError):
+        SymCSRMatrix.from_coo(4, [0, 7], [1, 2], [1.0, 2.0])
+
+
+def test_from_coo_negative_index_raises():
+    with pytest.raises(AcgError):
+        SymCSRMatrix.from_coo(4, [0, -1], [1, 2], [1.0, 2.0])
+
+
+def test_native_expand_col_out_of_range_raises():
+    # hand-built packed matrix with a broken column: native sym_expand_full
+    # must raise a Python exception, not scribble past its buffers
+    A = SymCSRMatrix(3, np.array([0, 1, 2, 3]), np.array([0, 1, 2]),
+                     np.array([4.0, 4.0, 4.0]))
+    A.colidx[1] = 9  # corrupt
+    with pytest.raises(Exception):
+        A.to_full_csr()
+
+
+# ---- numfmt ' thousands flag (reference fmtspec honours it) ----
+
+def test_numfmt_thousands_flag():
+    f = parse_numfmt("%'.2f")
+    assert f(1234567.891) == "1,234,567.89"
+    f2 = parse_numfmt("%'14.1f")
+    assert f2(1234567.891) == "   1,234,567.9"
+    # grouping has no visible effect on scientific notation (like C)
+    f3 = parse_numfmt("%'.3e")
+    assert f3(1234.5) == "1.234e+03"
+
+
+# ---- diff tolerances ----
+
+def test_cpu_diff_stopping_matches_gpu_semantics():
+    # plumbing-level check of the shared semantics: CPU solve stops on
+    # |alpha|*||p|| <= dtol even with residual tolerances off
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.solvers.cpu import CGSolverCPU
+
+    A = stencil_global(16, 16, 1, STENCIL_5PT_2D)
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    solver = CGSolverCPU(S)
+    b = torch.ones(S.nowned, dtype=torch.float64)
+    x = torch.zeros(S.nowned, dtype=torch.float64)
+    res = solver.solve(b, x, maxits=2000, res_rtol=0.0, diff_atol=1e-10)
+    assert res.converged
+    assert res.niterations < 2000
+
+
+def test_cli_diff_rejected_for_pipelined_gpu_solver(tmp_path, monkeypatch):
+    # without a GPU the CLI errors before the diff check; only meaningful
+    # to assert the NOT_SUPPORTED path when solver construction succeeds.
+    # Here: the CPU solver accepts diff tolerances (rc 0 on convergence).
+    from acg_amd import cli
+
+    _A, path = _poisson_mtx(tmp_path)
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    monkeypatch.delenv("RANK", raising=False)
+    rc = cli.main([str(path), "--solver", "cpu", "--diff-atol", "1e-10",
+                   "--residual-rtol", "0", "--max-iterations", "2000", "-q"])
+    assert rc == 0
+
+
+# ---- CLI exit codes ----
+
+def test_cli_exit2_on_atol_only_nonconvergence(tmp_path, monkeypatch):
+    from acg_amd import cli
+
+    _A, path = _poisson_mtx(tmp_path)
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    monkeypatch.delenv("RANK", raising=False)
+    # residual-rtol 0 but atol active and unreachable in 2 iterations
+    rc = cli.main([str(path), "--solver", "cpu", "--residual-rtol", "0",
+                   "--residual-atol", "1e-14", "--max-iterations", "2", "-q"])
+    assert rc == 2
+
+
+def test_cli_exit0_no_criterion(tmp_path, monkeypatch):
+    from acg_amd import cli
+
+    _A, path = _poisson_mtx(tmp_path)
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    monkeypatch.delenv("RANK", raising=False)
+    rc = cli.main([str(path), "--solver", "cpu", "--residual-rtol", "0",
+                   "--max-iterations", "2", "-q"])
+    assert rc == 0
+
+
+# ---- RHS format handling ----
+
+def test_cli_coordinate_rhs_scattered_by_rowidx(tmp_path, capsys, monkeypatch):
+    from acg_amd import cli
+
+    A, path = _poisson_mtx(tmp_path)
+    n = A.n
+    # sparse RHS with nnz == n but PERMUTED rowidx: positional assignment
+    # would silently permute b; rowidx scatter must not
+    rng = np.random.default_rng(3)
+    perm = rng.permutation(n)
+    vals = rng.standard_normal(n)
+    mb = MtxFile(object="matrix", format="coordinate", field_="real",
+                 symmetry="general", nrows=n, ncols=1, nnz=n,
+                 rowidx=perm.astype(np.int64),
+                 colidx=np.zeros(n, dtype=np.int64), a=vals)
+    bpath = tmp_path / "b.mtx"
+    write_mtx(bpath, mb)
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    monkeypatch.delenv("RANK", raising=False)
+    rc = cli.main([str(path), str(bpath), "--solver", "cpu",
+                   "--max-iterations", "3000", "--residual-rtol", "1e-12"])
+    out = capsys.readouterr()
+    assert rc == 0, out.err
+    # solve A x = b with the correctly-scattered b and compare
+    b = np.zeros(n)
+    b[perm] = vals
+    xs = np.linalg.solve(A.to_scipy_full().toarray(), b)
+    got = np.array([float(v) for v in out.out.strip().splitlines()[2:]])
+    assert np.allclose(got, xs, atol=1e-8)
+
+
+def test_cli_rhs_length_mismatch_raises(tmp_path, monkeypatch):
+    from acg_amd import cli
+
+    _A, path = _poisson_mtx(tmp_path)
+    mb = MtxFile(object="matrix", format="array", field_="real",
+                 symmetry="general", nrows=7, ncols=1, nnz=7,
+                 a=np.ones(7))
+    bpath = tmp_path / "b.mtx"
+    write_mtx(bpath, mb)
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    monkeypatch.delenv("RANK", raising=False)
+    with pytest.raises(AcgError):
+        cli.main([str(path), str(bpath), "--solver", "cpu", "-q"])
