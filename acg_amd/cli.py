@@ -217,6 +217,14 @@ def main(argv=None) -> int:
     log(f"scattered subdomains ({time.perf_counter() - t0:.2f}s)")
     if args.verbose and args.verbose > 1:
         S.dump(file=sys.stderr)
+    if comm and comm.size > 1:
+        # collective dry-run audit of the halo pattern: fail loudly at
+        # setup (pairing symmetry, ghost-tail global-id agreement) instead
+        # of deadlocking in the first exchange
+        from .dist.verify import verify_halo
+
+        verify_halo(S, comm)
+        log("halo audit passed")
 
     if args.output_comm_matrix and comm:
         counts = comm.gather_object(

@@ -273,7 +273,7 @@ class CGSolverHIP:
 
     def solve(self, b: torch.Tensor, x: torch.Tensor, maxits: int = 100,
               res_atol: float = 0.0, res_rtol: float = 1e-9,
-              use_graph: bool = False,
+              use_graph: bool | None = None,
               fold_daypx: bool | None = None,
               diff_atol: float = 0.0, diff_rtol: float = 0.0) -> SolveResult:
         """Classic CG (reference acgsolverhip_solvempi, cghip.c:402-1159).
@@ -281,6 +281,18 @@ class CGSolverHIP:
         ``x`` must be an nlocal vector (ghost tail included); ``b`` nowned.
         The host convergence test runs every iteration on the lag-1
         pipeline (see solve_pipelined).
+
+        ``use_graph`` default None = measured policy (same as pipelined):
+        OFF serial (eager wins by ~8 us/it), ON multi-GPU -- the WHOLE
+        distributed iteration (halo fork, split SpMV, both allreduces,
+        fused update, daypx) is captured once and replayed so the
+        ~100 us/it of host-side c10d/launch work leaves the critical
+        path.  Unlike pipelined no in-graph scalar snapshot is needed:
+        the body ENDS with scal[S_RR] holding the allreduced new (r,r)
+        (daypx reads but never writes it), so the post-replay lagged D2H
+        reads a rank-consistent value by stream order.  Any capture
+        failure (e.g. gloo) permanently falls back to eager;
+        ACG_DIST_GRAPH=0 is the kill switch.
 
         ``diff_atol``/``diff_rtol`` stop on the solution change
         ``|alpha|*||p|| <= max(diff_atol, diff_rtol*||b||)`` -- a
@@ -338,8 +350,18 @@ class CGSolverHIP:
         if fold:
             p2 = self._workspace("classic", [("p2", True)])["p2"]
             scal[S.S_RR_PREV] = math.inf
+        if use_graph is None:
+            use_graph = not serial  # measured policy (see docstring)
         graph_ok = use_graph and serial and not self.prof.enabled and not fold
         graph = self._graphs.get("classic") if graph_ok else None
+        # multi-GPU: capture the whole distributed iteration (see docstring)
+        dist_key = "classic:dist"
+        dist_failed_key = "classic:capture_failed"
+        dist_graph_ok = (use_graph and not serial and not self.prof.enabled
+                         and not fold
+                         and os.environ.get("ACG_DIST_GRAPH", "1") != "0"
+                         and not self._graphs.get(dist_failed_key, False))
+        dgraph = self._graphs.get(dist_key) if dist_graph_ok else None
         # lag-2 convergence pipeline (+ optional hipGraph replay), mirroring
         # solve_pipelined (the host test runs for every iteration; the host
         # reads the value two iterations late).  Classic's rr copy lands at
@@ -399,6 +421,8 @@ class CGSolverHIP:
                 break
             if graph is not None:
                 graph.replay()
+            elif dgraph is not None and k > 0:
+                dgraph.replay()
             else:
                 body(k)
                 if graph_ok and k == 1:
@@ -406,6 +430,23 @@ class CGSolverHIP:
                     with torch.cuda.graph(graph):
                         body()
                     self._graphs["classic"] = graph
+                elif dist_graph_ok and k == 2 and dgraph is None:
+                    # RCCL channels are warm after 2 eager iterations;
+                    # capture records without executing (rank-local)
+                    try:
+                        gg = torch.cuda.CUDAGraph()
+                        with torch.cuda.graph(gg):
+                            body()
+                        dgraph = gg
+                        self._graphs[dist_key] = gg
+                    except Exception as e:
+                        self._graphs[dist_failed_key] = True
+                        torch.cuda.synchronize(self.device)
+                        if self.comm is None or self.comm.rank == 0:
+                            print(f"[acg_amd] classic distributed graph "
+                                  f"capture unavailable ({type(e).__name__});"
+                                  f" eager iterations",
+                                  file=__import__("sys").stderr)
             # lagged rr D2H on the SAME stream: cross-stream event chains
             # measured ~20 us each in fire->launch latency (45 us/iter of
             # the iteration period); the in-order 8-byte copy costs ~3 us

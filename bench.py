@@ -104,34 +104,64 @@ def main() -> int:
     solver = CGSolverHIP(S, comm=comm, device=device, lanes=args.lanes,
                          matfree=args.matfree)
 
+    # dry-run halo audit: collective cross-rank check of pairing symmetry
+    # and in-place ghost-tail global-id agreement, so the first N>=2 run
+    # fails loudly at setup instead of deadlocking mid-solve
+    if comm is not None and comm.size > 1:
+        from acg_amd.dist.verify import verify_halo
+
+        verify_halo(S, comm)
+
     rloc = np.random.default_rng(10_000 + rank)
     b = torch.from_numpy(rloc.standard_normal(S.nowned)).to(device)
     x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device=device)
 
+    def _probe_pair():
+        """Time classic vs pipelined briefly and agree on the winner
+        collectively (MAX over ranks per arm, identical collective
+        sequence on every rank)."""
+        def _probe(fn):
+            fn(b, x.clone(), maxits=5, res_rtol=0.0)
+            torch.cuda.synchronize(device)
+            if comm is not None:
+                comm.barrier()
+            t0 = time.perf_counter()
+            fn(b, x.clone(), maxits=30, res_rtol=0.0)
+            torch.cuda.synchronize(device)
+            el = time.perf_counter() - t0
+            if comm is not None:
+                import torch.distributed as dist
+
+                et = torch.tensor([el], dtype=torch.float64, device=red_dev)
+                dist.all_reduce(et, op=dist.ReduceOp.MAX)
+                el = float(et.item())
+            return el
+        tc = _probe(solver.solve)
+        tp = _probe(solver.solve_pipelined)
+        kind_ = "classic" if tc <= tp else "pipelined"
+        if rank == 0:
+            print(f"# auto probe: classic {tc * 1e6 / 30:.1f} us/it, "
+                  f"pipelined {tp * 1e6 / 30:.1f} -> {kind_}",
+                  file=sys.stderr, flush=True)
+        return kind_
+
+    red_dev = device if kind == "rccl" else "cpu"
     if args.solver == "auto":
-        if ngpus > 1 or solver.megafuse_auto:
-            # multi-GPU: single overlapped (graph-captured) allreduce;
-            # narrow rows: megafused pipelined (both measured policy)
+        if solver.megafuse_auto:
+            # narrow rows: megafused pipelined (measured policy)
             solver_kind = "pipelined"
+        elif ngpus > 1:
+            # multi-GPU: both solvers are graph-captured; their ordering
+            # depends on halo/allreduce overlap on the actual fabric --
+            # probe both (collective decision, MAX over ranks)
+            solver_kind = _probe_pair()
         else:
             # serial wide rows: classic and pipelined are within ~10% of
             # each other and the WINNER flips with the per-instance
             # allocation-placement lottery (profiles/RESULTS.md) -- probe
             # both briefly and keep the faster (selection + its warmup
             # happen outside the timed region, like any autotuner)
-            def _probe(fn):
-                fn(b, x.clone(), maxits=5, res_rtol=0.0)
-                torch.cuda.synchronize(device)
-                t0 = time.perf_counter()
-                fn(b, x.clone(), maxits=30, res_rtol=0.0)
-                torch.cuda.synchronize(device)
-                return time.perf_counter() - t0
-            tc = _probe(solver.solve)
-            tp = _probe(solver.solve_pipelined)
-            solver_kind = "classic" if tc <= tp else "pipelined"
-            print(f"# auto probe: classic {tc * 1e6 / 30:.1f} us/it, "
-                  f"pipelined {tp * 1e6 / 30:.1f} -> {solver_kind}",
-                  file=sys.stderr, flush=True)
+            solver_kind = _probe_pair()
     else:
         solver_kind = args.solver
     solve = solver.solve_pipelined if solver_kind == "pipelined" else solver.solve
@@ -182,6 +212,24 @@ def main() -> int:
     recur_rel = res.rnrm2 / res.bnrm2 if res.bnrm2 > 0 else float("nan")
     assert np.isfinite(true_rel), "non-finite true residual"
 
+    # per-rank overlap diagnosis (outside the timed region): a short
+    # profiled EAGER pass records hipEvent spans (halo / spmvA / spmvO /
+    # allreduce / update) per rank and ships them on the JSON line, so the
+    # first real N=8 run is diagnosable -- not just a single number.
+    # (Profiling forces the eager path; the timed number above still
+    # reflects the captured-graph hot loop.)
+    rank_spans = None
+    if ngpus > 1 or os.environ.get("ACG_BENCH_SPANS"):
+        solver.prof.enabled = True
+        pres = solve(b, x.clone(), maxits=16, res_rtol=0.0)
+        solver.prof.enabled = False
+        mine = {nm: [round(st.seconds * 1e6 / max(st.count, 1), 1), st.count]
+                for nm, st in (pres.ops or {}).items()}
+        mine["halo_B_per_it"] = 8 * (S.halo.sendsize + S.halo.recvsize)
+        allspans = comm.gather_object(mine) if comm is not None else [mine]
+        if rank == 0:
+            rank_spans = {f"r{i}": d for i, d in enumerate(allspans)}
+
     nrows_global = dof * G * G * G
     nnz_local = S.nnzA + S.nnzO
     if comm is not None:
@@ -226,6 +274,7 @@ def main() -> int:
                 "rel_residual_recursion": recur_rel,
                 "gflops": args.steps * (2.0 * nnz_global + 10.0 * nrows_global) / elapsed / 1e9,
                 "parallelism": f"slab{ngpus}-rccl",
+                "rank_spans_us_per_call": rank_spans,
             },
         }
         print(json.dumps(out), flush=True)

@@ -158,6 +158,33 @@ def test_halo_exchange_gloo_ws2():
     _run_dist("_body_halo", world=2, port=29601)
 
 
+def _body_halo_audit(comm):
+    """Collective halo dry-run audit (dist.verify) on a 4-way rgb
+    partition: passes clean, and a corrupted pattern raises on ALL ranks."""
+    from acg_amd.dist.verify import verify_halo
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.utils.errors import AcgError
+
+    A = stencil_global(6, 6, 6, STENCIL_27PT_3D)
+    part = partition_rows(A, comm.size, method="rgb", seed=1)
+    S = extract_subdomains(A, part, comm.size)[comm.rank]
+    verify_halo(S, comm)  # must pass
+    # corrupt ONE rank's ghost ordering: every rank must raise
+    if comm.rank == 1 and S.nghost >= 2:
+        S.ghost_global[:2] = S.ghost_global[:2][::-1].copy()
+    try:
+        verify_halo(S, comm)
+        return ("no-raise",)
+    except AcgError:
+        return ("raised",)
+
+
+def test_halo_audit_ws4():
+    results = _run_dist("_body_halo_audit", world=4, port=29606)
+    for rank, (status,) in results.items():
+        assert status == "raised", f"rank {rank}: corrupted halo not detected"
+
+
 @pytest.mark.parametrize("body,port", [("_body_cg", 29602),
                                        ("_body_cg_pipelined", 29603)])
 def test_distributed_cg_matches_serial(body, port):

@@ -165,3 +165,30 @@ def test_cli_rhs_length_mismatch_raises(tmp_path, monkeypatch):
     monkeypatch.delenv("RANK", raising=False)
     with pytest.raises(AcgError):
         cli.main([str(path), str(bpath), "--solver", "cpu", "-q"])
+
+
+# ---- halo audit (serial, all parts at once) ----
+
+def test_halo_audit_serial_all_parts():
+    from acg_amd.dist.verify import _audit, halo_descriptor
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.gen import STENCIL_27PT_3D
+
+    A = stencil_global(6, 6, 6, STENCIL_27PT_3D)
+    part = partition_rows(A, 4, method="rgb", seed=1)
+    systems = extract_subdomains(A, part, 4)
+    descs = [halo_descriptor(S) for S in systems]
+    _audit(descs)  # clean pattern passes
+    # asymmetric pairing: drop one recv entry
+    bad = [halo_descriptor(S) for S in systems]
+    victim = next(d for d in bad if d["recv"])
+    victim["recv"].pop(next(iter(victim["recv"])))
+    with pytest.raises(AcgError):
+        _audit(bad)
+    # count mismatch
+    bad2 = [halo_descriptor(S) for S in systems]
+    victim = next(d for d in bad2 if d["send"])
+    q = next(iter(victim["send"]))
+    victim["send"][q] = victim["send"][q][:-1]
+    with pytest.raises(AcgError):
+        _audit(bad2)
