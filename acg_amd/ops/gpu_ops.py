@@ -82,6 +82,59 @@ def spmv(rowptr: torch.Tensor, colidx: torch.Tensor, vals: torch.Tensor,
            scal.data_ptr() if fuse else 0, dotslot, dot_accum, _stream())
 
 
+def build_row_bins(rowptr, max_lanes: int = 64):
+    """Sort rows into lane-count bins by length for the hybrid binned SpMV.
+
+    Returns (rowlist int32 numpy, bins [(start, count, lanes)]).  Bin
+    thresholds follow the pick_lanes rule (~6 nnz per lane); rows inside
+    a bin are ordered longest-first so grid-stride waves retire the
+    heavy tail early."""
+    import numpy as np
+
+    rowptr = np.asarray(rowptr)
+    lens = np.diff(rowptr)
+    order = np.argsort(-lens, kind="stable").astype(np.int32)
+    slens = lens[order.astype(np.int64)]
+    bins = []
+    lo = 0
+    # descending order: long rows first; thresholds from the high end
+    for lanes in (64, 32, 16, 8, 4):
+        if lanes > max_lanes:
+            continue
+        if lanes == 4:
+            hi = len(slens)
+        else:
+            # rows longer than (lanes/2)*6 stay in this bin
+            thresh = (lanes // 2) * 6
+            hi = int(np.searchsorted(-slens, -thresh - 1, side="right"))
+        if hi > lo:
+            bins.append((lo, hi - lo, lanes))
+            lo = hi
+    return order, bins
+
+
+def spmv_binned(rowptr: torch.Tensor, colidx: torch.Tensor, vals: torch.Tensor,
+                rowlist: torch.Tensor, bins, x: torch.Tensor, y: torch.Tensor,
+                *, rowbase: int = 0, accum: bool = False,
+                partials: torch.Tensor | None = None,
+                scal: torch.Tensor | None = None, dotslot: int = -1,
+                dot_accum: bool = True) -> None:
+    """Row-binned hybrid CSR SpMV (load balance for power-law rows).
+    ``rowlist`` int32 rows sorted longest-first; ``bins`` from
+    :func:`build_row_bins`."""
+    if rowlist.numel() == 0:
+        return
+    assert rowlist.dtype == torch.int32
+    fuse = scal is not None and dotslot >= 0
+    K.spmv_binned(rowbase, rowptr.data_ptr(), colidx.data_ptr(),
+                  1 if colidx.dtype == torch.int64 else 0,
+                  vals.data_ptr(), x.data_ptr(), y.data_ptr(),
+                  rowlist.data_ptr(), [list(b) for b in bins], accum,
+                  partials.data_ptr() if fuse else 0,
+                  scal.data_ptr() if fuse else 0, dotslot, dot_accum,
+                  _stream())
+
+
 # SELL kernel variant bits (see kernels.hip): +1 non-temporal vals/cols,
 # +2 XCD-aware block swizzle, +4 unroll-8.
 SELL_NT, SELL_SWZ, SELL_U8 = 1, 2, 4

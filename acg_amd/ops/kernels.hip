@@ -45,7 +45,10 @@
 
 #include <hip/hip_runtime.h>
 #include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+#include <array>
 #include <cstdint>
+#include <vector>
 #include <stdexcept>
 #include <string>
 
@@ -111,7 +114,12 @@ k_reduce_partials(const double* __restrict__ partials, int nblocks,
 // ---------------------------------------------------------------------------
 // CSR SpMV, vector kernel: LANES lanes cooperate on one row (irregular rows,
 // matO).  FUSE_DOT: also reduce dot(x, y_contrib) into partials[blockIdx].
-template <typename ColT, int LANES, bool ACCUM, bool FUSE_DOT>
+// ROWLIST: indirect row ids (rowlist[r]) -- the row-binned hybrid launches
+// this kernel once per length bin with bin-appropriate LANES, the MI355X
+// answer to the load-balance problem the reference's merge-path SpMV
+// solves (cg-kernels-hip.hip:348-1175): per-row work stays proportional
+// to row length, no fp64 atomics, fully deterministic.
+template <typename ColT, int LANES, bool ACCUM, bool FUSE_DOT, bool ROWLIST = false>
 __global__ void __launch_bounds__(BLOCK)
 spmv_csr_vector(long nrows, long rowbase,
                 const long* __restrict__ rowptr,
@@ -119,12 +127,14 @@ spmv_csr_vector(long nrows, long rowbase,
                 const double* __restrict__ vals,
                 const double* __restrict__ x,
                 double* __restrict__ y,
-                double* __restrict__ partials) {
+                double* __restrict__ partials,
+                const int* __restrict__ rowlist = nullptr) {
     const int lane = threadIdx.x & (LANES - 1);
     const long group = ((long)blockIdx.x * BLOCK + threadIdx.x) / LANES;
     const long ngroups = (long)gridDim.x * BLOCK / LANES;
     double dacc = 0.0;
-    for (long r = group; r < nrows; r += ngroups) {
+    for (long i = group; i < nrows; i += ngroups) {
+        const long r = ROWLIST ? (long)rowlist[i] : i;
         const long k0 = rowptr[r], k1 = rowptr[r + 1];
         double sum = 0.0;
         for (long k = k0 + lane; k < k1; k += LANES)
@@ -1573,6 +1583,60 @@ void spmv(long nrows, long rowbase, uintptr_t rowptr, uintptr_t colidx,
         reduce_partials(partials, (int)blocks, scal, dotslot, dot_accum, stream);
 }
 
+// Row-binned hybrid CSR SpMV: rows pre-sorted by length into bins, one
+// launch per bin with bin-appropriate LANES (4..64).  Every row is fully
+// reduced by one lane group => deterministic, no atomics; each launch
+// writes its own partials window, one finalize over the union.  This is
+// the MI355X-native load-balancer for power-law rows (reference analog:
+// merge-path csrgemv_merge, cg-kernels-hip.hip:348-1175).
+// bins: (start_in_rowlist, count, lanes) triples.
+void spmv_binned(long rowbase, uintptr_t rowptr, uintptr_t colidx, int col64,
+                 uintptr_t vals, uintptr_t x, uintptr_t y, uintptr_t rowlist,
+                 const std::vector<std::array<long, 3>>& bins,
+                 bool accum, uintptr_t partials, uintptr_t scal,
+                 int dotslot, bool dot_accum, uintptr_t stream) {
+    const bool fuse = partials != 0 && dotslot >= 0;
+    long poff = 0;
+    for (const auto& bin : bins) {
+        const long start = bin[0], count = bin[1];
+        const int lanes = (int)bin[2];
+        if (count == 0) continue;
+        const int rows_per_block = BLOCK / lanes;
+        long blocks = (count + rows_per_block - 1) / rows_per_block;
+        if (blocks > 3072) blocks = 3072;  // grid-stride within the bin
+        // (5 bins x 3072 <= MAXG partials)
+        if (poff + blocks > MAXG)
+            throw std::runtime_error("spmv_binned: partials overflow");
+        dim3 g((unsigned)blocks), b(BLOCK);
+        const int* rl = (const int*)rowlist + start;
+        double* pp = (double*)partials + poff;
+        #define LAUNCH_BIN(CT, L, AC, FD) \
+            hipLaunchKernelGGL((spmv_csr_vector<CT, L, AC, FD, true>), g, b, 0, \
+                S(stream), count, rowbase, (const long*)rowptr, \
+                (const CT*)colidx, (const double*)vals, (const double*)x, \
+                (double*)y, pp, rl)
+        #define DISPATCH_BL(CT, AC, FD) \
+            switch (lanes) { \
+                case 4:  LAUNCH_BIN(CT, 4,  AC, FD); break; \
+                case 8:  LAUNCH_BIN(CT, 8,  AC, FD); break; \
+                case 16: LAUNCH_BIN(CT, 16, AC, FD); break; \
+                case 32: LAUNCH_BIN(CT, 32, AC, FD); break; \
+                case 64: LAUNCH_BIN(CT, 64, AC, FD); break; \
+                default: throw std::runtime_error("spmv_binned: lanes 4/8/16/32/64"); }
+        #define DISPATCH_BA(CT) \
+            if (accum) { if (fuse) { DISPATCH_BL(CT, true, true) } else { DISPATCH_BL(CT, true, false) } } \
+            else       { if (fuse) { DISPATCH_BL(CT, false, true) } else { DISPATCH_BL(CT, false, false) } }
+        if (col64) { DISPATCH_BA(long) } else { DISPATCH_BA(int) }
+        #undef DISPATCH_BA
+        #undef DISPATCH_BL
+        #undef LAUNCH_BIN
+        if (fuse) poff += blocks;
+    }
+    check_hip("spmv_binned");
+    if (fuse && poff > 0)
+        reduce_partials(partials, (int)poff, scal, dotslot, dot_accum, stream);
+}
+
 void spmv_sell(long nslices, long nrows, long rowbase, uintptr_t sellptr,
                uintptr_t cols, int col64, uintptr_t vals, uintptr_t x,
                uintptr_t y, bool accum, uintptr_t partials, uintptr_t scal,
@@ -1810,6 +1874,7 @@ void pack_gather(uintptr_t sendbuf, uintptr_t x, uintptr_t idx, int idx64, long 
 PYBIND11_MODULE(_acg_kernels, m) {
     m.doc() = "acg_amd gfx950 HIP kernels";
     m.def("spmv", &spmv);
+    m.def("spmv_binned", &spmv_binned);
     m.def("spmv_sell", &spmv_sell);
     m.def("zero_scalars", &zero_scalars);
     m.def("cg_prep_pt", &cg_prep_pt);

@@ -135,12 +135,20 @@ def _sort_rows_cols(rowptr, colidx, vals, nrows):
     return colidx[order], vals[order]
 
 
-def extract_subdomains(A, part: np.ndarray, nparts: int, eps: float = 0.0) -> list:
+def extract_subdomains(A, part: np.ndarray, nparts: int, eps: float = 0.0,
+                       only_parts=None) -> list:
     """Split a SymCSRMatrix into per-rank LocalSystems + halo patterns.
 
     Reference call stack analog: acgsymcsrmatrix_partition (symcsrmatrix.c:685)
     -> acggraph_partition (graph.c:813) -> acgsymcsrmatrix_dsymv_init
     (symcsrmatrix.c:760) -> acgsymcsrmatrix_halo/acggraph_halo (graph.c:1898).
+
+    The structure pass (interior/border split, per-part ghost lists) is ONE
+    vectorised sweep over the full sparsity (no per-part scipy slicing);
+    the value pass streams one part at a time.  ``only_parts`` limits the
+    value pass to the listed parts (other list slots are None) -- every
+    rank of a deterministic-input job can extract just its own subdomain
+    with O(nnz/nparts) extra memory beyond the shared structure arrays.
     """
     import scipy.sparse as sp
 
@@ -148,39 +156,45 @@ def extract_subdomains(A, part: np.ndarray, nparts: int, eps: float = 0.0) -> li
     part = np.asarray(part, dtype=np.int32)
     if len(part) != n:
         raise AcgError(ErrCode.INVALID_VALUE, "partition vector length mismatch")
+    if int(part.min()) < 0 or int(part.max()) >= nparts:
+        raise AcgError(ErrCode.INVALID_VALUE, "partition id outside [0, nparts)")
     f = A.to_full_csr(eps=eps)
     F = sp.csr_matrix((f.vals, f.colidx.astype(np.int64), f.rowptr), shape=(n, n))
 
-    # pass 1: per-part structure
-    owned_globals, ghost_globals, locmats = [], [], []
+    # pass 1 (structure, one global sweep): border rows + per-part ghost
+    # lists from the foreign entries (row part != col part)
+    rows_all = np.repeat(np.arange(n, dtype=np.int64), np.diff(f.rowptr))
+    colpart_all = part[f.colidx]
+    foreign_all = part[rows_all] != colpart_all
+    nforeign = np.bincount(rows_all[foreign_all], minlength=n)
+    is_border = nforeign > 0
+    fr = part[rows_all[foreign_all]].astype(np.int64)  # part of the row
+    fc = f.colidx[foreign_all].astype(np.int64)        # ghost global id
+    del rows_all, colpart_all, foreign_all
+    ukey = np.unique(fr * n + fc)
+    del fr, fc
+    gpart = (ukey // n).astype(np.int32)
+    ggid = (ukey % n).astype(np.int64)
+    del ukey
+    gorder = np.lexsort((ggid, part[ggid], gpart))  # (part | owner, gid)
+    gpart, ggid = gpart[gorder], ggid[gorder]
+    gsplit = np.searchsorted(gpart, np.arange(nparts + 1))
+    ghost_globals = [ggid[gsplit[p]:gsplit[p + 1]] for p in range(nparts)]
+    owned_globals, ninteriors = [], []
     for p in range(nparts):
         rows_p = np.where(part == p)[0].astype(np.int64)
-        Fp = F[rows_p]
-        colpart = part[Fp.indices]
-        foreign = colpart != p
-        nnz_per_row = np.diff(Fp.indptr)
-        rowid = np.repeat(np.arange(len(rows_p), dtype=np.int64), nnz_per_row)
-        nforeign = np.bincount(rowid[foreign], minlength=len(rows_p))
-        is_border = nforeign > 0
-        interior_g = rows_p[~is_border]
-        border_g = rows_p[is_border]
-        owned_global = np.concatenate([interior_g, border_g])
-        ghosts = np.unique(Fp.indices[foreign].astype(np.int64))
-        gowner = part[ghosts]
-        gorder = np.lexsort((ghosts, gowner))
-        ghost_global = ghosts[gorder]
-        owned_globals.append(owned_global)
-        ghost_globals.append(ghost_global)
-        locmats.append((rows_p, Fp, interior_g, border_g))
+        bmask = is_border[rows_p]
+        owned_globals.append(np.concatenate([rows_p[~bmask], rows_p[bmask]]))
+        ninteriors.append(int((~bmask).sum()))
 
-    systems = []
-    for p in range(nparts):
-        rows_p, Fp, interior_g, border_g = locmats[p]
+    want = list(range(nparts)) if only_parts is None else list(only_parts)
+    systems: list = [None] * nparts
+    for p in want:
         owned_global = owned_globals[p]
         ghost_global = ghost_globals[p]
         nowned = len(owned_global)
-        ninterior = len(interior_g)
-        nborder = len(border_g)
+        ninterior = ninteriors[p]
+        nborder = nowned - ninterior
         nghost = len(ghost_global)
         nlocal = nowned + nghost
 
@@ -247,11 +261,11 @@ def extract_subdomains(A, part: np.ndarray, nparts: int, eps: float = 0.0) -> li
             sdispls=sdispls,
             sendidx=sendidx.astype(cdt),
         )
-        systems.append(LocalSystem(
+        systems[p] = LocalSystem(
             rank=p, nparts=nparts, n_global=n,
             nowned=nowned, ninterior=ninterior, nborder=nborder, nghost=nghost,
             A_rowptr=A_rowptr, A_colidx=A_colidx, A_vals=A_vals,
             O_rowptr=O_rowptr, O_colidx=O_colidx, O_vals=O_vals,
             owned_global=owned_global, ghost_global=ghost_global, halo=halo,
-        ))
+        )
     return systems

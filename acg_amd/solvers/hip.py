@@ -41,7 +41,8 @@ class CGSolverHIP:
 
     def __init__(self, local: LocalSystem, comm=None, device=None,
                  lanes: int | None = None, use_sell: bool = True,
-                 profile: bool = False, matfree: bool = False):
+                 profile: bool = False, matfree: bool = False,
+                 force_format: str | None = None):
         self.local = local
         self.comm = comm
         if device is None:
@@ -63,6 +64,7 @@ class CGSolverHIP:
         self.O_rowptr = self.O_colidx = self.O_vals = None
         self.lanesA = self.lanesO = lanes or 16
         self.bsell = None
+        self.hybrid = None  # (rowlist int32, bins) row-binned CSR
         # matrix-free analytic operator (dof=1 stencils, opt-in): the SpMV
         # reads NO matrix data -- see ops/kernels.hip k_stencil_spmv.
         self.matfree = None
@@ -92,46 +94,7 @@ class CGSolverHIP:
             self.lanesA = lanes or ops.pick_lanes(mean_nnz)
             mean_nnzO = L.nnzO / max(L.nborder, 1)
             self.lanesO = lanes or ops.pick_lanes(mean_nnzO)
-            # SELL-C-64 fast path for matA when rows are regular enough that
-            # padding stays small (stencil/FEM); CSR-vector otherwise.
-            if use_sell and L.nowned > 0:
-                from ..ops.torch_ref import sell_from_csr
-
-                self.sell_perm = None
-                sellptr, scols, svals = sell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals)
-                waste = (int(sellptr[-1]) - L.nnzA) / max(L.nnzA, 1)
-                if waste > 0.3:
-                    # irregular rows: sigma-sort within 16-slice windows
-                    out = sell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals,
-                                        sigma=16)
-                    sellptr, scols, svals, perm = out
-                    waste = (int(sellptr[-1]) - L.nnzA) / max(L.nnzA, 1)
-                    if waste <= 0.5:
-                        self.sell_perm = up(perm)
-                if waste <= 0.5:
-                    self.sell = (up(sellptr), up(scols), up(svals))
-                    if L.nnzO > 0:
-                        optr, ocols, ovals = sell_from_csr(L.O_rowptr, L.O_colidx,
-                                                           L.O_vals)
-                        self.sellO = (up(optr), up(ocols), up(ovals))
-                # Block-SELL when the matrix has dense dof x dof blocks
-                # (FEM/structural, e.g. Queen_4147's 3-dof nodes): one int32
-                # index per block.  Worth it when block density is high
-                # enough that the index saving beats the zero-fill.
-                from ..ops.torch_ref import bsell_from_csr
-
-                for dof_try in (3, 2):
-                    out = bsell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals,
-                                         dof_try)
-                    if out is None:
-                        continue
-                    bptr_h, bcol_h, bvals_h, density = out
-                    if density >= 0.75:
-                        self.bsell = (up(bptr_h), up(bcol_h), up(bvals_h),
-                                      dof_try)
-                        break
-                if self.sell is not None or self.bsell is not None:
-                    self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR
+            self._pick_format(L, up, use_sell, force_format)
         # megafused pipelined iteration needs SELL everywhere + int32 cols.
         # Measured on MI355X: for wide rows (~80 nnz, Queen-shaped) the fused
         # epilogue's 6 vector streams cost the SpMV more x-gather locality
@@ -173,6 +136,84 @@ class CGSolverHIP:
         self._ws: dict = {}
         self._graphs: dict = {}
         self._nstag = 0
+
+    def _pick_format(self, L, up, use_sell: bool, force: str | None) -> None:
+        """Choose the matA operator format (reference analog: the choice
+        between hipsparse ALG_DEFAULT and the hand merge-path kernel,
+        cghip.c:533-585 / cg-kernels-hip.hip:348).
+
+        Auto ladder by measured fit:
+          1. plain SELL-C-64 when padding waste <= 0.3 (stencil/FEM rows),
+          2. sigma-SELL (window 16) when waste <= 0.5 (mildly irregular),
+          3. wide-window sigma-SELL (4096) when that gets waste <= 0.35,
+          4. row-binned hybrid CSR otherwise (power-law rows: per-bin lane
+             counts keep per-row work proportional to row length),
+        plus Block-SELL whenever dense dof x dof block structure is found
+        (density >= 0.75) -- it wins on index bytes.
+        ``force`` in {csr, sell, sigma, bsell, hybrid} overrides for A/B
+        measurement (bench --format)."""
+        from ..ops.torch_ref import bsell_from_csr, sell_from_csr
+
+        if L.nowned == 0:
+            return
+
+        def mk_hybrid():
+            rowlist, bins = ops.build_row_bins(L.A_rowptr)
+            self.hybrid = (up(rowlist), bins)
+
+        def mk_sell(sigma):
+            out = sell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals, sigma=sigma)
+            if sigma > 1:
+                sellptr, scols, svals, perm = out
+                self.sell_perm = up(perm)
+            else:
+                sellptr, scols, svals = out
+                self.sell_perm = None
+            self.sell = (up(sellptr), up(scols), up(svals))
+            if L.nnzO > 0:
+                optr, ocols, ovals = sell_from_csr(L.O_rowptr, L.O_colidx,
+                                                   L.O_vals)
+                self.sellO = (up(optr), up(ocols), up(ovals))
+            return (int(sellptr[-1]) - L.nnzA) / max(L.nnzA, 1)
+
+        def mk_bsell():
+            for dof_try in (3, 2):
+                out = bsell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals, dof_try)
+                if out is None:
+                    continue
+                bptr_h, bcol_h, bvals_h, density = out
+                if density >= 0.75:
+                    self.bsell = (up(bptr_h), up(bcol_h), up(bvals_h), dof_try)
+                    return True
+            return False
+
+        if force == "csr":
+            return
+        if force == "hybrid":
+            mk_hybrid()
+            return
+        if force == "sell":
+            mk_sell(1)
+        elif force == "sigma":
+            mk_sell(4096)
+        elif force == "bsell":
+            if not mk_bsell():
+                raise ValueError("force_format=bsell: no dense dof x dof "
+                                 "block structure (density < 0.75)")
+        elif use_sell:  # auto ladder
+            waste = mk_sell(1)
+            if waste > 0.3:
+                self.sell = self.sellO = self.sell_perm = None
+                waste = mk_sell(16)
+                if waste > 0.5:
+                    self.sell = self.sellO = self.sell_perm = None
+                    waste = mk_sell(4096)
+                    if waste > 0.35:
+                        self.sell = self.sellO = self.sell_perm = None
+                        mk_hybrid()
+            mk_bsell()
+        if self.sell is not None or self.bsell is not None:
+            self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR
 
     def _workspace(self, key: str, names) -> dict:
         """Per-method persistent vectors, base-address STAGGERED: equal-size
@@ -231,6 +272,11 @@ class CGSolverHIP:
                 ops.spmv_sell(sellptr, scols, svals, self.n, xfull, y,
                               accum=False, perm=self.sell_perm,
                               dot_accum=False, **fuse)
+            elif self.hybrid is not None:
+                rowlist, bins = self.hybrid
+                ops.spmv_binned(self.A_rowptr, self.A_colidx, self.A_vals,
+                                rowlist, bins, xfull, y, accum=False,
+                                dot_accum=False, **fuse)
             else:
                 ops.spmv(self.A_rowptr, self.A_colidx, self.A_vals, xfull, y,
                          lanes=self.lanesA, accum=False, dot_accum=False,

@@ -36,10 +36,20 @@ def main() -> int:
                          "(overlapped allreduce) for multi-GPU; on 1 GPU "
                          "wide rows a 30-it warmup probe picks classic vs "
                          "pipelined (their ordering flips per instance)")
-    ap.add_argument("--config", choices=["queen", "flan", "poisson7"], default="queen",
+    ap.add_argument("--config", choices=["queen", "flan", "poisson7", "irregular"],
+                    default="queen",
                     help="queen: 27-pt dof-3 Queen_4147-shaped (BASELINE configs 3-4); "
                          "flan: Flan_1565-shaped, 1.56M rows (BASELINE config 2); "
-                         "poisson7: 7-pt 3D Poisson (BASELINE config 5 sizing)")
+                         "poisson7: 7-pt 3D Poisson (BASELINE config 5 sizing); "
+                         "irregular: power-law-degree SPD (the merge-path "
+                         "regime -- heavy-tailed rows, no block structure)")
+    ap.add_argument("--rows", type=int, default=2_000_000,
+                    help="irregular config: global row count")
+    ap.add_argument("--mean-nnz", type=float, default=40.0,
+                    help="irregular config: target mean nonzeros/row")
+    ap.add_argument("--format", choices=["auto", "sell", "sigma", "csr",
+                                         "bsell", "hybrid"], default="auto",
+                    help="force the matA operator format (A/B measurement)")
     ap.add_argument("--grid", type=int, default=None,
                     help="grid edge G (queen default 111 -> 4.10M rows; "
                          "poisson7 default 512)")
@@ -86,6 +96,9 @@ def main() -> int:
         G = args.grid or 80
         spec = queen_like_spec(dof)
         model = f"flan1565-like-27pt-dof{dof}-G{G}"
+    elif args.config == "irregular":
+        dof, G = 1, 0
+        model = f"powerlaw-n{args.rows}-m{args.mean_nnz:g}"
     else:
         dof = args.dof or 1
         G = args.grid or 512
@@ -94,15 +107,40 @@ def main() -> int:
         model = f"poisson3d-7pt-G{G}"
     if args.matfree and (dof != 1 or args.gen != "device"):
         raise SystemExit("--matfree needs a dof=1 stencil config with --gen device")
-    if args.gen == "device":
+    if args.config == "irregular":
+        # host-generated power-law SPD, deterministic across ranks: every
+        # rank builds the same global matrix and extracts its own part
+        # through the GENERIC extractor (the path real .mtx input takes)
+        from acg_amd.gen.irregular import powerlaw_spd
+        from acg_amd.part import extract_subdomains, partition_rows
+
+        A = powerlaw_spd(args.rows, mean_nnz=args.mean_nnz, seed=12345)
+        part = partition_rows(A, ngpus, seed=0,
+                              method="auto" if ngpus > 1 else "block")
+        S = extract_subdomains(A, part, ngpus, only_parts=[rank])[rank] \
+            if ngpus > 1 else extract_subdomains(A, part, 1)[0]
+        del A
+        nrows_global = args.rows
+    elif args.gen == "device":
         from acg_amd.gen.device_slab import device_stencil_slab
 
         S = device_stencil_slab(G, G, G, spec, rank, ngpus, device,
                                 operator=not args.matfree)
+        nrows_global = dof * G * G * G
     else:
         S = stencil_local_slab(G, G, G, spec, rank, ngpus)
+        nrows_global = dof * G * G * G
     solver = CGSolverHIP(S, comm=comm, device=device, lanes=args.lanes,
-                         matfree=args.matfree)
+                         matfree=args.matfree,
+                         force_format=None if args.format == "auto"
+                         else args.format)
+    if rank == 0:
+        fmt = ("matfree" if solver.matfree is not None else
+               "bsell" if solver.bsell is not None else
+               "sigma-sell" if solver.sell_perm is not None else
+               "sell" if solver.sell is not None else
+               "hybrid-binned" if solver.hybrid is not None else "csr-vector")
+        print(f"# operator format: {fmt}", file=sys.stderr, flush=True)
 
     # dry-run halo audit: collective cross-rank check of pairing symmetry
     # and in-place ghost-tail global-id agreement, so the first N>=2 run
