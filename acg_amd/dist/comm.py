@@ -57,6 +57,15 @@ class Comm:
         if kind == "rccl" and device is None:
             self.device = torch.device("cuda", self.rank % max(torch.cuda.device_count(), 1))
 
+    @property
+    def can_capture(self) -> bool:
+        """True when this backend's collectives are hipGraph-capturable
+        (RCCL on-stream).  The gloo test backend stages CUDA tensors
+        through blocking host copies -- attempting capture there not only
+        fails but can invalidate cross-stream event state mid-body, so
+        callers must not even TRY (solvers consult this before capture)."""
+        return self.kind == "rccl"
+
     # -- collectives ------------------------------------------------------
 
     def allreduce_(self, t: torch.Tensor) -> torch.Tensor:

@@ -405,6 +405,7 @@ class CGSolverHIP:
         dist_failed_key = "classic:capture_failed"
         dist_graph_ok = (use_graph and not serial and not self.prof.enabled
                          and not fold
+                         and getattr(self.comm, "can_capture", False)
                          and os.environ.get("ACG_DIST_GRAPH", "1") != "0"
                          and not self._graphs.get(dist_failed_key, False))
         dgraph = self._graphs.get(dist_key) if dist_graph_ok else None
@@ -800,6 +801,7 @@ class CGSolverHIP:
         # ACG_DIST_GRAPH=0 is the operational kill switch for the captured
         # distributed iteration (falls back to the eager path everywhere)
         dist_graph_ok = (use_graph and not serial and not self.prof.enabled
+                         and getattr(self.comm, "can_capture", False)
                          and os.environ.get("ACG_DIST_GRAPH", "1") != "0"
                          and not self._graphs.get(dist_failed_key, False))
         dgraph = self._graphs.get(dist_key) if dist_graph_ok and not mega else None
