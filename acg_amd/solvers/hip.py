@@ -145,9 +145,10 @@ class CGSolverHIP:
         Auto ladder by measured fit:
           1. plain SELL-C-64 when padding waste <= 0.3 (stencil/FEM rows),
           2. sigma-SELL (window 16) when waste <= 0.5 (mildly irregular),
-          3. wide-window sigma-SELL (4096) when that gets waste <= 0.35,
-          4. row-binned hybrid CSR otherwise (power-law rows: per-bin lane
-             counts keep per-row work proportional to row length),
+          3. row-binned hybrid CSR otherwise (power-law rows; MEASURED on
+             MI355X 1M-row power-law: hybrid 665 us/it vs single-lane CSR
+             855 vs wide-sigma SELL 2627 -- slice-length imbalance makes
+             wide sigma lose badly, so it is force-only),
         plus Block-SELL whenever dense dof x dof block structure is found
         (density >= 0.75) -- it wins on index bytes.
         ``force`` in {csr, sell, sigma, bsell, hybrid} overrides for A/B
@@ -207,10 +208,7 @@ class CGSolverHIP:
                 waste = mk_sell(16)
                 if waste > 0.5:
                     self.sell = self.sellO = self.sell_perm = None
-                    waste = mk_sell(4096)
-                    if waste > 0.35:
-                        self.sell = self.sellO = self.sell_perm = None
-                        mk_hybrid()
+                    mk_hybrid()
             mk_bsell()
         if self.sell is not None or self.bsell is not None:
             self.A_rowptr = self.A_colidx = self.A_vals = None  # free CSR
