@@ -268,7 +268,6 @@ def main() -> int:
         if rank == 0:
             rank_spans = {f"r{i}": d for i, d in enumerate(allspans)}
 
-    nrows_global = dof * G * G * G
     nnz_local = S.nnzA + S.nnzO
     if comm is not None:
         import torch.distributed as dist
@@ -287,6 +286,7 @@ def main() -> int:
             "metric": {"queen": "CG iter/s (whole node), Queen_4147-shaped fp64",
                        "flan": "CG iter/s (whole node), Flan_1565-shaped fp64",
                        "poisson7": "CG iter/s (whole node), 7-pt 3D Poisson fp64",
+                       "irregular": "CG iter/s (whole node), power-law SPD fp64",
                        }[args.config],
             "value": iters_per_s,
             "unit": "iter/s",
@@ -301,6 +301,8 @@ def main() -> int:
             "data": {"queen": "synthetic SPD (27-pt dof-3 stencil, Queen_4147 shape; random RHS)",
                      "flan": "synthetic SPD (27-pt dof-3 stencil, Flan_1565 shape; random RHS)",
                      "poisson7": "synthetic SPD (7-pt 3D Poisson; random RHS)",
+                     "irregular": "synthetic SPD (power-law degrees, "
+                                  "Laplace-local columns; random RHS)",
                      }[args.config],
             "config": {
                 "model": model,
