@@ -113,6 +113,37 @@ def build_row_bins(rowptr, max_lanes: int = 64):
     return order, bins
 
 
+def build_sellcsr_hybrid(rowptr, colidx, vals, cut: int = 96):
+    """Host prep for the SELL+CSR hybrid: short rows (len <= cut) as
+    sigma-sorted SELL, long rows as a longest-first 64/32-lane binned CSR
+    list.  Returns (sellptr, cols, svals, perm, rowlist_long, bins) numpy
+    arrays (sellptr None when every row is long)."""
+    import numpy as np
+
+    from .torch_ref import sellcsr_split
+
+    sellptr, cols, svals, perm, long_rows, nshort = sellcsr_split(
+        rowptr, colidx, vals, cut=cut)
+    lens = np.diff(np.asarray(rowptr))
+    longlens = lens[long_rows.astype(np.int64)]  # descending
+    bins = []
+    lo = 0
+    for lanes in (64, 32, 16, 8, 4):
+        if lanes == 4:
+            hi = len(longlens)
+        else:
+            hi = int(np.searchsorted(-longlens, -((lanes // 2) * 6) - 1,
+                                     side="right"))
+        if hi > lo:
+            bins.append((lo, hi - lo, lanes))
+            lo = hi
+        if lo >= len(longlens):
+            break
+    if nshort == 0:
+        sellptr = None
+    return sellptr, cols, svals, perm, long_rows, bins
+
+
 def spmv_binned(rowptr: torch.Tensor, colidx: torch.Tensor, vals: torch.Tensor,
                 rowlist: torch.Tensor, bins, x: torch.Tensor, y: torch.Tensor,
                 *, rowbase: int = 0, accum: bool = False,

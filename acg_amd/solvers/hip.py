@@ -158,9 +158,28 @@ class CGSolverHIP:
         if L.nowned == 0:
             return
 
-        def mk_hybrid():
+        def mk_hybrid(cut: int = 96):
+            """SELL+CSR split (short rows SELL, long tail binned CSR).
+            MEASURED (MI355X 1M-row power-law): the pure binned hybrid's
+            4/8-lane short-row bins cost 400 us/it of its 587 us SpMV --
+            SELL's lockstep 512 B line loads serve those rows instead.
+            Costs ~1.75x operator memory (CSR kept for the long rows)."""
+            sp_, cols, svals, perm, rowlist, bins = ops.build_sellcsr_hybrid(
+                L.A_rowptr, L.A_colidx, L.A_vals, cut=cut)
+            self.hybrid = {
+                "sellptr": up(sp_) if sp_ is not None else None,
+                "cols": up(cols) if sp_ is not None else None,
+                "svals": up(svals) if sp_ is not None else None,
+                "perm": up(perm) if sp_ is not None else None,
+                "rowlist": up(rowlist) if len(rowlist) else None,
+                "bins": bins,
+            }
+
+        def mk_binned():
             rowlist, bins = ops.build_row_bins(L.A_rowptr)
-            self.hybrid = (up(rowlist), bins)
+            self.hybrid = {"sellptr": None, "cols": None, "svals": None,
+                           "perm": None, "rowlist": up(rowlist),
+                           "bins": bins}
 
         def mk_sell(sigma):
             out = sell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals, sigma=sigma)
@@ -192,6 +211,9 @@ class CGSolverHIP:
             return
         if force == "hybrid":
             mk_hybrid()
+            return
+        if force == "binned":
+            mk_binned()
             return
         if force == "sell":
             mk_sell(1)
@@ -271,10 +293,20 @@ class CGSolverHIP:
                               accum=False, perm=self.sell_perm,
                               dot_accum=False, **fuse)
             elif self.hybrid is not None:
-                rowlist, bins = self.hybrid
-                ops.spmv_binned(self.A_rowptr, self.A_colidx, self.A_vals,
-                                rowlist, bins, xfull, y, accum=False,
-                                dot_accum=False, **fuse)
+                h = self.hybrid
+                have_sell = h["sellptr"] is not None
+                if have_sell:
+                    ops.spmv_sell(h["sellptr"], h["cols"], h["svals"],
+                                  self.n, xfull, y, accum=False,
+                                  perm=h["perm"], dot_accum=False, **fuse)
+                if h["rowlist"] is not None:
+                    # rows are disjoint from the SELL set: y writes stay
+                    # overwrite-mode; the fused dot ACCUMULATES on top of
+                    # the SELL part's contribution
+                    ops.spmv_binned(self.A_rowptr, self.A_colidx,
+                                    self.A_vals, h["rowlist"], h["bins"],
+                                    xfull, y, accum=False,
+                                    dot_accum=have_sell, **fuse)
             else:
                 ops.spmv(self.A_rowptr, self.A_colidx, self.A_vals, xfull, y,
                          lanes=self.lanesA, accum=False, dot_accum=False,

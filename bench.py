@@ -48,7 +48,8 @@ def main() -> int:
     ap.add_argument("--mean-nnz", type=float, default=40.0,
                     help="irregular config: target mean nonzeros/row")
     ap.add_argument("--format", choices=["auto", "sell", "sigma", "csr",
-                                         "bsell", "hybrid"], default="auto",
+                                         "bsell", "hybrid", "binned"],
+                    default="auto",
                     help="force the matA operator format (A/B measurement)")
     ap.add_argument("--grid", type=int, default=None,
                     help="grid edge G (queen default 111 -> 4.10M rows; "

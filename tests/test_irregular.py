@@ -104,3 +104,48 @@ def test_hybrid_torch_fallback_matches_spmv():
         rows = rowlist[start:start + count].astype(np.int64)
         y[rows] = M[rows] @ x
     np.testing.assert_allclose(y, M @ x, rtol=1e-12)
+
+
+def test_sellcsr_split_structure():
+    """Every entry lands exactly once in SELL or stays CSR-long; perm
+    sentinel and bin thresholds consistent."""
+    from acg_amd.ops.gpu_ops import build_sellcsr_hybrid
+
+    A = powerlaw_spd(8000, mean_nnz=30, seed=2)
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    cut = 96
+    sellptr, cols, svals, perm, rowlist, bins = build_sellcsr_hybrid(
+        S.A_rowptr, S.A_colidx, S.A_vals, cut=cut)
+    n = S.nowned
+    lens = np.diff(S.A_rowptr)
+    short = set(np.where(lens <= cut)[0].tolist())
+    long_ = set(np.where(lens > cut)[0].tolist())
+    assert set(rowlist.tolist()) == long_
+    sell_rows = set(int(r) for r in perm if r < n)
+    assert sell_rows == short
+    # SELL holds exactly the short rows' values
+    assert float(np.abs(svals).sum()) == pytest.approx(
+        float(np.abs(S.A_vals[np.isin(np.repeat(np.arange(n), lens),
+                                      list(short))]).sum()), rel=1e-12)
+    # reconstruct y = A x from the two halves on CPU
+    import scipy.sparse as sp
+
+    M = sp.csr_matrix((S.A_vals, S.A_colidx.astype(np.int64), S.A_rowptr),
+                      shape=(n, n))
+    x = np.random.default_rng(0).standard_normal(n)
+    y = np.zeros(n)
+    C = 64
+    nslices = len(sellptr) - 1
+    for s in range(nslices):
+        base = int(sellptr[s])
+        L = (int(sellptr[s + 1]) - base) // C
+        blk = (svals[base:base + L * C].reshape(L, C)
+               * x[cols[base:base + L * C].astype(np.int64)].reshape(L, C)
+               ).sum(axis=0)
+        for lane in range(min(C, nslices * C - s * C)):
+            r = int(perm[s * C + lane])
+            if r < n:
+                y[r] = blk[lane]
+    lr = rowlist.astype(np.int64)
+    y[lr] = M[lr] @ x
+    np.testing.assert_allclose(y, M @ x, rtol=1e-12)

@@ -26,7 +26,7 @@ def main() -> int:
     ap.add_argument("--clip", type=int, default=8192)
     ap.add_argument("--reps", type=int, default=5)
     ap.add_argument("--its", type=int, default=30)
-    ap.add_argument("--formats", default="sigma,hybrid,csr")
+    ap.add_argument("--formats", default="hybrid,binned,csr")
     args = ap.parse_args()
 
     from acg_amd.gen.irregular import degree_stats, powerlaw_spd
