@@ -396,9 +396,50 @@ def test_powerlaw_auto_format_is_hybrid_or_sigma(dev):
     assert solver.hybrid is not None or solver.sell_perm is not None
 
 
+def test_sellcsr_split_spmv_exact(dev):
+    """SELL+CSR split SpMV (sell part + binned long part, fused dot) is
+    EXACT vs the torch reference on the heavy tail -- the per-iteration
+    building block, free of CG rounding amplification."""
+    from acg_amd.gen.irregular import powerlaw_spd
+    from acg_amd.ops import gpu_ops, torch_ref
+    from acg_amd.part import extract_subdomains, partition_rows
+
+    A = powerlaw_spd(50_000, mean_nnz=35, seed=11)
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    n = S.nowned
+    sellptr, cols, svals, perm, rowlist, bins = gpu_ops.build_sellcsr_hybrid(
+        S.A_rowptr, S.A_colidx, S.A_vals)
+    assert sellptr is not None and len(rowlist)
+    rowptr_t = torch.from_numpy(S.A_rowptr)
+    colidx_t = torch.from_numpy(S.A_colidx.astype(np.int32))
+    vals_t = torch.from_numpy(S.A_vals)
+    x = torch.randn(n, dtype=torch.float64)
+    y_ref = torch.zeros(n, dtype=torch.float64)
+    torch_ref.spmv(rowptr_t, colidx_t, vals_t, x, y_ref)
+    xg = x.to(dev)
+    yg = torch.zeros(n, dtype=torch.float64, device=dev)
+    scal = gpu_ops.alloc_scalars(dev)
+    partials = gpu_ops.alloc_partials(dev)
+    gpu_ops.spmv_sell(torch.from_numpy(sellptr).to(dev),
+                      torch.from_numpy(cols).to(dev),
+                      torch.from_numpy(svals).to(dev), n, xg, yg,
+                      perm=torch.from_numpy(perm).to(dev),
+                      partials=partials, scal=scal,
+                      dotslot=gpu_ops.S_PT, dot_accum=False)
+    gpu_ops.spmv_binned(rowptr_t.to(dev), colidx_t.to(dev), vals_t.to(dev),
+                        torch.from_numpy(rowlist).to(dev), bins, xg, yg,
+                        partials=partials, scal=scal,
+                        dotslot=gpu_ops.S_PT, dot_accum=True)
+    torch.testing.assert_close(yg.cpu(), y_ref, rtol=1e-12, atol=1e-10)
+    want = float(torch.dot(x, y_ref))
+    assert abs(float(scal[gpu_ops.S_PT]) - want) < 1e-8 * max(1.0, abs(want))
+
+
 def test_sellcsr_hybrid_solver_heavy_tail(dev):
-    """SELL+CSR split format: GPU solve on the HEAVY tail matrix matches
-    the pure-binned format bit-for-bit-level (same system, fixed its)."""
+    """Solver-level split-format run on the HEAVY tail: converges like the
+    pure-binned format (cross-format iterates only agree to rounding-
+    amplified tolerance at kappa~5e3, so compare residual QUALITY; exact
+    SpMV equality is test_sellcsr_split_spmv_exact)."""
     from acg_amd.gen.irregular import powerlaw_spd
     from acg_amd.part import extract_subdomains, partition_rows
     from acg_amd.solvers.hip import CGSolverHIP
@@ -407,7 +448,7 @@ def test_sellcsr_hybrid_solver_heavy_tail(dev):
     S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
     rng = np.random.default_rng(1)
     b_np = rng.standard_normal(S.nowned)
-    xs = {}
+    rnorm = {}
     for fmt in ("hybrid", "binned"):
         solver = CGSolverHIP(S, device=dev, force_format=fmt)
         if fmt == "hybrid":
@@ -417,10 +458,9 @@ def test_sellcsr_hybrid_solver_heavy_tail(dev):
         x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device=dev)
         res = solver.solve(b, x, maxits=120, res_rtol=0.0)
         assert res.niterations == 120
-        xs[fmt] = x[:S.nowned].cpu().numpy()
-        # true residual must have dropped
-        r = b_np - (A.to_scipy_full() @ xs[fmt])
-        assert np.linalg.norm(r) < 1e-2 * np.linalg.norm(b_np), fmt
-    # same Krylov trajectory up to rounding: iterates agree tightly
-    np.testing.assert_allclose(xs["hybrid"], xs["binned"], rtol=1e-6,
-                               atol=1e-8)
+        xh = x[:S.nowned].cpu().numpy()
+        rnorm[fmt] = np.linalg.norm(b_np - (A.to_scipy_full() @ xh))
+        assert rnorm[fmt] < 1e-2 * np.linalg.norm(b_np), fmt
+    # equivalent convergence quality (same algorithm, different rounding)
+    ratio = rnorm["hybrid"] / rnorm["binned"]
+    assert 0.5 < ratio < 2.0, rnorm
