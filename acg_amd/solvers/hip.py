@@ -158,12 +158,15 @@ class CGSolverHIP:
         if L.nowned == 0:
             return
 
-        def mk_hybrid(cut: int = 96):
+        def mk_hybrid(cut: int | None = None):
             """SELL+CSR split (short rows SELL, long tail binned CSR).
             MEASURED (MI355X 1M-row power-law): the pure binned hybrid's
             4/8-lane short-row bins cost 400 us/it of its 587 us SpMV --
             SELL's lockstep 512 B line loads serve those rows instead.
-            Costs ~1.75x operator memory (CSR kept for the long rows)."""
+            Costs ~1.75x operator memory (CSR kept for the long rows).
+            ACG_HYBRID_CUT overrides the split length (tuning sweeps)."""
+            if cut is None:
+                cut = int(os.environ.get("ACG_HYBRID_CUT", "96"))
             sp_, cols, svals, perm, rowlist, bins = ops.build_sellcsr_hybrid(
                 L.A_rowptr, L.A_colidx, L.A_vals, cut=cut)
             self.hybrid = {
