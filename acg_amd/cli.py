@@ -43,7 +43,8 @@ def make_parser() -> argparse.ArgumentParser:
                    help="precomputed partition vector (mtx integer array)")
     p.add_argument("--binary-partition", action="store_true",
                    help="partition file is in binary Matrix Market format")
-    p.add_argument("--partition-method", choices=("block", "rgb"), default="block")
+    p.add_argument("--partition-method",
+                   choices=("block", "rgb", "ml", "auto"), default="auto")
     p.add_argument("--seed", type=int, default=0, help="partitioner seed")
     p.add_argument("--solver", default=None,
                    choices=("acg", "acg-pipelined", "acg-device", "cpu",

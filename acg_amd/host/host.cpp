@@ -343,6 +343,42 @@ py::tuple bsell_blocks(py::array_t<i64, py::array::c_style | py::array::forcecas
     return py::make_tuple(counts_a, bcols_a, entry_a);
 }
 
+// ---------------------------------------------------------------------------
+// Heavy-edge matching for multilevel partitioning (reference: the METIS
+// coarsening stage wrapped by acg/metis.c:80-436).  Greedy sequential:
+// visit vertices in the caller's (random) order, match each unmatched
+// vertex with its unmatched neighbour of maximum edge weight; ties break
+// toward the first-seen neighbour.  match[v] = partner (== v when no
+// unmatched neighbour exists).
+py::array_t<i64> hem_match(py::array_t<i64, py::array::c_style | py::array::forcecast> rowptr,
+                           py::array_t<i64, py::array::c_style | py::array::forcecast> colidx,
+                           py::array_t<double, py::array::c_style | py::array::forcecast> wts,
+                           py::array_t<i64, py::array::c_style | py::array::forcecast> order) {
+    const i64* rp = rowptr.data();
+    const i64* ci = colidx.data();
+    const double* w = wts.data();
+    const i64* ord = order.data();
+    const i64 n = (i64)rowptr.shape(0) - 1;
+    py::array_t<i64> match_a(n);
+    i64* match = match_a.mutable_data();
+    for (i64 i = 0; i < n; ++i) match[i] = -1;
+    for (i64 k = 0; k < n; ++k) {
+        i64 v = ord[k];
+        if (v < 0 || v >= n) throw std::out_of_range("hem_match: order");
+        if (match[v] >= 0) continue;
+        i64 best = -1;
+        double bw = -1.0;
+        for (i64 e = rp[v]; e < rp[v + 1]; ++e) {
+            i64 u = ci[e];
+            if (u == v || u < 0 || u >= n || match[u] >= 0) continue;
+            if (w[e] > bw) { bw = w[e]; best = u; }
+        }
+        if (best >= 0) { match[v] = best; match[best] = v; }
+        else match[v] = v;
+    }
+    return match_a;
+}
+
 // in-place scans (reference acgprefixsum_inplace_*, prefixsum.h:72-116)
 py::array_t<i64> prefix_sum(py::array_t<i64, py::array::c_style | py::array::forcecast> a,
                             bool inclusive) {
@@ -375,5 +411,6 @@ PYBIND11_MODULE(_acg_host, m) {
     m.def("coo_to_sym_csr", &coo_to_sym_csr);
     m.def("prefix_sum", &prefix_sum, py::arg("a"), py::arg("inclusive") = true);
     m.def("bsell_blocks", &bsell_blocks);
+    m.def("hem_match", &hem_match);
     m.def("num_threads", &num_threads);
 }

