@@ -148,16 +148,19 @@ def main(argv=None) -> int:
 
     numfmt = parse_numfmt(args.numfmt) if args.numfmt else None
 
-    # ---- root: read + partition + extract (reference acg-hip.c:1267-1652)
-    systems = None
-    b_pieces = None
+    # ---- root: read + partition + structure pass (acg-hip.c:1267-1652)
+    ex = None
+    b_global = None
+    x0_global = None
+    xsol = None
     n_global = None
     err = None
     try:
         if rank == 0:
             from .core.symcsr import SymCSRMatrix
             from .io.mtx import read_mtx
-            from .part import extract_subdomains, partition_rows, read_partition_file
+            from .part import partition_rows, read_partition_file
+            from .part.subdomain import SubdomainExtractor
 
             t0 = time.perf_counter()
             m = read_mtx(args.A, gzipped=args.gzip, binary=args.binary,
@@ -177,8 +180,8 @@ def main(argv=None) -> int:
                 part = partition_rows(A, nparts, seed=args.seed,
                                       method=args.partition_method)
             t0 = time.perf_counter()
-            systems = extract_subdomains(A, part, nparts, eps=args.epsilon)
-            log(f"partitioned into {nparts} subdomains "
+            ex = SubdomainExtractor(A, part, nparts, eps=args.epsilon)
+            log(f"structure pass for {nparts} subdomains "
                 f"({time.perf_counter() - t0:.2f}s)")
             n_global = A.n
 
@@ -191,30 +194,32 @@ def main(argv=None) -> int:
             elif args.b:
                 mb = read_mtx(args.b, gzipped=args.gzip)
                 b_global = _vector_from_mtx(mb, A.n, "b")
-                xsol = None
             else:
                 b_global = np.ones(A.n, dtype=np.float64)
-                xsol = None
-            x0_global = None
             if args.x0:
                 mx = read_mtx(args.x0, gzipped=args.gzip)
                 x0_global = _vector_from_mtx(mx, A.n, "x0")
-            b_pieces = [(b_global[S.owned_global],
-                         None if x0_global is None else x0_global[S.owned_global])
-                        for S in systems]
     except Exception as e:  # collective error agreement (acgerrmpi)
         err = e
     collective_raise(comm, err)
 
-    # ---- scatter (reference acgsymcsrmatrix_scatter, acg-hip.c:1752)
+    # ---- scatter: parts streamed one at a time, fields as chunked
+    # tensors (reference acgsymcsrmatrix_scatter field-by-field with
+    # MPI_Send64 chunks, graph.c:1529-1893; never whole-object pickles)
     t0 = time.perf_counter()
-    if comm:
-        S = comm.scatter_object(systems)
-        b_local, x0_local = comm.scatter_object(b_pieces)
+    if comm and comm.size > 1:
+        S = comm.scatter_systems((lambda p: ex.build(p)) if rank == 0 else None)
+        b_local = comm.scatter_rows(
+            (lambda p: b_global[ex.owned_globals[p]]) if rank == 0 else None)
+        has_x0 = comm.bcast_object(x0_global is not None if rank == 0 else None)
+        x0_local = comm.scatter_rows(
+            (lambda p: x0_global[ex.owned_globals[p]]) if rank == 0 else None) \
+            if has_x0 else None
         n_global = comm.bcast_object(n_global)
     else:
-        S = systems[0]
-        b_local, x0_local = b_pieces[0]
+        S = ex.build(0)
+        b_local = b_global[S.owned_global]
+        x0_local = None if x0_global is None else x0_global[S.owned_global]
     log(f"scattered subdomains ({time.perf_counter() - t0:.2f}s)")
     if args.verbose and args.verbose > 1:
         S.dump(file=sys.stderr)

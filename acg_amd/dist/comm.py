@@ -56,6 +56,14 @@ class Comm:
         self.group = dist.group.WORLD
         if kind == "rccl" and device is None:
             self.device = torch.device("cuda", self.rank % max(torch.cuda.device_count(), 1))
+        # host-side bulk transport group: setup-phase scatters move large
+        # CPU arrays (CSR fields), which RCCL cannot carry -- a parallel
+        # gloo group serves them (reference: the MPI side of the RCCL
+        # build plays this role)
+        if kind == "rccl" and self.size > 1:
+            self._host_group = dist.new_group(backend="gloo")
+        else:
+            self._host_group = self.group
 
     @property
     def can_capture(self) -> bool:
@@ -102,6 +110,116 @@ class Comm:
             self._dist.barrier(device_ids=[self.device.index])
         else:
             self._dist.barrier()
+
+    # -- setup-phase bulk transport (chunked tensor send/recv) ------------
+    # Reference: acgsymcsrmatrix_scatter / acggraph_scatter field-by-field
+    # MPI_Send/Recv with chunked MPI_Send64 for >2^31 counts
+    # (graph.c:1529-1893, symcsrmatrix.c:1005-1330).  Pickled-object
+    # scatters cap out around 100M rows (4 GB pickles through the store);
+    # these move raw numpy buffers as uint8 tensor chunks over the host
+    # (gloo) group.
+
+    CHUNK_BYTES = 1 << 30
+    _DTYPES = [np.dtype(t) for t in
+               ("int64", "int32", "float64", "float32", "uint8")]
+
+    def _send_array(self, a, dst: int) -> None:
+        a = np.ascontiguousarray(a)
+        code = next(i for i, d in enumerate(self._DTYPES) if d == a.dtype)
+        hdr = torch.tensor([code, a.size], dtype=torch.int64)
+        self._dist.send(hdr, dst, group=self._host_group)
+        if a.size:
+            buf = torch.from_numpy(a.view(np.uint8))
+            for off in range(0, buf.numel(), self.CHUNK_BYTES):
+                self._dist.send(buf[off:off + self.CHUNK_BYTES], dst,
+                                group=self._host_group)
+
+    def _recv_array(self, src: int) -> np.ndarray:
+        hdr = torch.zeros(2, dtype=torch.int64)
+        self._dist.recv(hdr, src, group=self._host_group)
+        code, count = int(hdr[0]), int(hdr[1])
+        if code < 0:
+            from ..utils.errors import AcgError, ErrCode
+
+            raise AcgError(ErrCode.ERRNO, "scatter aborted on the root rank")
+        dt = self._DTYPES[code]
+        out = np.empty(count * dt.itemsize, dtype=np.uint8)
+        t = torch.from_numpy(out)
+        for off in range(0, t.numel(), self.CHUNK_BYTES):
+            self._dist.recv(t[off:off + self.CHUNK_BYTES], src,
+                            group=self._host_group)
+        return out.view(dt)
+
+    _SYS_ARRAYS = ("A_rowptr", "A_colidx", "A_vals", "O_rowptr", "O_colidx",
+                   "O_vals", "owned_global", "ghost_global")
+    _HALO_ARRAYS = ("senders", "recvcounts", "rdispls", "recipients",
+                    "sendcounts", "sdispls", "sendidx")
+
+    def scatter_systems(self, factory, src: int = 0):
+        """Stream per-rank LocalSystems from root, field by field.
+
+        Root calls with ``factory(p) -> LocalSystem`` and builds ONE part
+        at a time (peak memory: global operator + one part); other ranks
+        pass ``factory=None`` and receive.  A root-side build failure
+        poisons the remaining ranks' headers so nobody hangs."""
+        if self.kind == "none" or self.size == 1:
+            return factory(0)
+        from ..part.subdomain import HaloPattern, LocalSystem
+
+        if self.rank == src:
+            keep = None
+            sent = []
+            try:
+                for p in range(self.size):
+                    S = factory(p)
+                    if p == src:
+                        keep = S
+                        continue
+                    meta = torch.tensor(
+                        [S.nparts, S.n_global, S.nowned, S.ninterior,
+                         S.nborder, S.nghost], dtype=torch.int64)
+                    self._dist.send(meta, p, group=self._host_group)
+                    sent.append(p)
+                    for nm in self._SYS_ARRAYS:
+                        self._send_array(getattr(S, nm), p)
+                    for nm in self._HALO_ARRAYS:
+                        self._send_array(getattr(S.halo, nm), p)
+                    del S
+            except Exception:
+                poison = torch.tensor([-1, -1], dtype=torch.int64)
+                for p in range(self.size):
+                    if p != src and p not in sent:
+                        self._dist.send(poison, p, group=self._host_group)
+                raise
+            return keep
+        meta = torch.zeros(6, dtype=torch.int64)
+        self._dist.recv(meta, src, group=self._host_group)
+        if int(meta[0]) < 0:
+            from ..utils.errors import AcgError, ErrCode
+
+            raise AcgError(ErrCode.ERRNO, "scatter aborted on the root rank")
+        arrs = {nm: self._recv_array(src) for nm in self._SYS_ARRAYS}
+        halo = HaloPattern(**{nm: self._recv_array(src)
+                              for nm in self._HALO_ARRAYS})
+        return LocalSystem(
+            rank=self.rank, nparts=int(meta[0]), n_global=int(meta[1]),
+            nowned=int(meta[2]), ninterior=int(meta[3]),
+            nborder=int(meta[4]), nghost=int(meta[5]), halo=halo, **arrs)
+
+    def scatter_rows(self, factory, src: int = 0):
+        """Scatter one numpy array per rank (root: factory(p) -> array)."""
+        if self.kind == "none" or self.size == 1:
+            return factory(0)
+        if self.rank == src:
+            keep = None
+            for p in range(self.size):
+                a = factory(p)
+                if p == src:
+                    keep = a
+                else:
+                    self._send_array(np.ascontiguousarray(a), p)
+            return keep
+        return self._recv_array(src)
 
     # -- setup-phase object transport (root-centric scatter) --------------
 

@@ -179,6 +179,53 @@ def _body_halo_audit(comm):
         return ("raised",)
 
 
+def _body_chunked_scatter(comm):
+    """Streaming field-by-field scatter (comm.scatter_systems) must equal
+    direct local extraction; tiny CHUNK_BYTES exercises the chunk loop
+    (reference MPI_Send64 chunking, graph.c:1529-1893)."""
+    from acg_amd.dist.comm import Comm
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.part.subdomain import SubdomainExtractor
+
+    Comm.CHUNK_BYTES = 64  # force many chunks per array
+    A = stencil_global(6, 6, 6, STENCIL_27PT_3D)
+    part = partition_rows(A, comm.size, method="rgb", seed=1)
+    if comm.rank == 0:
+        ex = SubdomainExtractor(A, part, comm.size)
+        S = comm.scatter_systems(lambda p: ex.build(p))
+        rng = np.random.default_rng(3)
+        bg = rng.standard_normal(A.n)
+        b = comm.scatter_rows(lambda p: bg[ex.owned_globals[p]])
+    else:
+        S = comm.scatter_systems(None)
+        b = comm.scatter_rows(None)
+    # oracle: local extraction of the same part
+    ref = extract_subdomains(A, part, comm.size,
+                             only_parts=[comm.rank])[comm.rank]
+    ok = True
+    for fld in ("nowned", "ninterior", "nborder", "nghost", "n_global"):
+        ok &= getattr(S, fld) == getattr(ref, fld)
+    import numpy as _np
+
+    for fld in ("A_rowptr", "A_colidx", "A_vals", "O_rowptr", "O_colidx",
+                "O_vals", "owned_global", "ghost_global"):
+        a, r = getattr(S, fld), getattr(ref, fld)
+        ok &= a.dtype == r.dtype and _np.array_equal(a, r)
+    for fld in ("senders", "recvcounts", "rdispls", "recipients",
+                "sendcounts", "sdispls", "sendidx"):
+        ok &= _np.array_equal(getattr(S.halo, fld), getattr(ref.halo, fld))
+    rng = np.random.default_rng(3)
+    bg = rng.standard_normal(A.n)
+    ok &= _np.array_equal(b, bg[ref.owned_global])
+    return ("ok" if ok else "mismatch",)
+
+
+def test_chunked_scatter_ws4():
+    results = _run_dist("_body_chunked_scatter", world=4, port=29607)
+    for rank, (status,) in results.items():
+        assert status == "ok", f"rank {rank}: {status}"
+
+
 def test_halo_audit_ws4():
     results = _run_dist("_body_halo_audit", world=4, port=29606)
     for rank, (status,) in results.items():
