@@ -42,6 +42,42 @@ class DeviceSlabSystem:
         return self._nnzO
 
 
+def estimate_slab_gib(gx: int, gy: int, gz: int, spec: dict, rank: int,
+                      nranks: int, matfree: bool = False) -> float:
+    """Pre-flight per-rank GPU memory estimate (GiB) for a device-generated
+    stencil slab + pipelined solver workspace.
+
+    Calibrated against measured peaks on MI355X (tools/slab_probe.py,
+    profiles/RESULTS.md): 2048^3 7-pt rank 0/8 = 1.074B rows measured
+    157.2 GiB assembled / 80.9 GiB matrix-free.  Model: assembled
+    operator ~ 12 B/nnz (8 B vals + 4 B int32 cols incl. SELL padding)
+    + 73 B/row of vectors/transients; matrix-free 81 B/row flat."""
+    dof = spec["dof"]
+    pts = len(spec["offsets"]) + 1  # neighbours + self
+    z0, z1 = _slab_bounds(gz, rank, nranks)
+    nghost_planes = (1 if rank > 0 else 0) + (1 if rank < nranks - 1 else 0)
+    nlocal = (z1 - z0 + nghost_planes) * gx * gy * dof
+    bytes_per_row = 81.0 if matfree else 12.0 * pts * dof + 73.0
+    return nlocal * bytes_per_row / 2**30
+
+
+def preflight_slab(gx: int, gy: int, gz: int, spec: dict, rank: int,
+                   nranks: int, matfree: bool, total_bytes: int,
+                   headroom: float = 0.94) -> float:
+    """Assert the slab + solver fits this GPU BEFORE any allocation
+    (BASELINE config-5 rehearsal: fail in seconds, not after a 40-GiB
+    partial generation OOMs the box).  Returns the estimate in GiB."""
+    est = estimate_slab_gib(gx, gy, gz, spec, rank, nranks, matfree)
+    budget = headroom * total_bytes / 2**30
+    if est > budget:
+        raise MemoryError(
+            f"rank {rank}/{nranks} of {gx}x{gy}x{gz} needs ~{est:.0f} GiB "
+            f"({'matfree' if matfree else 'assembled'}) but this GPU has "
+            f"{total_bytes / 2**30:.0f} GiB ({budget:.0f} usable); use more "
+            f"ranks, --matfree, or a smaller --grid")
+    return est
+
+
 def device_stencil_slab(gx: int, gy: int, gz: int, spec: dict,
                         rank: int, nranks: int, device,
                         operator: bool = True) -> DeviceSlabSystem:

@@ -123,8 +123,16 @@ def main() -> int:
         del A
         nrows_global = args.rows
     elif args.gen == "device":
-        from acg_amd.gen.device_slab import device_stencil_slab
+        from acg_amd.gen.device_slab import device_stencil_slab, preflight_slab
 
+        # pre-flight memory assertion (config-5 rehearsal): fail in
+        # seconds with a clear message instead of OOMing the box mid-way
+        est = preflight_slab(G, G, G, spec, rank, ngpus, args.matfree,
+                             torch.cuda.get_device_properties(device).total_memory)
+        if rank == 0:
+            print(f"# preflight: ~{est:.1f} GiB/rank "
+                  f"({'matfree' if args.matfree else 'assembled'})",
+                  file=sys.stderr, flush=True)
         S = device_stencil_slab(G, G, G, spec, rank, ngpus, device,
                                 operator=not args.matfree)
         nrows_global = dof * G * G * G

@@ -192,3 +192,34 @@ def test_halo_audit_serial_all_parts():
     victim["send"][q] = victim["send"][q][:-1]
     with pytest.raises(AcgError):
         _audit(bad2)
+
+
+# ---- config-5 pre-flight memory assertions ----
+
+def test_slab_estimate_matches_measured_2048():
+    """Calibration check: the estimator reproduces the slab_probe-measured
+    peaks for rank 0/8 of 2048^3 7-pt (157.2 GiB assembled, 80.9 GiB
+    matfree, profiles/RESULTS.md) within a few percent."""
+    from acg_amd.gen import STENCIL_7PT_3D
+    from acg_amd.gen.device_slab import estimate_slab_gib
+
+    est_a = estimate_slab_gib(2048, 2048, 2048, dict(STENCIL_7PT_3D), 0, 8)
+    est_m = estimate_slab_gib(2048, 2048, 2048, dict(STENCIL_7PT_3D), 0, 8,
+                              matfree=True)
+    assert abs(est_a - 157.2) < 8, est_a
+    assert abs(est_m - 80.9) < 4, est_m
+
+
+def test_preflight_rejects_oversized_grid():
+    from acg_amd.gen import STENCIL_7PT_3D
+    from acg_amd.gen.device_slab import preflight_slab
+
+    total = 288 * 2**30
+    # 2048^3 / 8 assembled fits 288 GiB
+    preflight_slab(2048, 2048, 2048, dict(STENCIL_7PT_3D), 0, 8, False, total)
+    # 2048^3 / 4 assembled (~314 GiB) must be rejected with a clear error
+    with pytest.raises(MemoryError):
+        preflight_slab(2048, 2048, 2048, dict(STENCIL_7PT_3D), 0, 4, False,
+                       total)
+    # ... but fits matrix-free
+    preflight_slab(2048, 2048, 2048, dict(STENCIL_7PT_3D), 0, 4, True, total)
