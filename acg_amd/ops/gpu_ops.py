@@ -367,23 +367,34 @@ def pipelined_fused(z, t, p, x, r, w, q, scal: torch.Tensor,
                       nt_update, _stream())
 
 
+BAR_STATE_WORDS = 400  # shared with kernels.hip (flat + hierarchical v3)
+
+
 def cg_device(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
               nrows: int, b: torch.Tensor, x: torch.Tensor, r: torch.Tensor,
               p: torch.Tensor, t: torch.Tensor, scal: torch.Tensor,
               partials: torch.Tensor, out2: torch.Tensor,
               barrier_state: torch.Tensor, maxits: int,
-              res_atol: float, res_rtol: float) -> int:
+              res_atol: float, res_rtol: float,
+              hier: bool | None = None) -> int:
     """Monolithic device-side CG: one cooperative launch runs the whole
-    solve.  ``barrier_state``: 3 zeroed uint32 words (grid-barrier counter,
-    generation, fail flag -- must be zeroed before EVERY launch).
-    Returns the grid size used."""
+    solve.  ``barrier_state``: BAR_STATE_WORDS zeroed uint32 words (must
+    be zeroed before EVERY launch).  ``hier`` selects the hierarchical
+    (per-XCD then global) grid barrier; default = env ACG_DEVCG_HIER
+    (on).  Returns the grid size used."""
     nslices = sellptr.numel() - 1
+    assert barrier_state.numel() >= BAR_STATE_WORDS
+    if hier is None:
+        import os
+
+        hier = os.environ.get("ACG_DEVCG_HIER", "1") != "0"
     return K.cg_device(nslices, nrows, sellptr.data_ptr(), cols.data_ptr(),
                        1 if cols.dtype == torch.int64 else 0, vals.data_ptr(),
                        b.data_ptr(), x.data_ptr(), r.data_ptr(), p.data_ptr(),
                        t.data_ptr(), scal.data_ptr(), partials.data_ptr(),
                        out2.data_ptr(), barrier_state.data_ptr(),
-                       maxits, res_atol, res_rtol, _stream())
+                       maxits, res_atol, res_rtol, _stream(),
+                       1 if hier else 0)
 
 
 def pack_gather(sendbuf: torch.Tensor, x: torch.Tensor, idx: torch.Tensor) -> None:

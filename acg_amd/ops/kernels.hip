@@ -1354,10 +1354,78 @@ typedef __attribute__((address_space(1))) unsigned int gu32;
 // [16+16*x]=generation copy for XCD x (64 B stride: distinct lines).
 #define BAR_GEN0 16
 #define BAR_GENSTRIDE 16
+// hierarchical-barrier extension words (v3): per-XCD arrival counters on
+// distinct cachelines, one set per parity: [144+16x] even, [272+16x] odd.
+#define BAR_XCNT0 144
+#define BAR_XCNT_PAR 128
+#define BAR_STATE_WORDS 400
 
 __device__ __forceinline__ void st_sc1_f64(double* p, double v) {
     __hip_atomic_store(reinterpret_cast<unsigned long long*>(p),
                        (unsigned long long)__double_as_longlong(v), RLX_AGENT);
+}
+
+// Hierarchical barrier (v3): arrivals first bump a PER-XCD counter line
+// (blocks land on XCD b%8, so the ~128 RMWs per line proceed on 8 lines
+// in parallel instead of 1024 serialised on one line -- the ~35 us cost
+// of the flat barrier at 1024 blocks); each XCD's last arriver bumps the
+// global counter (<= 8 RMWs); the global last resets BOTH levels of the
+// current parity and bumps every generation copy.  Parity safety is the
+// same argument as the flat barrier: the words being reset belong to the
+// phase that no block will touch again until two barriers later.
+__device__ __forceinline__ bool grid_barrier_hier(unsigned* state) {
+    gu32* fail = (gu32*)(state + 1);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    bool ok = true;
+    if (threadIdx.x == 0) {
+        const int myxcd = (int)(blockIdx.x & 7);
+        const unsigned nb = gridDim.x;
+        const unsigned q = nb >> 3, rmd = nb & 7u;
+        const unsigned blocks_on_mine = q + ((unsigned)myxcd < rmd ? 1u : 0u);
+        const unsigned nxcd = nb < 8u ? nb : 8u;
+        gu32* mygen = (gu32*)(state + BAR_GEN0 + BAR_GENSTRIDE * myxcd);
+        const unsigned g = __hip_atomic_load(mygen, RLX_AGENT);
+        const unsigned par = g & 1u;
+        gu32* xcnt = (gu32*)(state + BAR_XCNT0 + BAR_XCNT_PAR * par
+                             + BAR_GENSTRIDE * myxcd);
+        const unsigned ax = __hip_atomic_fetch_add(xcnt, 1u, RLX_AGENT) + 1u;
+        bool releaser = false;
+        if (ax == blocks_on_mine) {
+            gu32* gcnt = (gu32*)(state + (par ? 8 : 0));
+            const unsigned ag = __hip_atomic_fetch_add(gcnt, 1u, RLX_AGENT) + 1u;
+            if (ag == nxcd) {
+                __hip_atomic_store(gcnt, 0u, RLX_AGENT);
+                #pragma unroll
+                for (int xx = 0; xx < 8; ++xx)
+                    __hip_atomic_store(
+                        (gu32*)(state + BAR_XCNT0 + BAR_XCNT_PAR * par
+                                + BAR_GENSTRIDE * xx), 0u, RLX_AGENT);
+                #pragma unroll
+                for (int xx = 0; xx < 8; ++xx)
+                    __hip_atomic_store(
+                        (gu32*)(state + BAR_GEN0 + BAR_GENSTRIDE * xx),
+                        g + 1u, RLX_AGENT);
+                releaser = true;
+            }
+        }
+        if (!releaser) {
+            unsigned spins = 0;
+            while (__hip_atomic_load(mygen, RLX_AGENT) == g) {
+                if (spins < 32) __builtin_amdgcn_s_sleep(2);
+                else __builtin_amdgcn_s_sleep(64);
+                if (++spins > 4000000u) {
+                    __hip_atomic_store(fail, 1u, RLX_AGENT);
+                    ok = false;
+                    break;
+                }
+                if (__hip_atomic_load(fail, RLX_AGENT)) { ok = false; break; }
+            }
+        }
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    }
+    __syncthreads();
+    return ok;
 }
 
 __device__ __forceinline__ bool grid_barrier(unsigned* state) {
@@ -1413,7 +1481,7 @@ k_cg_device(long nslices, long nrows,
             double* __restrict__ t,
             double* __restrict__ scal, double* __restrict__ partials,
             int* __restrict__ out2, unsigned* __restrict__ barrier_state,
-            int maxits, double res_atol, double res_rtol) {
+            int maxits, double res_atol, double res_rtol, int hier) {
     const long tid = (long)blockIdx.x * BLOCK + threadIdx.x;
     const long nth = (long)gridDim.x * BLOCK;
     const int lane = threadIdx.x & (WAVE - 1);
@@ -1421,7 +1489,8 @@ k_cg_device(long nslices, long nrows,
     const long nw = nth >> 6;
     bool alive = true;
     auto gsync = [&]() {
-        if (alive && !grid_barrier(barrier_state)) alive = false;
+        if (alive && !(hier ? grid_barrier_hier(barrier_state)
+                            : grid_barrier(barrier_state))) alive = false;
         return alive;
     };
 
@@ -1816,7 +1885,7 @@ int cg_device(long nslices, long nrows, uintptr_t sellptr, uintptr_t cols,
               uintptr_t p, uintptr_t t, uintptr_t scal, uintptr_t partials,
               uintptr_t out2, uintptr_t barrier_state,
               int maxits, double res_atol, double res_rtol,
-              uintptr_t stream) {
+              uintptr_t stream, int hier) {
     int dev = 0;
     hipGetDevice(&dev);
     hipDeviceProp_t props;
@@ -1843,7 +1912,7 @@ int cg_device(long nslices, long nrows, uintptr_t sellptr, uintptr_t cols,
                         (void*)&vals, (void*)&b, (void*)&x, (void*)&r,
                         (void*)&p, (void*)&t, (void*)&scal, (void*)&partials,
                         (void*)&out2, (void*)&barrier_state,
-                        &maxits, &res_atol, &res_rtol};
+                        &maxits, &res_atol, &res_rtol, &hier};
         hipError_t e = hipLaunchCooperativeKernel(
             kern, dim3((unsigned)grid), dim3(BLOCK), args, 0, S(stream));
         if (e == hipSuccess) break;

@@ -630,7 +630,7 @@ class CGSolverHIP:
         t = self._vec()
         p = self._vec(nghost=True)
         out2 = torch.zeros(2, dtype=torch.int32, device=self.device)
-        barrier_state = torch.zeros(16 + 16 * 8, dtype=torch.int32,
+        barrier_state = torch.zeros(ops.BAR_STATE_WORDS, dtype=torch.int32,
                                     device=self.device)
         sellptr, scols, svals = self.sell
         torch.cuda.synchronize(self.device)
