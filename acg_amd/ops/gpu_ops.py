@@ -381,20 +381,23 @@ def cg_device(sellptr: torch.Tensor, cols: torch.Tensor, vals: torch.Tensor,
     solve.  ``barrier_state``: BAR_STATE_WORDS zeroed uint32 words (must
     be zeroed before EVERY launch).  ``hier`` selects the hierarchical
     (per-XCD then global) grid barrier; default = env ACG_DEVCG_HIER
-    (on).  Returns the grid size used."""
+    ("auto": hierarchical from 64 blocks up -- measured crossover,
+    tools/devcg_barrier_ab.py).  Returns the grid size used."""
     nslices = sellptr.numel() - 1
     assert barrier_state.numel() >= BAR_STATE_WORDS
     if hier is None:
         import os
 
-        hier = os.environ.get("ACG_DEVCG_HIER", "1") != "0"
+        v = os.environ.get("ACG_DEVCG_HIER", "auto")
+        ih = -1 if v == "auto" else (0 if v == "0" else 1)
+    else:
+        ih = 1 if hier else 0
     return K.cg_device(nslices, nrows, sellptr.data_ptr(), cols.data_ptr(),
                        1 if cols.dtype == torch.int64 else 0, vals.data_ptr(),
                        b.data_ptr(), x.data_ptr(), r.data_ptr(), p.data_ptr(),
                        t.data_ptr(), scal.data_ptr(), partials.data_ptr(),
                        out2.data_ptr(), barrier_state.data_ptr(),
-                       maxits, res_atol, res_rtol, _stream(),
-                       1 if hier else 0)
+                       maxits, res_atol, res_rtol, _stream(), ih)
 
 
 def pack_gather(sendbuf: torch.Tensor, x: torch.Tensor, idx: torch.Tensor) -> None:

@@ -1905,6 +1905,10 @@ int cg_device(long nslices, long nrows, uintptr_t sellptr, uintptr_t cols,
     long need = (nslices * WAVE + BLOCK - 1) / BLOCK;
     if (grid > need) grid = need;
     if (grid > MAXG) grid = MAXG;
+    // hier < 0 = auto: per-XCD staging pays beyond ~64 blocks (measured
+    // MI355X: 1024 blocks flat 192 us/it vs hier 76; 16 blocks flat
+    // 15.6 vs 18.7 -- tools/devcg_barrier_ab.py)
+    if (hier < 0) hier = grid >= 64 ? 1 : 0;
     // the occupancy API can over-report by one block/CU (guide §1); the
     // cooperative launch checks residency -- shrink and retry on rejection.
     for (;;) {
