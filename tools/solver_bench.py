@@ -60,7 +60,7 @@ def main():
             b, x.clone(), maxits=args.steps, res_rtol=0.0, megafuse=False)
         arms["pipelined-mega"] = lambda: solver.solve_pipelined(
             b, x.clone(), maxits=args.steps, res_rtol=0.0, megafuse=True)
-    if S.nowned <= 200_000 and solver.sell is not None:
+    if S.nowned <= 1_100_000 and solver.sell is not None:
         arms["device"] = lambda: solver.solve_device(b, x.clone(),
                                                      maxits=args.steps,
                                                      res_rtol=0.0)
