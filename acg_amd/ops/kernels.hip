@@ -1717,18 +1717,25 @@ void spmv_sell(long nslices, long nrows, long rowbase, uintptr_t sellptr,
     const bool fuse = partials != 0 && dotslot >= 0;
     dim3 g((unsigned)blocks), b(BLOCK);
     if (perm != 0) {
-        // sigma-sorted SELL (irregular rows): fixed NT, no swizzle, unroll 4
-        #define LAUNCH_PERM(CT, AC, FD) \
-            hipLaunchKernelGGL((k_spmv_sell<CT, AC, FD, true, false, 4, true>), \
+        // sigma-sorted SELL (irregular rows): fixed NT, no swizzle;
+        // UNROLL honours the U8 variant bit (deeper unroll = more
+        // outstanding x gathers to hide the random-access latency that
+        // dominates this path)
+        #define LAUNCH_PERM_U(CT, AC, FD, U) \
+            hipLaunchKernelGGL((k_spmv_sell<CT, AC, FD, true, false, U, true>), \
                 g, b, 0, S(stream), nslices, nrows, rowbase, \
                 (const long*)sellptr, (const CT*)cols, (const double*)vals, \
                 (const double*)x, (double*)y, (double*)partials, (const int*)perm)
+        #define LAUNCH_PERM(CT, AC, FD) \
+            if (variant & SELL_U8) { LAUNCH_PERM_U(CT, AC, FD, 8); } \
+            else { LAUNCH_PERM_U(CT, AC, FD, 4); }
         #define DISPP(CT) \
             if (accum) { if (fuse) { LAUNCH_PERM(CT, true, true); } else { LAUNCH_PERM(CT, true, false); } } \
             else       { if (fuse) { LAUNCH_PERM(CT, false, true); } else { LAUNCH_PERM(CT, false, false); } }
         if (col64) { DISPP(long) } else { DISPP(int) }
         #undef DISPP
         #undef LAUNCH_PERM
+        #undef LAUNCH_PERM_U
         check_hip("spmv_sell_perm");
         if (fuse)
             reduce_partials(partials, (int)blocks, scal, dotslot, dot_accum, stream);
