@@ -177,6 +177,11 @@ class CGSolverHIP:
                 "rowlist": up(rowlist) if len(rowlist) else None,
                 "bins": bins,
             }
+            if L.nnzO > 0:
+                # matO rows inherit the same power-law tail: bin them too
+                rlO, binsO = ops.build_row_bins(L.O_rowptr)
+                self.hybrid["rowlistO"] = up(rlO)
+                self.hybrid["binsO"] = binsO
 
         def mk_binned():
             rowlist, bins = ops.build_row_bins(L.A_rowptr)
@@ -330,6 +335,11 @@ class CGSolverHIP:
                     optr, ocols, ovals = self.sellO
                     ops.spmv_sell(optr, ocols, ovals, L.nborder, xfull, y,
                                   rowbase=L.ninterior, accum=True, **fuse)
+                elif self.hybrid is not None and "rowlistO" in self.hybrid:
+                    ops.spmv_binned(self.O_rowptr, self.O_colidx, self.O_vals,
+                                    self.hybrid["rowlistO"],
+                                    self.hybrid["binsO"], xfull, y,
+                                    rowbase=L.ninterior, accum=True, **fuse)
                 else:
                     ops.spmv(self.O_rowptr, self.O_colidx, self.O_vals, xfull, y,
                              rowbase=L.ninterior, lanes=self.lanesO, accum=True,
