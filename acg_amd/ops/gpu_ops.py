@@ -113,7 +113,8 @@ def build_row_bins(rowptr, max_lanes: int = 64):
     return order, bins
 
 
-def build_sellcsr_hybrid(rowptr, colidx, vals, cut: int = 96):
+def build_sellcsr_hybrid(rowptr, colidx, vals, cut: int = 96,
+                         window: int = 0):
     """Host prep for the SELL+CSR hybrid: short rows (len <= cut) as
     sigma-sorted SELL, long rows as a longest-first 64/32-lane binned CSR
     list.  Returns (sellptr, cols, svals, perm, rowlist_long, bins) numpy
@@ -123,7 +124,7 @@ def build_sellcsr_hybrid(rowptr, colidx, vals, cut: int = 96):
     from .torch_ref import sellcsr_split
 
     sellptr, cols, svals, perm, long_rows, nshort = sellcsr_split(
-        rowptr, colidx, vals, cut=cut)
+        rowptr, colidx, vals, cut=cut, window=window)
     lens = np.diff(np.asarray(rowptr))
     longlens = lens[long_rows.astype(np.int64)]  # descending
     bins = []

@@ -108,17 +108,24 @@ def sell_from_csr(rowptr, colidx, vals, C: int = 64, sigma: int = 1):
     return sellptr, cols, svals
 
 
-def sellcsr_split(rowptr, colidx, vals, cut: int = 96, C: int = 64):
+def sellcsr_split(rowptr, colidx, vals, cut: int = 96, C: int = 64,
+                  window: int = 0):
     """Split a CSR matrix by row length for the SELL+CSR hybrid format.
 
     Rows with len <= ``cut`` (the regular majority) go into a sigma-SELL
-    structure built in globally length-descending order (minimal padding,
-    longest slices scheduled first); rows with len > cut (the power-law
-    tail) stay CSR and are listed longest-first for the 64-lane vector
-    kernel.  MEASURED motive (MI355X, 1M-row power-law): the pure binned
-    hybrid spends 241 us/it in its 4-lane short-row bin -- 4-lane groups
-    issue 16 scattered row streams per wave, while SELL's 64-lane slices
-    load one contiguous 512 B line set per step.
+    structure; rows with len > cut (the power-law tail) stay CSR and are
+    listed longest-first for the 64-lane vector kernel.  MEASURED motive
+    (MI355X, 1M-row power-law): the pure binned hybrid spends 241 us/it
+    in its 4-lane short-row bin -- 4-lane groups issue 16 scattered row
+    streams per wave, while SELL's 64-lane slices load one contiguous
+    512 B line set per step.
+
+    ``window``: 0 = short rows globally length-descending (minimal
+    padding, longest slices first).  W > 0 = sort by length only WITHIN
+    windows of W consecutive short rows (original index order preserved
+    across windows): slightly more padding, but the x gather keeps the
+    source ordering's column locality -- the lever when the gather, not
+    the vals/cols stream, bounds the SELL part.
 
     Returns (sellptr i64, cols, svals f64, perm int32 [sentinel n],
     rowlist_long int32 desc, nshort).  SELL part empty => sellptr len 1.
@@ -135,6 +142,14 @@ def sellcsr_split(rowptr, colidx, vals, cut: int = 96, C: int = 64):
     nlong = int(np.searchsorted(-slens, -int(cut) - 1, side="right"))
     long_rows = order[:nlong].astype(np.int32)
     short_rows = order[nlong:].astype(np.int64)  # descending lengths
+    if window and len(short_rows):
+        sr = np.sort(short_rows)  # original index order
+        out = np.empty_like(sr)
+        for w0 in range(0, len(sr), window):
+            w1 = min(w0 + window, len(sr))
+            sub = np.argsort(-lens[sr[w0:w1]], kind="stable")
+            out[w0:w1] = sr[w0:w1][sub]
+        short_rows = out
     nshort = len(short_rows)
     nslices = (nshort + C - 1) // C
     # SELL arrays over the short subset, slice lengths from the sorted order

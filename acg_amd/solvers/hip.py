@@ -167,8 +167,9 @@ class CGSolverHIP:
             ACG_HYBRID_CUT overrides the split length (tuning sweeps)."""
             if cut is None:
                 cut = int(os.environ.get("ACG_HYBRID_CUT", "96"))
+            window = int(os.environ.get("ACG_HYBRID_WINDOW", "0"))
             sp_, cols, svals, perm, rowlist, bins = ops.build_sellcsr_hybrid(
-                L.A_rowptr, L.A_colidx, L.A_vals, cut=cut)
+                L.A_rowptr, L.A_colidx, L.A_vals, cut=cut, window=window)
             self.hybrid = {
                 "sellptr": up(sp_) if sp_ is not None else None,
                 "cols": up(cols) if sp_ is not None else None,
