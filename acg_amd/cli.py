@@ -69,7 +69,10 @@ def make_parser() -> argparse.ArgumentParser:
                    help="b := A x* for random x*; report error norms")
     p.add_argument("--numfmt", default=None, help="printf format for output values")
     p.add_argument("--output-comm-matrix", action="store_true",
-                   help="write the rank x rank halo send-count matrix to stderr")
+                   help="print the rank x rank halo send-count matrix to "
+                        "standard output (reference acg-hip.c:1683-1742)")
+    p.add_argument("--no-output-comm-matrix", dest="output_comm_matrix",
+                   action="store_false", help=argparse.SUPPRESS)
     p.add_argument("--profile", action="store_true",
                    help="per-op hipEvent timing (reference ACG_ENABLE_PROFILING)")
     p.add_argument("-q", "--quiet", action="store_true",
@@ -236,12 +239,13 @@ def main(argv=None) -> int:
         counts = comm.gather_object(
             {int(q): int(c) for q, c in zip(S.halo.recipients, S.halo.sendcounts)})
         if rank == 0:
-            print("%%MatrixMarket matrix coordinate integer general", file=sys.stderr)
+            # to STDOUT like the reference (before the solution vector)
+            print("%%MatrixMarket matrix coordinate integer general")
             entries = [(p, q, c) for p, row in enumerate(counts)
                        for q, c in row.items()]
-            print(f"{nparts} {nparts} {len(entries)}", file=sys.stderr)
+            print(f"{nparts} {nparts} {len(entries)}")
             for p, q, c in entries:
-                print(f"{p + 1} {q + 1} {c}", file=sys.stderr)
+                print(f"{p + 1} {q + 1} {c}")
 
     import torch as _t
 

@@ -379,6 +379,93 @@ py::array_t<i64> hem_match(py::array_t<i64, py::array::c_style | py::array::forc
     return match_a;
 }
 
+// ---------------------------------------------------------------------------
+// Graph contraction for multilevel partitioning: given a (symmetric,
+// both-triangle) fine CSR adjacency and a fine->coarse vertex map, build
+// the coarse CSR with parallel-edge weights summed and self-loops
+// dropped.  Same machinery as coo_to_sym_csr (atomic counting scatter +
+// per-row sort + dedup + compact), OpenMP-parallel -- the numpy
+// np.unique route costs a full 80M-key sort per level.
+py::tuple contract_graph(py::array_t<i64, py::array::c_style | py::array::forcecast> rowptr,
+                         py::array_t<i64, py::array::c_style | py::array::forcecast> cols,
+                         py::array_t<double, py::array::c_style | py::array::forcecast> wts,
+                         py::array_t<i64, py::array::c_style | py::array::forcecast> cmap,
+                         i64 nc) {
+    const i64* rp = rowptr.data();
+    const i64* ci = cols.data();
+    const double* w = wts.data();
+    const i64* cm = cmap.data();
+    const i64 n = (i64)rowptr.shape(0) - 1;
+
+    std::vector<std::atomic<i64>> cnt(nc);
+    for (i64 i = 0; i < nc; ++i) cnt[i].store(0, std::memory_order_relaxed);
+    #pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < n; ++i) {
+        const i64 ciu = cm[i];
+        i64 local = 0;
+        for (i64 k = rp[i]; k < rp[i + 1]; ++k)
+            if (cm[ci[k]] != ciu) ++local;
+        if (local) cnt[ciu].fetch_add(local, std::memory_order_relaxed);
+    }
+    std::vector<i64> rowc(nc + 1);
+    rowc[0] = 0;
+    for (i64 i = 0; i < nc; ++i)
+        rowc[i + 1] = rowc[i] + cnt[i].load(std::memory_order_relaxed);
+    const i64 nnz_c = rowc[nc];
+    std::vector<i64> tc(nnz_c);
+    std::vector<double> tw(nnz_c);
+    for (i64 i = 0; i < nc; ++i) cnt[i].store(0, std::memory_order_relaxed);
+    #pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < n; ++i) {
+        const i64 ciu = cm[i];
+        for (i64 k = rp[i]; k < rp[i + 1]; ++k) {
+            const i64 cj = cm[ci[k]];
+            if (cj == ciu) continue;
+            const i64 pos = rowc[ciu] + cnt[ciu].fetch_add(1, std::memory_order_relaxed);
+            tc[pos] = cj;
+            tw[pos] = w[k];
+        }
+    }
+    std::vector<i64> outcnt(nc);
+    #pragma omp parallel
+    {
+        std::vector<std::pair<i64, double>> buf;
+        #pragma omp for schedule(dynamic, 1024)
+        for (i64 i = 0; i < nc; ++i) {
+            const i64 b = rowc[i], e = rowc[i + 1];
+            const i64 len = e - b;
+            buf.resize(len);
+            for (i64 k = 0; k < len; ++k) buf[k] = {tc[b + k], tw[b + k]};
+            std::sort(buf.begin(), buf.end(),
+                      [](const auto& x, const auto& y) { return x.first < y.first; });
+            i64 m = 0;
+            for (i64 k = 0; k < len; ++k) {
+                if (m > 0 && buf[m - 1].first == buf[k].first)
+                    buf[m - 1].second += buf[k].second;
+                else
+                    buf[m++] = buf[k];
+            }
+            for (i64 k = 0; k < m; ++k) { tc[b + k] = buf[k].first; tw[b + k] = buf[k].second; }
+            outcnt[i] = m;
+        }
+    }
+    py::array_t<i64> rowptr_c(nc + 1);
+    i64* rpo = rowptr_c.mutable_data();
+    rpo[0] = 0;
+    for (i64 i = 0; i < nc; ++i) rpo[i + 1] = rpo[i] + outcnt[i];
+    const i64 out_nnz = rpo[nc];
+    py::array_t<i64> cols_c(out_nnz);
+    py::array_t<double> w_c(out_nnz);
+    i64* co = cols_c.mutable_data();
+    double* wo = w_c.mutable_data();
+    #pragma omp parallel for schedule(static)
+    for (i64 i = 0; i < nc; ++i) {
+        std::memcpy(co + rpo[i], tc.data() + rowc[i], outcnt[i] * sizeof(i64));
+        std::memcpy(wo + rpo[i], tw.data() + rowc[i], outcnt[i] * sizeof(double));
+    }
+    return py::make_tuple(rowptr_c, cols_c, w_c);
+}
+
 // in-place scans (reference acgprefixsum_inplace_*, prefixsum.h:72-116)
 py::array_t<i64> prefix_sum(py::array_t<i64, py::array::c_style | py::array::forcecast> a,
                             bool inclusive) {
@@ -412,5 +499,6 @@ PYBIND11_MODULE(_acg_host, m) {
     m.def("prefix_sum", &prefix_sum, py::arg("a"), py::arg("inclusive") = true);
     m.def("bsell_blocks", &bsell_blocks);
     m.def("hem_match", &hem_match);
+    m.def("contract_graph", &contract_graph);
     m.def("num_threads", &num_threads);
 }
