@@ -195,12 +195,15 @@ def main(argv=None) -> int:
                 xsol /= np.linalg.norm(xsol)
                 b_global = A.dsymv(xsol)
             elif args.b:
-                mb = read_mtx(args.b, gzipped=args.gzip)
+                # --binary applies to b/x0 too (reference acg-hip.c:1796)
+                mb = read_mtx(args.b, gzipped=args.gzip, binary=args.binary,
+                              idxsize=args.idxsize)
                 b_global = _vector_from_mtx(mb, A.n, "b")
             else:
                 b_global = np.ones(A.n, dtype=np.float64)
             if args.x0:
-                mx = read_mtx(args.x0, gzipped=args.gzip)
+                mx = read_mtx(args.x0, gzipped=args.gzip, binary=args.binary,
+                              idxsize=args.idxsize)
                 x0_global = _vector_from_mtx(mx, A.n, "x0")
     except Exception as e:  # collective error agreement (acgerrmpi)
         err = e

@@ -223,3 +223,33 @@ def test_preflight_rejects_oversized_grid():
                        total)
     # ... but fits matrix-free
     preflight_slab(2048, 2048, 2048, dict(STENCIL_7PT_3D), 0, 4, True, total)
+
+
+def test_cli_binary_rhs(tmp_path, capsys, monkeypatch):
+    """--binary applies to b too (reference acg-hip.c:1796)."""
+    from acg_amd import cli
+    from acg_amd.io.mtx import MtxFile, write_mtx
+
+    A, apath = _poisson_mtx(tmp_path)
+    bpath = tmp_path / "b.bin"
+    rng = np.random.default_rng(4)
+    bvals = rng.standard_normal(A.n)
+    mb = MtxFile(object="matrix", format="array", field_="real",
+                 symmetry="general", nrows=A.n, ncols=1, nnz=A.n, a=bvals)
+    write_mtx(bpath, mb, binary=True, idxsize=64)
+    # matrix stays text: write a binary A too so one flag covers both
+    abin = tmp_path / "A.bin"
+    rows = np.repeat(np.arange(A.n), np.diff(A.rowptr))
+    ma = MtxFile(object="matrix", format="coordinate", field_="real",
+                 symmetry="symmetric", nrows=A.n, ncols=A.n,
+                 nnz=A.nnz_stored, rowidx=rows, colidx=A.colidx, a=A.vals)
+    write_mtx(abin, ma, binary=True, idxsize=64)
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    monkeypatch.delenv("RANK", raising=False)
+    rc = cli.main([str(abin), str(bpath), "--binary", "--solver", "cpu",
+                   "--max-iterations", "3000", "--residual-rtol", "1e-11"])
+    out = capsys.readouterr()
+    assert rc == 0, out.err
+    got = np.array([float(v) for v in out.out.strip().splitlines()[2:]])
+    xs = np.linalg.solve(A.to_scipy_full().toarray(), bvals)
+    assert np.allclose(got, xs, atol=1e-7)
