@@ -113,7 +113,7 @@ def build_row_bins(rowptr, max_lanes: int = 64):
     return order, bins
 
 
-def build_sellcsr_hybrid(rowptr, colidx, vals, cut: int = 96,
+def build_sellcsr_hybrid(rowptr, colidx, vals, cut: int = 192,
                          window: int = 0):
     """Host prep for the SELL+CSR hybrid: short rows (len <= cut) as
     sigma-sorted SELL, long rows as a longest-first 64/32-lane binned CSR

@@ -108,7 +108,7 @@ def sell_from_csr(rowptr, colidx, vals, C: int = 64, sigma: int = 1):
     return sellptr, cols, svals
 
 
-def sellcsr_split(rowptr, colidx, vals, cut: int = 96, C: int = 64,
+def sellcsr_split(rowptr, colidx, vals, cut: int = 192, C: int = 64,
                   window: int = 0):
     """Split a CSR matrix by row length for the SELL+CSR hybrid format.
 

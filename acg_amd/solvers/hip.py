@@ -166,7 +166,7 @@ class CGSolverHIP:
             Costs ~1.75x operator memory (CSR kept for the long rows).
             ACG_HYBRID_CUT overrides the split length (tuning sweeps)."""
             if cut is None:
-                cut = int(os.environ.get("ACG_HYBRID_CUT", "96"))
+                cut = int(os.environ.get("ACG_HYBRID_CUT", "192"))
             window = int(os.environ.get("ACG_HYBRID_WINDOW", "0"))
             sp_, cols, svals, perm, rowlist, bins = ops.build_sellcsr_hybrid(
                 L.A_rowptr, L.A_colidx, L.A_vals, cut=cut, window=window)
