@@ -172,7 +172,7 @@ def _csr_arrays(G):
 
 def _refine_kway(u: np.ndarray, v: np.ndarray, w: np.ndarray, n: int,
                  vwts: np.ndarray, part: np.ndarray, nparts: int,
-                 passes: int = 3, eps: float = 0.05) -> np.ndarray:
+                 passes: int = 10, eps: float = 0.05) -> np.ndarray:
     """Greedy k-way boundary refinement (the FM-style refinement stage of
     the multilevel scheme, reference METIS refinement inside
     METIS_PartGraphRecursive) on raw edge arrays (u, v, w).
@@ -188,7 +188,14 @@ def _refine_kway(u: np.ndarray, v: np.ndarray, w: np.ndarray, n: int,
     lo = (1.0 - eps) * target
     part = part.astype(np.int32).copy()
     rng_n = np.arange(n)
+    cut_prev = None
     for _ in range(passes):
+        # adaptive stop: passes beyond convergence cost a full-edge sweep
+        # each for ~nothing (banded 2M: 7 extra passes bought 0.06% once)
+        cut_now = float(w[part[u] != part[v]].sum())
+        if cut_prev is not None and cut_prev - cut_now < 2e-3 * max(cut_prev, 1.0):
+            break
+        cut_prev = cut_now
         idx = u * nparts + part[v]
         W = np.bincount(idx, weights=w, minlength=n * nparts) \
             .reshape(n, nparts)

@@ -221,5 +221,6 @@ def test_bench_2proc_end_to_end(tmp_path):
     line = next(l for l in r.stdout.splitlines() if l.startswith("{"))
     d = json.loads(line)
     assert d["n_gpus"] == 2 and d["steps"] == 5
-    assert d["value"] > 0 and d["config"]["solver"] == "cg-pipelined"
+    assert d["value"] > 0
+    assert d["config"]["solver"] in ("cg-pipelined", "cg-classic")  # auto probes
     assert d["config"]["rows"] == 3 * 14 ** 3
