@@ -1589,6 +1589,25 @@ k_cg_device(long nslices, long nrows,
     (void)bnrm2sqr;
 }
 
+// fp64 atomic-scatter throughput probe (design study for a
+// symmetric-storage SpMV: the transpose half's y updates would be HW
+// atomic adds; whether that can beat the full-storage roofline depends
+// entirely on sustained global_atomic_add_f64 throughput at realistic
+// target distributions).  MODE 0: atomic add; 1: plain store (upper
+// bound); 2: gather-read (calibration).
+__global__ void __launch_bounds__(BLOCK)
+k_atomic_probe(double* __restrict__ y, const int* __restrict__ idx,
+               const double* __restrict__ v, long nnz, int mode) {
+    const long stride = (long)gridDim.x * BLOCK;
+    double acc = 0.0;
+    for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < nnz; i += stride) {
+        if (mode == 0) unsafeAtomicAdd(&y[idx[i]], v[i]);
+        else if (mode == 1) y[idx[i]] = v[i];
+        else acc += y[idx[i]] * v[i];
+    }
+    if (mode == 2 && acc == -1.0) y[0] = acc;  // keep the reads live
+}
+
 // halo pack: sendbuf[i] = x[sendidx[i]]
 // (reference acghalo_pack_hip_double, halo-kernels-hip.hip:48-103; no unpack
 // kernel exists -- ghosts are received in place, see dist/halo.py)
@@ -1938,6 +1957,15 @@ int cg_device(long nslices, long nrows, uintptr_t sellptr, uintptr_t cols,
     return (int)grid;
 }
 
+void atomic_probe(uintptr_t y, uintptr_t idx, uintptr_t v, long nnz,
+                  int mode, uintptr_t stream) {
+    const long blocks = elem_grid(nnz, 8);
+    hipLaunchKernelGGL(k_atomic_probe, dim3((unsigned)blocks), dim3(BLOCK), 0,
+                       S(stream), (double*)y, (const int*)idx,
+                       (const double*)v, nnz, mode);
+    check_hip("atomic_probe");
+}
+
 void pack_gather(uintptr_t sendbuf, uintptr_t x, uintptr_t idx, int idx64, long n,
                  uintptr_t stream) {
     if (n == 0) return;
@@ -1968,6 +1996,7 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.def("sell_pipe", &sell_pipe);
     m.def("pipelined_finalize", &pipelined_finalize);
     m.def("pack_gather", &pack_gather);
+    m.def("atomic_probe", &atomic_probe);
     m.def("cg_device", &cg_device);
     m.def("stencil_rowlen", &stencil_rowlen);
     m.def("stencil_fill", &stencil_fill);
