@@ -149,3 +149,37 @@ def test_sellcsr_split_structure():
     lr = rowlist.astype(np.int64)
     y[lr] = M[lr] @ x
     np.testing.assert_allclose(y, M @ x, rtol=1e-12)
+
+
+@pytest.mark.parametrize("seed,cut", [(2, 48), (5, 96), (9, 192)])
+def test_sellcsr_split_fuzz(seed, cut):
+    """Split reconstruction equals the plain SpMV across seeds and cuts."""
+    from acg_amd.ops.gpu_ops import build_sellcsr_hybrid
+
+    A = powerlaw_spd(4000, mean_nnz=26, seed=seed)
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    n = S.nowned
+    sellptr, cols, svals, perm, rowlist, bins = build_sellcsr_hybrid(
+        S.A_rowptr, S.A_colidx, S.A_vals, cut=cut)
+    import scipy.sparse as sp
+
+    M = sp.csr_matrix((S.A_vals, S.A_colidx.astype(np.int64), S.A_rowptr),
+                      shape=(n, n))
+    x = np.random.default_rng(0).standard_normal(n)
+    y = np.zeros(n)
+    if sellptr is not None:
+        C = 64
+        for s in range(len(sellptr) - 1):
+            base = int(sellptr[s])
+            L = (int(sellptr[s + 1]) - base) // C
+            blk = (svals[base:base + L * C].reshape(L, C)
+                   * x[cols[base:base + L * C].astype(np.int64)].reshape(L, C)
+                   ).sum(axis=0)
+            for lane in range(C):
+                r = int(perm[s * C + lane])
+                if r < n:
+                    y[r] = blk[lane]
+    lr = rowlist.astype(np.int64)
+    if len(lr):
+        y[lr] = M[lr] @ x
+    np.testing.assert_allclose(y, M @ x, rtol=1e-12)

@@ -112,3 +112,80 @@ def test_ml_end_to_end_solve():
     from acg_amd.dist.verify import _audit, halo_descriptor
 
     _audit([halo_descriptor(S) for S in systems])
+
+
+def test_contract_graph_native_matches_numpy():
+    """C++ contract_graph == the numpy unique-key contraction."""
+    H = pytest.importorskip("acg_amd.host._acg_host")
+    from acg_amd.part.partition import _csr_arrays, _full_adjacency_weighted
+
+    rng = np.random.default_rng(3)
+    A = powerlaw_spd(4000, mean_nnz=18, seed=3)
+    G = _full_adjacency_weighted(A)
+    rowptr, cols, w, u = _csr_arrays(G)
+    n = G.shape[0]
+    nc = n // 3
+    cmap = rng.integers(0, nc, n).astype(np.int64)
+    rp_c, c_c, w_c = (np.asarray(a) for a in
+                      H.contract_graph(rowptr, cols, w, cmap, nc))
+    # numpy oracle
+    cu, cv = cmap[u], cmap[cols]
+    keep = cu != cv
+    key = cu[keep] * nc + cv[keep]
+    uk, inv = np.unique(key, return_inverse=True)
+    wc_ref = np.bincount(inv, weights=w[keep])
+    u_ref = (uk // nc).astype(np.int64)
+    c_ref = (uk % nc).astype(np.int64)
+    rp_ref = np.searchsorted(u_ref, np.arange(nc + 1, dtype=np.int64))
+    np.testing.assert_array_equal(rp_c, rp_ref)
+    np.testing.assert_array_equal(c_c, c_ref)
+    np.testing.assert_allclose(w_c, wc_ref, rtol=1e-13)
+
+
+def test_refine_kway_preserves_partition_validity():
+    from acg_amd.part.partition import (_csr_arrays, _full_adjacency_weighted,
+                                        _refine_kway)
+
+    rng = np.random.default_rng(1)
+    A = powerlaw_spd(3000, mean_nnz=14, seed=2)
+    G = _full_adjacency_weighted(A)
+    rowptr, cols, w, u = _csr_arrays(G)
+    n = G.shape[0]
+    for k in (2, 5, 8):
+        part0 = rng.integers(0, k, n).astype(np.int32)
+        vw = np.ones(n)
+        part = _refine_kway(u, cols, w, n, vw, part0.copy(), k)
+        assert part.min() >= 0 and part.max() < k
+        # refinement never leaves the (1+eps) balance corridor it enforces
+        sizes = np.bincount(part, minlength=k)
+        # it can only start violating if part0 already did; with random
+        # init sizes are near-balanced, so the corridor holds loosely
+        assert sizes.max() <= 1.2 * n / k + 64
+
+
+def test_mtxpartition_tool_ml(tmp_path):
+    import subprocess
+    import sys as _sys
+    from pathlib import Path
+
+    from acg_amd.io.mtx import MtxFile, write_mtx
+
+    A = powerlaw_spd(1500, mean_nnz=12, seed=6)
+    rows = np.repeat(np.arange(A.n), np.diff(A.rowptr))
+    m = MtxFile(object="matrix", format="coordinate", field_="real",
+                symmetry="symmetric", nrows=A.n, ncols=A.n,
+                nnz=A.nnz_stored, rowidx=rows, colidx=A.colidx, a=A.vals)
+    path = tmp_path / "A.mtx"
+    write_mtx(path, m)
+    repo = Path(__file__).resolve().parent.parent
+    out = tmp_path / "part.mtx"
+    r = subprocess.run([_sys.executable, str(repo / "tools/mtxpartition.py"),
+                        str(path), "--parts", "4", "--method", "ml",
+                        "--seed", "1", "--output", str(out)],
+                       capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr
+    from acg_amd.part import read_partition_file
+
+    part = read_partition_file(out, A.n)
+    assert part.min() == 0 and part.max() == 3
+    assert len(np.unique(part)) == 4
