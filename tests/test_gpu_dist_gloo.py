@@ -224,3 +224,26 @@ def test_bench_2proc_end_to_end(tmp_path):
     assert d["value"] > 0
     assert d["config"]["solver"] in ("cg-pipelined", "cg-classic")  # auto probes
     assert d["config"]["rows"] == 3 * 14 ** 3
+
+
+@pytest.mark.gpu
+def test_run_poisson2048_script_smoke(tmp_path):
+    """The config-5 push-button script works end-to-end at a toy grid
+    (1 proc, GRID=64): torchrun launch, preflight assert, one JSON line."""
+    import json
+    import subprocess
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    env = dict(os.environ)
+    env.update(GRID="64", STEPS="3", WARMUP="1", NGPUS="1",
+               MASTER_PORT="29650")
+    r = subprocess.run(["bash", str(repo / "tools/run_poisson2048.sh")],
+                       capture_output=True, text=True, cwd=repo, env=env,
+                       timeout=420)
+    assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
+    line = next(l for l in r.stdout.splitlines() if l.startswith("{"))
+    d = json.loads(line)
+    assert d["config"]["model"] == "poisson3d-7pt-G64"
+    assert d["steps"] == 3 and d["value"] > 0
+    assert "preflight" in r.stderr
