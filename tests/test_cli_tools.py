@@ -1,7 +1,6 @@
 """CLI driver, IO formats, tools, oracle solver (CPU)."""
 
 import gzip
-import io
 import sys
 
 import numpy as np
@@ -9,7 +8,7 @@ import pytest
 import torch
 
 from acg_amd.gen import STENCIL_5PT_2D, stencil_global
-from acg_amd.io.mtx import MtxFile, read_mtx, vector_to_mtx, write_mtx
+from acg_amd.io.mtx import MtxFile, read_mtx, write_mtx
 from acg_amd.utils.numfmt import parse_numfmt
 
 

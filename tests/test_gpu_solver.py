@@ -542,3 +542,39 @@ def test_cli_gpu_irregular_file_hybrid(tmp_path, capsys, monkeypatch):
     vals = np.array([float(v) for v in out.out.strip().splitlines()[2:]])
     x_ref = spla.spsolve(A.to_scipy_full().tocsc(), np.ones(A.n))
     np.testing.assert_allclose(vals, x_ref, rtol=1e-6, atol=1e-8)
+
+
+def test_dist_graph_capture_replay_megafused_fakecomm():
+    """The MEGAFUSED pipelined distributed path (what config-5 poisson
+    runs at N=8: narrow rows => megafuse auto-on) with ping-pong graphs:
+    capture + replay must match the serial solve."""
+    import numpy as np
+    import torch
+
+    from acg_amd.gen import STENCIL_7PT_3D, stencil_global
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    A = stencil_global(16, 16, 16, STENCIL_7PT_3D)
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    rng = np.random.default_rng(0)
+    b = torch.from_numpy(rng.standard_normal(S.nowned)).cuda()
+
+    ser = CGSolverHIP(S, comm=None, device="cuda:0")
+    assert ser.megafuse_auto  # 7-pt rows are narrow: mega path engages
+    xs = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
+    rs = ser.solve_pipelined(b, xs, maxits=200, res_rtol=1e-10)
+    assert rs.converged
+
+    comm = _FakeCaptureComm(torch.device("cuda", 0))
+    dist = CGSolverHIP(S, comm=comm, device="cuda:0")
+    xd = torch.zeros_like(xs)
+    rd = dist.solve_pipelined(b, xd, maxits=200, res_rtol=1e-10)
+    assert rd.converged
+    assert not dist._graphs.get("pipelined_mega:capture_failed", False)
+    graphs = dist._graphs.get("pipelined_mega:dist")
+    assert graphs is not None and all(g is not None for g in graphs), \
+        "megafused distributed ping-pong graphs were not captured"
+    np.testing.assert_allclose(xd[:S.nowned].cpu().numpy(),
+                               xs[:S.nowned].cpu().numpy(),
+                               rtol=1e-9, atol=1e-11)

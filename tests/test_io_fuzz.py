@@ -93,7 +93,6 @@ def test_numfmt_grammar_fuzz():
     """Random printf specifiers over the reference grammar (fmtspec.c):
     every accepted spec must format like C/python %-formatting; malformed
     ones must be rejected, never crash."""
-    import itertools
 
     from acg_amd.utils.numfmt import FmtSpec, parse_numfmt
 

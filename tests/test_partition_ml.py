@@ -89,10 +89,8 @@ def test_hem_match_native_properties():
 def test_ml_end_to_end_solve():
     """Solver correctness on an ml partition (4 parts, serial extraction)."""
     import scipy.sparse.linalg as spla
-    import torch
 
     from acg_amd.part import extract_subdomains
-    from acg_amd.solvers.cpu import CGSolverCPU
 
     A = _permuted(powerlaw_spd(4_000, mean_nnz=14, locality=100, seed=4))
     part = partition_rows(A, 4, seed=1, method="ml")

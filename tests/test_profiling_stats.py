@@ -3,7 +3,6 @@
 import io
 
 import numpy as np
-import torch
 
 from acg_amd.gen import STENCIL_5PT_2D, stencil_global
 from acg_amd.part import (extract_subdomains, partition_rows,
