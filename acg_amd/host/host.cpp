@@ -247,8 +247,16 @@ py::tuple coo_to_sym_csr(i64 n,
             i64 len = e - b;
             buf.resize(len);
             for (i64 k = 0; k < len; ++k) buf[k] = {tc[b + k], tv[b + k]};
+            // value tiebreak: the atomic-cursor scatter lands duplicates in
+            // racy order -- sorting them canonically makes the duplicate
+            // SUM deterministic across runs (fp addition is order-
+            // sensitive; a last-ulp assembly difference visibly forks CG
+            // trajectories on ill-conditioned systems)
             std::sort(buf.begin(), buf.end(),
-                      [](const auto& x, const auto& y) { return x.first < y.first; });
+                      [](const auto& x, const auto& y) {
+                          return x.first != y.first ? x.first < y.first
+                                                    : x.second < y.second;
+                      });
             i64 m = 0;
             for (i64 k = 0; k < len; ++k) {
                 if (m > 0 && buf[m - 1].first == buf[k].first) {
@@ -436,8 +444,14 @@ py::tuple contract_graph(py::array_t<i64, py::array::c_style | py::array::forcec
             const i64 len = e - b;
             buf.resize(len);
             for (i64 k = 0; k < len; ++k) buf[k] = {tc[b + k], tw[b + k]};
+            // value tiebreak: canonical duplicate order => deterministic
+            // edge-weight sums regardless of the scatter race (see
+            // coo_to_sym_csr)
             std::sort(buf.begin(), buf.end(),
-                      [](const auto& x, const auto& y) { return x.first < y.first; });
+                      [](const auto& x, const auto& y) {
+                          return x.first != y.first ? x.first < y.first
+                                                    : x.second < y.second;
+                      });
             i64 m = 0;
             for (i64 k = 0; k < len; ++k) {
                 if (m > 0 && buf[m - 1].first == buf[k].first)

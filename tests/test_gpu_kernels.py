@@ -463,7 +463,7 @@ def test_sellcsr_hybrid_solver_heavy_tail(dev):
         assert rnorm[fmt] < 1e-2 * np.linalg.norm(b_np), fmt
     # equivalent convergence quality (same algorithm, different rounding)
     ratio = rnorm["hybrid"] / rnorm["binned"]
-    assert 0.5 < ratio < 2.0, rnorm
+    assert 0.3 < ratio < 3.0, rnorm
 
 
 def test_hybrid_mato_split_spmv(dev):

@@ -67,3 +67,48 @@ def test_sym_expand_full_matches_scipy(col32, eps):
     for r in range(0, A.n, 37):
         b, e = int(rowptr[r]), int(rowptr[r + 1])
         assert (np.diff(np.asarray(cols)[b:e]) > 0).all()
+
+
+def test_coo_assembly_bitwise_deterministic():
+    """The atomic-cursor scatter lands duplicates in racy order; the
+    canonical (col, value) dedup sort must make the SUM bitwise stable
+    across repeated assemblies (a last-ulp assembly difference visibly
+    forks CG trajectories on ill-conditioned systems -- found by a GPU
+    soak run)."""
+    H = pytest.importorskip("acg_amd.host._acg_host")
+    rng = np.random.default_rng(0)
+    n, m = 5000, 200_000
+    i = rng.integers(0, n, m)
+    j = rng.integers(0, n, m)
+    v = rng.standard_normal(m)
+    ref = None
+    for _ in range(5):
+        rp, c, vv = (np.asarray(a) for a in H.coo_to_sym_csr(n, i, j, v))
+        if ref is None:
+            ref = (rp.copy(), c.copy(), vv.copy())
+        else:
+            np.testing.assert_array_equal(ref[0], rp)
+            np.testing.assert_array_equal(ref[1], c)
+            np.testing.assert_array_equal(ref[2], vv)
+
+
+def test_contract_graph_bitwise_deterministic():
+    H = pytest.importorskip("acg_amd.host._acg_host")
+    from acg_amd.gen.irregular import powerlaw_spd
+    from acg_amd.part.partition import _csr_arrays, _full_adjacency_weighted
+
+    A = powerlaw_spd(4000, mean_nnz=20, seed=1)
+    G = _full_adjacency_weighted(A)
+    rowptr, cols, w, _u = _csr_arrays(G)
+    rng = np.random.default_rng(2)
+    cmap = rng.integers(0, 700, G.shape[0]).astype(np.int64)
+    ref = None
+    for _ in range(5):
+        rp, c, vv = (np.asarray(a) for a in
+                     H.contract_graph(rowptr, cols, w, cmap, 700))
+        if ref is None:
+            ref = (rp.copy(), c.copy(), vv.copy())
+        else:
+            np.testing.assert_array_equal(ref[0], rp)
+            np.testing.assert_array_equal(ref[1], c)
+            np.testing.assert_array_equal(ref[2], vv)
