@@ -206,7 +206,7 @@ class CGSolverHIP:
             return (int(sellptr[-1]) - L.nnzA) / max(L.nnzA, 1)
 
         def mk_bsell():
-            for dof_try in (3, 2):
+            for dof_try in (4, 3, 2):  # kernel dispatch covers 2/3/4
                 out = bsell_from_csr(L.A_rowptr, L.A_colidx, L.A_vals, dof_try)
                 if out is None:
                     continue

@@ -578,3 +578,35 @@ def test_dist_graph_capture_replay_megafused_fakecomm():
     np.testing.assert_allclose(xd[:S.nowned].cpu().numpy(),
                                xs[:S.nowned].cpu().numpy(),
                                rtol=1e-9, atol=1e-11)
+
+
+def test_bsell_dof4_detected_and_correct():
+    """dof=4 dense-block matrices take the Block-SELL path (dispatch
+    covers 2/3/4; auto-detect previously only tried 3 and 2)."""
+    import numpy as np
+    import scipy.sparse.linalg as spla
+    import torch
+
+    from acg_amd.gen import STENCIL_27PT_3D, stencil_global
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.solvers.hip import CGSolverHIP
+
+    spec = dict(STENCIL_27PT_3D)
+    spec["dof"] = 4
+    M = np.array([[1.0, .3, .2, .1], [.3, 1.0, .3, .2],
+                  [.2, .3, 1.0, .3], [.1, .2, .3, 1.0]])
+    spec["offblock"] = M
+    spec["diagblock"] = 60.0 * np.eye(4) + 0.5 * (M - np.eye(4))
+    A = stencil_global(8, 8, 8, spec)
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    solver = CGSolverHIP(S, device="cuda:0")
+    assert solver.bsell is not None and solver.bsell[3] == 4
+    rng = np.random.default_rng(0)
+    b_np = rng.standard_normal(S.nowned)
+    b = torch.from_numpy(b_np).cuda()
+    x = torch.zeros(S.nowned + S.nghost, dtype=torch.float64, device="cuda")
+    res = solver.solve(b, x, maxits=400, res_rtol=1e-11)
+    assert res.converged
+    x_ref = spla.spsolve(A.to_scipy_full().tocsc(), b_np)
+    np.testing.assert_allclose(x[:S.nowned].cpu().numpy(), x_ref,
+                               rtol=1e-7, atol=1e-9)
