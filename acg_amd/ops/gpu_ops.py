@@ -54,12 +54,18 @@ def alloc_partials(device) -> torch.Tensor:
 def pick_lanes(mean_nnz_per_row: float) -> int:
     """Lanes-per-row heuristic for the CSR vector kernel (CDNA4: wave=64).
 
-    Measured on MI355X (Queen-shaped, ~80 nnz/row): 16 lanes (≈5 nnz/lane)
-    beats 64 lanes (≈1.2 nnz/lane) by ~25% — deep-enough per-lane runs
-    amortize the shuffle reduction and keep loads contiguous."""
-    for lanes in (4, 8, 16, 32):
-        if mean_nnz_per_row <= lanes * 6.0:
-            return lanes
+    MEASURED mapping (tools/lanes_sweep.py, fixed-length rows, 30M nnz):
+    the optimum sits near 1-2 nnz/lane, NOT the round-1 ~6 rule-of-thumb
+    (len-24 rows: 32 lanes 372 us vs 4 lanes 557; len-16: 16 lanes 432
+    vs 4 lanes 588).  len<=8 -> 8, <=16 -> 16, <=96 -> 32, else 64."""
+    if mean_nnz_per_row <= 4.0:
+        return 4
+    if mean_nnz_per_row <= 8.0:
+        return 8
+    if mean_nnz_per_row <= 16.0:
+        return 16
+    if mean_nnz_per_row <= 96.0:
+        return 32
     return 64
 
 
@@ -98,14 +104,14 @@ def build_row_bins(rowptr, max_lanes: int = 64):
     bins = []
     lo = 0
     # descending order: long rows first; thresholds from the high end
-    for lanes in (64, 32, 16, 8, 4):
+    # (measured mapping, see pick_lanes: >96 -> 64, >16 -> 32, >8 -> 16,
+    # >4 -> 8, rest 4)
+    for lanes, thresh in ((64, 96), (32, 16), (16, 8), (8, 4), (4, None)):
         if lanes > max_lanes:
             continue
-        if lanes == 4:
+        if thresh is None:
             hi = len(slens)
         else:
-            # rows longer than (lanes/2)*6 stay in this bin
-            thresh = (lanes // 2) * 6
             hi = int(np.searchsorted(-slens, -thresh - 1, side="right"))
         if hi > lo:
             bins.append((lo, hi - lo, lanes))
@@ -129,12 +135,11 @@ def build_sellcsr_hybrid(rowptr, colidx, vals, cut: int = 192,
     longlens = lens[long_rows.astype(np.int64)]  # descending
     bins = []
     lo = 0
-    for lanes in (64, 32, 16, 8, 4):
-        if lanes == 4:
+    for lanes, thresh in ((64, 96), (32, 16), (16, 8), (8, 4), (4, None)):
+        if thresh is None:
             hi = len(longlens)
         else:
-            hi = int(np.searchsorted(-longlens, -((lanes // 2) * 6) - 1,
-                                     side="right"))
+            hi = int(np.searchsorted(-longlens, -thresh - 1, side="right"))
         if hi > lo:
             bins.append((lo, hi - lo, lanes))
             lo = hi

@@ -73,16 +73,19 @@ def test_build_row_bins_partitions_all_rows():
     total = sum(c for _, c, _ in bins)
     assert total == n
     lens = np.diff(S.A_rowptr)
-    # bin lane assignment follows the ~6 nnz/lane rule and rows are
+    # bin lane assignment follows the MEASURED mapping (lanes_sweep:
+    # >96 -> 64, >16 -> 32, >8 -> 16, >4 -> 8, rest 4); rows are
     # longest-first within the global order
     slens = lens[rowlist.astype(np.int64)]
     assert np.all(np.diff(slens) <= 0)
+    upper = {64: None, 32: 96, 16: 16, 8: 8, 4: 4}
+    lower = {64: 96, 32: 16, 16: 8, 8: 4, 4: None}
     for start, count, lanes in bins:
         seg = slens[start:start + count]
-        if lanes < 64:
-            assert seg.max() <= lanes * 6
-        if lanes > 4:
-            assert seg.min() > (lanes // 2) * 6
+        if upper[lanes] is not None:
+            assert seg.max() <= upper[lanes]
+        if lower[lanes] is not None:
+            assert seg.min() > lower[lanes]
 
 
 def test_hybrid_torch_fallback_matches_spmv():
