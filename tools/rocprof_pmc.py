@@ -33,11 +33,19 @@ def main() -> int:
         idcol = "id" if "id" in cols else cols[0]
         for pid, nm in cur.execute(f"SELECT {idcol}, {namecol} FROM {pmc_info}"):
             pmc_names[pid] = str(nm)
-    # event_id -> kernel name
-    q = f"""SELECT d.event_id, s.string FROM {kd} d
-            JOIN {ki} k ON d.kernel_id = k.id
-            JOIN {strt} s ON k.display_name = s.id"""
-    ev2k = dict(cur.execute(q))
+    # event_id -> kernel name (display_name holds the string directly in
+    # pmc-mode DBs; string-table indirection only exists in trace DBs)
+    try:
+        q = f"""SELECT d.event_id, s.string FROM {kd} d
+                JOIN {ki} k ON d.kernel_id = k.id
+                JOIN {strt} s ON k.display_name = s.id"""
+        ev2k = dict(cur.execute(q))
+        if ev2k and all(isinstance(v, int) for v in ev2k.values()):
+            raise ValueError
+    except Exception:
+        q = f"""SELECT d.event_id, k.display_name FROM {kd} d
+                JOIN {ki} k ON d.kernel_id = k.id"""
+        ev2k = dict(cur.execute(q))
     agg = defaultdict(float)
     cnt = defaultdict(int)
     for eid, pid, val in cur.execute(
