@@ -40,7 +40,7 @@ def main() -> int:
                 JOIN {ki} k ON d.kernel_id = k.id
                 JOIN {strt} s ON k.display_name = s.id"""
         ev2k = dict(cur.execute(q))
-        if ev2k and all(isinstance(v, int) for v in ev2k.values()):
+        if not ev2k or all(isinstance(v, int) for v in ev2k.values()):
             raise ValueError
     except Exception:
         q = f"""SELECT d.event_id, k.display_name FROM {kd} d
