@@ -109,7 +109,7 @@ def sell_from_csr(rowptr, colidx, vals, C: int = 64, sigma: int = 1):
 
 
 def sellcsr_split(rowptr, colidx, vals, cut: int = 192, C: int = 64,
-                  window: int = 0):
+                  window: int = 0, bucket: int = 0):
     """Split a CSR matrix by row length for the SELL+CSR hybrid format.
 
     Rows with len <= ``cut`` (the regular majority) go into a sigma-SELL
@@ -150,6 +150,17 @@ def sellcsr_split(rowptr, colidx, vals, cut: int = 192, C: int = 64,
             sub = np.argsort(-lens[sr[w0:w1]], kind="stable")
             out[w0:w1] = sr[w0:w1][sub]
         short_rows = out
+    elif bucket and len(short_rows):
+        # (length-bucket desc, first column asc): padding bounded by the
+        # bucket width while rows inside a slice share nearby columns --
+        # their 128 B x-line fetches overlap (the gather is the measured
+        # bound: 50% L2 miss with 8-of-128-byte line use, see
+        # profiles/irr_pmc_tcc_r02.txt)
+        sl = lens[short_rows]
+        firstcol = np.where(sl > 0, colidx[rowptr[short_rows]], 0)
+        bkt = (sl + bucket - 1) // bucket
+        order2 = np.lexsort((firstcol, -bkt))
+        short_rows = short_rows[order2]
     nshort = len(short_rows)
     nslices = (nshort + C - 1) // C
     # SELL arrays over the short subset, slice lengths from the sorted order
