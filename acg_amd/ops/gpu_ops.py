@@ -120,7 +120,7 @@ def build_row_bins(rowptr, max_lanes: int = 64):
 
 
 def build_sellcsr_hybrid(rowptr, colidx, vals, cut: int = 192,
-                         window: int = 0, bucket: int = 0):
+                         window: int = 0, bucket: int = 8):
     """Host prep for the SELL+CSR hybrid: short rows (len <= cut) as
     sigma-sorted SELL, long rows as a longest-first 64/32-lane binned CSR
     list.  Returns (sellptr, cols, svals, perm, rowlist_long, bins) numpy

@@ -109,7 +109,7 @@ def sell_from_csr(rowptr, colidx, vals, C: int = 64, sigma: int = 1):
 
 
 def sellcsr_split(rowptr, colidx, vals, cut: int = 192, C: int = 64,
-                  window: int = 0, bucket: int = 0):
+                  window: int = 0, bucket: int = 8):
     """Split a CSR matrix by row length for the SELL+CSR hybrid format.
 
     Rows with len <= ``cut`` (the regular majority) go into a sigma-SELL

@@ -168,7 +168,7 @@ class CGSolverHIP:
             if cut is None:
                 cut = int(os.environ.get("ACG_HYBRID_CUT", "192"))
             window = int(os.environ.get("ACG_HYBRID_WINDOW", "0"))
-            bucket = int(os.environ.get("ACG_HYBRID_BUCKET", "0"))
+            bucket = int(os.environ.get("ACG_HYBRID_BUCKET", "8"))
             sp_, cols, svals, perm, rowlist, bins = ops.build_sellcsr_hybrid(
                 L.A_rowptr, L.A_colidx, L.A_vals, cut=cut, window=window,
                 bucket=bucket)
