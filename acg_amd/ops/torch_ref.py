@@ -142,6 +142,15 @@ def sellcsr_split(rowptr, colidx, vals, cut: int = 192, C: int = 64,
     nlong = int(np.searchsorted(-slens, -int(cut) - 1, side="right"))
     long_rows = order[:nlong].astype(np.int32)
     short_rows = order[nlong:].astype(np.int64)  # descending lengths
+    if bucket and nlong and cut >= 96:
+        # same line-sharing ordering for the tail.  Only when every long
+        # row gets 64 lanes (cut >= 96): the bin builder derives bin
+        # boundaries by searchsorted on descending lengths, which this
+        # reorder would break for multi-bin tails.
+        ll = lens[long_rows.astype(np.int64)]
+        fc = colidx[rowptr[long_rows.astype(np.int64)]]
+        bw = max(bucket * 8, 64)
+        long_rows = long_rows[np.lexsort((fc, -((ll + bw - 1) // bw)))]
     if window and len(short_rows):
         sr = np.sort(short_rows)  # original index order
         out = np.empty_like(sr)
