@@ -47,8 +47,9 @@ def make_parser() -> argparse.ArgumentParser:
                    choices=("block", "rgb", "ml", "auto"), default="auto")
     p.add_argument("--seed", type=int, default=0, help="partitioner seed")
     p.add_argument("--solver", default=None,
-                   choices=("acg", "acg-pipelined", "acg-device", "cpu",
-                            "cpu-pipelined", "scipy", "scipy-pipelined",
+                   choices=("acg", "acg-pipelined", "acg-device",
+                            "acg-jacobi", "cpu", "cpu-pipelined",
+                            "cpu-jacobi", "scipy", "scipy-pipelined",
                             "petsc", "petsc-pipelined"),
                    help="default: acg on GPU, cpu otherwise.  petsc[-pipelined] "
                         "are accepted for aCG compatibility and run the scipy "
@@ -297,6 +298,9 @@ def main(argv=None) -> int:
                 if solver_name == "acg-pipelined":
                     solver.solve_pipelined(b, x.clone(), maxits=args.warmup,
                                            res_rtol=0.0)
+                elif solver_name == "acg-jacobi":
+                    solver.solve_jacobi(b, x.clone(), maxits=args.warmup,
+                                        res_rtol=0.0)
                 elif solver_name == "acg-device":
                     # warm the actual cooperative kernel (reference warms the
                     # kernel it will time, cg-kernels-hip.hip:1925-1989)
@@ -308,6 +312,10 @@ def main(argv=None) -> int:
                 res = solver.solve_pipelined(b, x, maxits=args.max_iterations,
                                              res_atol=args.residual_atol,
                                              res_rtol=args.residual_rtol)
+            elif solver_name == "acg-jacobi":
+                res = solver.solve_jacobi(b, x, maxits=args.max_iterations,
+                                          res_atol=args.residual_atol,
+                                          res_rtol=args.residual_rtol)
             elif solver_name == "acg-device":
                 res = solver.solve_device(b, x, maxits=args.max_iterations,
                                           res_atol=args.residual_atol,
@@ -326,6 +334,10 @@ def main(argv=None) -> int:
                 res = solver.solve_pipelined(b, x, maxits=args.max_iterations,
                                              res_atol=args.residual_atol,
                                              res_rtol=args.residual_rtol)
+            elif solver_name == "cpu-jacobi":
+                res = solver.solve_jacobi(b, x, maxits=args.max_iterations,
+                                          res_atol=args.residual_atol,
+                                          res_rtol=args.residual_rtol)
             else:
                 res = solver.solve(b, x, maxits=args.max_iterations,
                                    res_atol=args.residual_atol,
