@@ -121,3 +121,92 @@ def test_cli_gpu_jacobi(tmp_path, monkeypatch, capsys):
                    "--residual-rtol", "1e-9", "-q"])
     out = capsys.readouterr()
     assert rc == 0, out.err
+
+
+# ---- jacobi SCALING (composes preconditioning with every solver) ----
+
+def test_jacobi_scale_pipelined_cpu_matches_scipy():
+    """Scaled system + plain PIPELINED CG == preconditioned solve: same
+    iteration-count class as explicit PCG, back-transformed solution
+    matches scipy."""
+    import scipy.sparse.linalg as spla
+
+    from acg_amd.solvers.precond import jacobi_scale_system
+
+    A = _ill_conditioned()
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    rng = np.random.default_rng(0)
+    b_np = rng.standard_normal(S.nowned)
+    Ss, s = jacobi_scale_system(S)
+    # scaled diagonal is exactly 1
+    rows = np.repeat(np.arange(Ss.nowned), np.diff(Ss.A_rowptr))
+    dm = rows == Ss.A_colidx
+    np.testing.assert_allclose(Ss.A_vals[dm], 1.0, rtol=1e-14)
+    solver = CGSolverCPU(Ss)
+    bs = torch.from_numpy(b_np * s)
+    xs = torch.zeros(Ss.nowned, dtype=torch.float64)
+    r = solver.solve_pipelined(bs, xs, maxits=3000, res_rtol=1e-10)
+    assert r.converged
+    # iteration-count class of explicit PCG (same spectrum)
+    plain = CGSolverCPU(S)
+    xj = torch.zeros(S.nowned, dtype=torch.float64)
+    rj = plain.solve_jacobi(torch.from_numpy(b_np), xj, maxits=3000,
+                            res_rtol=1e-10)
+    assert r.niterations <= rj.niterations * 2
+    x = xs.numpy() * s
+    x_ref = spla.spsolve(A.to_scipy_full().tocsc(), b_np)
+    np.testing.assert_allclose(x, x_ref, rtol=1e-6, atol=1e-8)
+
+
+def test_cli_jacobi_scale_pipelined(tmp_path, monkeypatch, capsys):
+    from acg_amd import cli
+    from acg_amd.io.mtx import MtxFile, write_mtx
+
+    A = _ill_conditioned(1500, seed=11)
+    rows = np.repeat(np.arange(A.n), np.diff(A.rowptr))
+    m = MtxFile(object="matrix", format="coordinate", field_="real",
+                symmetry="symmetric", nrows=A.n, ncols=A.n,
+                nnz=A.nnz_stored, rowidx=rows, colidx=A.colidx, a=A.vals)
+    p = tmp_path / "A.mtx"
+    write_mtx(p, m)
+    monkeypatch.delenv("WORLD_SIZE", raising=False)
+    monkeypatch.delenv("RANK", raising=False)
+    rc = cli.main([str(p), "--solver", "cpu-pipelined", "--jacobi-scale",
+                   "--max-iterations", "2000", "--residual-rtol", "1e-10"])
+    out = capsys.readouterr()
+    assert rc == 0, out.err
+    # output is the UNSCALED solution
+    got = np.array([float(v) for v in out.out.strip().splitlines()[2:]])
+    import scipy.sparse.linalg as spla
+
+    x_ref = spla.spsolve(A.to_scipy_full().tocsc(), np.ones(A.n))
+    np.testing.assert_allclose(got, x_ref, rtol=1e-5, atol=1e-7)
+
+
+@pytest.mark.gpu
+def test_gpu_jacobi_scale_pipelined_matches_explicit_pcg():
+    from acg_amd.solvers.hip import CGSolverHIP
+    from acg_amd.solvers.precond import jacobi_scale_system
+
+    A = _ill_conditioned()
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    rng = np.random.default_rng(0)
+    b_np = rng.standard_normal(S.nowned)
+    Ss, s = jacobi_scale_system(S)
+    gpu = CGSolverHIP(Ss, device="cuda:0")
+    bs = torch.from_numpy(b_np * s).cuda()
+    xs = torch.zeros(Ss.nowned + Ss.nghost, dtype=torch.float64,
+                     device="cuda")
+    r = gpu.solve_pipelined(bs, xs, maxits=3000, res_rtol=1e-10)
+    assert r.converged
+    x = xs[:S.nowned].cpu().numpy() * s
+    import scipy.sparse.linalg as spla
+
+    x_ref = spla.spsolve(A.to_scipy_full().tocsc(), b_np)
+    np.testing.assert_allclose(x, x_ref, rtol=1e-6, atol=1e-8)
+    # explicit-M PCG iteration class
+    plain = CGSolverHIP(S, device="cuda:0")
+    xj = torch.zeros_like(xs)
+    rj = plain.solve_jacobi(torch.from_numpy(b_np).cuda(), xj, maxits=3000,
+                            res_rtol=1e-10)
+    assert r.niterations <= rj.niterations * 2
