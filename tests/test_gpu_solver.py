@@ -419,7 +419,7 @@ def test_gpu_solver_survives_residual_underflow(problem, method):
     assert rel < 1e-10, rel
 
 
-@pytest.mark.parametrize("solver_name", ["acg", "acg-pipelined", "acg-device"])
+@pytest.mark.parametrize("solver_name", ["acg", "acg-pipelined", "acg-device", "acg-jacobi"])
 def test_cli_gpu_end_to_end(tmp_path, capsys, monkeypatch, solver_name):
     """Full CLI pipeline on a real GPU: mtx file -> assembly -> extraction
     -> CGSolverHIP (all three GPU solver types) -> solution to stdout."""
