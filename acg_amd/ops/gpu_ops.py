@@ -340,6 +340,15 @@ def cg_fused_update(r: torch.Tensor, x: torch.Tensor, p: torch.Tensor,
                       n, scal.data_ptr(), partials.data_ptr(), _stream())
 
 
+def pcg_fused_update(r, x, p, t, z, dinv, scal, partials, n: int) -> None:
+    """Jacobi-PCG fused epilogue: alpha = rz/pt (device); r -= alpha t;
+    x += alpha p; z = dinv*r; rotates rz -> RR_PREV and publishes the new
+    rz in S_RR plus the TRUE (r,r) in S_GAMMA."""
+    K.pcg_fused_update(r.data_ptr(), x.data_ptr(), p.data_ptr(), t.data_ptr(),
+                       z.data_ptr(), dinv.data_ptr(), n, scal.data_ptr(),
+                       partials.data_ptr(), _stream())
+
+
 def sell_pipe(sellptr, cols, vals, nrows_pass: int, rowbase: int,
               border_base: int, w_old, qpart, z, t, p, x, r, w_new,
               scal, first: bool, partials, partials_off: int,

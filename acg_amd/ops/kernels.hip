@@ -477,6 +477,58 @@ k_cg_finalize(const double* __restrict__ partials, int nblocks, double* scal) {
     }
 }
 
+// Jacobi-PCG fused iteration epilogue (beyond reference -- aCG has no
+// preconditioning): alpha = rz/(p,t) from the device slab, then ONE pass
+// computes r -= alpha t, x += alpha p, z = dinv .* r, and BOTH next
+// scalars rz' = (r,z) and rr' = (r,r).  Replaces axpy_ratio x2 + mul +
+// prep + dot x2 (6 launches, ~13n traffic) with 1 launch + a finalize.
+__global__ void __launch_bounds__(BLOCK)
+k_pcg_fused_update(double* __restrict__ r, double* __restrict__ x,
+                   const double* __restrict__ p, const double* __restrict__ t,
+                   double* __restrict__ z, const double* __restrict__ dinv,
+                   long n, const double* __restrict__ scal,
+                   double* __restrict__ partials) {
+    const double alpha = safe_div(scal[S_RR], scal[S_PT]);  // S_RR = rz
+    double arz = 0.0, arr = 0.0;
+    const long stride = (long)gridDim.x * BLOCK;
+    for (long i = (long)blockIdx.x * BLOCK + threadIdx.x; i < n; i += stride) {
+        const double rn = r[i] - alpha * ld_nt(t + i);
+        r[i] = rn;
+        __builtin_nontemporal_store(ld_nt(x + i) + alpha * p[i], x + i);
+        const double zi = dinv[i] * rn;
+        z[i] = zi;
+        arz += rn * zi;
+        arr += rn * rn;
+    }
+    arz = block_reduce(arz);
+    __syncthreads();  // block_reduce reuses its LDS scratch
+    arr = block_reduce(arr);
+    if (threadIdx.x == 0) {
+        partials[blockIdx.x] = arz;
+        partials[MAXG + blockIdx.x] = arr;
+    }
+}
+
+// finalize: rotate rz -> rz_prev (S_RR -> S_RR_PREV), publish rz' in
+// S_RR and the true rr' in S_GAMMA
+__global__ void __launch_bounds__(BLOCK)
+k_pcg_finalize(const double* __restrict__ partials, int nblocks,
+               double* scal) {
+    double vz = 0.0, vr = 0.0;
+    for (int i = threadIdx.x; i < nblocks; i += BLOCK) {
+        vz += partials[i];
+        vr += partials[MAXG + i];
+    }
+    vz = block_reduce(vz);
+    __syncthreads();
+    vr = block_reduce(vr);
+    if (threadIdx.x == 0) {
+        scal[S_RR_PREV] = scal[S_RR];
+        scal[S_RR] = vz;
+        scal[S_GAMMA] = vr;
+    }
+}
+
 // dot: partials[b] = block sum of x[i]*y[i]
 __global__ void __launch_bounds__(BLOCK)
 k_dot(const double* __restrict__ x, const double* __restrict__ y, long n,
@@ -1837,6 +1889,20 @@ void daypx_ratio(uintptr_t y, uintptr_t x, long n, uintptr_t scal, int num, int 
     check_hip("daypx_ratio");
 }
 
+void pcg_fused_update(uintptr_t r, uintptr_t x, uintptr_t p, uintptr_t t,
+                      uintptr_t z, uintptr_t dinv, long n, uintptr_t scal,
+                      uintptr_t partials, uintptr_t stream) {
+    long blocks = elem_grid(n);
+    hipLaunchKernelGGL(k_pcg_fused_update, dim3((unsigned)blocks), dim3(BLOCK),
+                       0, S(stream), (double*)r, (double*)x, (const double*)p,
+                       (const double*)t, (double*)z, (const double*)dinv,
+                       n, (const double*)scal, (double*)partials);
+    check_hip("pcg_fused_update");
+    hipLaunchKernelGGL(k_pcg_finalize, dim3(1), dim3(BLOCK), 0, S(stream),
+                       (const double*)partials, (int)blocks, (double*)scal);
+    check_hip("pcg_finalize");
+}
+
 void cg_fused_update(uintptr_t r, uintptr_t x, uintptr_t p, uintptr_t t, long n,
                      uintptr_t scal, uintptr_t partials, uintptr_t stream) {
     long blocks = elem_grid(n);
@@ -1992,6 +2058,7 @@ PYBIND11_MODULE(_acg_kernels, m) {
     m.def("axpy_ratio", &axpy_ratio);
     m.def("daypx_ratio", &daypx_ratio);
     m.def("cg_fused_update", &cg_fused_update);
+    m.def("pcg_fused_update", &pcg_fused_update);
     m.def("pipelined_fused", &pipelined_fused);
     m.def("sell_pipe", &sell_pipe);
     m.def("pipelined_finalize", &pipelined_finalize);
