@@ -22,6 +22,6 @@ Layer map (mirrors reference SURVEY.md §1):
   cli.py      acg-compatible driver          (hip/acg-hip.c)
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from .utils.errors import AcgError  # noqa: F401

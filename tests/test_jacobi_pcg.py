@@ -236,3 +236,26 @@ def test_cli_jacobi_scale_with_manufactured(tmp_path, monkeypatch, capsys):
     line = next(l for l in out.err.splitlines() if "manufactured" in l)
     enorm = float(line.split("=")[1].split("(")[0])
     assert enorm < 1e-6, line
+
+
+def test_jacobi_error_paths():
+    from acg_amd.solvers.precond import jacobi_scale_system
+    from acg_amd.core.symcsr import SymCSRMatrix
+    from acg_amd.utils.errors import AcgError
+
+    # zero diagonal -> clean error (both transform and explicit PCG)
+    A = SymCSRMatrix.from_coo(3, [0, 1, 0], [0, 1, 2], [1.0, 1.0, 0.5])
+    S = extract_subdomains(A, partition_rows(A, 1), 1)[0]
+    with pytest.raises(AcgError):
+        jacobi_scale_system(S)
+    solver = CGSolverCPU(S)
+    with pytest.raises(AcgError):
+        solver.solve_jacobi(torch.ones(3, dtype=torch.float64),
+                            torch.zeros(3, dtype=torch.float64), maxits=5)
+    # ghosts without a communicator -> clean error
+    B = powerlaw_spd(600, mean_nnz=10, seed=1)
+    part = partition_rows(B, 2, method="block")
+    S2 = extract_subdomains(B, part, 2)[0]
+    assert S2.nghost > 0
+    with pytest.raises(AcgError):
+        jacobi_scale_system(S2, comm=None)
