@@ -180,6 +180,7 @@ def main(argv=None) -> int:
             if args.partition:
                 part = read_partition_file(args.partition, A.n,
                                            binary=args.binary_partition,
+                                           gzipped=args.gzip,
                                            idxsize=args.idxsize)
                 if int(part.max()) >= nparts:
                     raise AcgError(ErrCode.INVALID_VALUE,
