@@ -99,6 +99,10 @@ def _body_cg_gpu_pipelined(comm):
     return _solve_gpu(comm, "solve_pipelined")
 
 
+def _body_cg_gpu_jacobi(comm):
+    return _solve_gpu(comm, "solve_jacobi")
+
+
 def _check_vs_direct(results):
     spec = queen_like_spec(3)
     A = stencil_global(5, 5, 8, spec)
@@ -164,6 +168,12 @@ def test_gpu_halo_exchange_2proc():
 
 def test_gpu_cg_classic_2proc():
     _check_vs_direct(_run_dist("_body_cg_gpu", world=2, port=29611))
+
+
+def test_gpu_cg_jacobi_2proc():
+    """2-process Jacobi-PCG on one GPU (gloo-staged): multi-rank
+    (r,z)/(p,t) allreduces + the fused PCG epilogue."""
+    _check_vs_direct(_run_dist("_body_cg_gpu_jacobi", world=2, port=29647))
 
 
 def test_gpu_cg_pipelined_2proc():
