@@ -187,6 +187,11 @@ def _refine_kway(u: np.ndarray, v: np.ndarray, w: np.ndarray, n: int,
     hi = (1.0 + eps) * target
     lo = (1.0 - eps) * target
     part = part.astype(np.int32).copy()
+    if n * nparts > (1 << 28):
+        # the per-pass W table is n*nparts doubles; for huge part counts
+        # at fine levels skip refinement there (coarser levels, where
+        # most of the cut is decided, still refine)
+        return part
     rng_n = np.arange(n)
     cut_prev = None
     for _ in range(passes):
