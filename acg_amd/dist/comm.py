@@ -125,7 +125,10 @@ class Comm:
 
     def _send_array(self, a, dst: int) -> None:
         a = np.ascontiguousarray(a)
-        code = next(i for i, d in enumerate(self._DTYPES) if d == a.dtype)
+        code = next((i for i, d in enumerate(self._DTYPES) if d == a.dtype),
+                    None)
+        if code is None:
+            raise TypeError(f"scatter transport: unsupported dtype {a.dtype}")
         hdr = torch.tensor([code, a.size], dtype=torch.int64)
         self._dist.send(hdr, dst, group=self._host_group)
         if a.size:
@@ -186,7 +189,9 @@ class Comm:
                         self._send_array(getattr(S.halo, nm), p)
                     del S
             except Exception:
-                poison = torch.tensor([-1, -1], dtype=torch.int64)
+                # poison header must MATCH the receivers' 6-element meta
+                # recv (gloo errors on size mismatch)
+                poison = torch.full((6,), -1, dtype=torch.int64)
                 for p in range(self.size):
                     if p != src and p not in sent:
                         self._dist.send(poison, p, group=self._host_group)

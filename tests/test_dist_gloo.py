@@ -313,3 +313,38 @@ def test_distributed_cg_matches_serial(body, port):
     x_ref = spla.spsolve(X.tocsc(), b_global)
     for rank, (owned_global, xloc, nit) in results.items():
         np.testing.assert_allclose(xloc, x_ref[owned_global], rtol=1e-6, atol=1e-8)
+
+
+def _body_scatter_poison(comm):
+    """Root-side factory failure mid-scatter must raise on EVERY rank
+    (matching-size poison headers), not hang the unsent ranks."""
+    from acg_amd.part import extract_subdomains, partition_rows
+    from acg_amd.part.subdomain import SubdomainExtractor
+    from acg_amd.utils.errors import AcgError
+
+    A = stencil_global(6, 6, 6, STENCIL_27PT_3D)
+    part = partition_rows(A, comm.size, method="rgb", seed=1)
+    if comm.rank == 0:
+        ex = SubdomainExtractor(A, part, comm.size)
+
+        def factory(p):
+            if p == 1:
+                raise RuntimeError("boom at part 1")
+            return ex.build(p)
+
+        try:
+            comm.scatter_systems(factory)
+            return ("no-raise",)
+        except RuntimeError:
+            return ("raised",)
+    try:
+        comm.scatter_systems(None)
+        return ("no-raise",)
+    except AcgError:
+        return ("raised",)
+
+
+def test_scatter_poison_ws4():
+    results = _run_dist("_body_scatter_poison", world=4, port=29610)
+    for rank, (status,) in results.items():
+        assert status == "raised", f"rank {rank}: {status}"
